@@ -1,0 +1,102 @@
+"""Compute-op dispatch: CDNA4 HIP kernels on GPU, eager PyTorch on CPU.
+
+The hot ops (the Beta-grid P(best) integral and the fused EIG pipeline,
+>95% of wall time in an acquisition step - reference hot loop at
+coda/coda.py:235-281) dispatch to hand-written gfx950 HIP kernels from
+`coda_amd/ops/hip/` when the input lives on a ROCm device. If the compiled
+extension is missing on a GPU machine the op raises instead of silently
+falling back to eager (set CODA_AMD_ALLOW_EAGER=1 to override, e.g. for
+kernel-vs-eager numerics comparisons on the GPU).
+
+Cheap glue ops (argmax, masks, prior construction) run as eager
+PyTorch-ROCm everywhere.
+"""
+from __future__ import annotations
+
+import os
+
+import torch
+
+from . import reference
+from .reference import (  # re-export cheap ops + constants
+    PBEST_NUM_POINTS, GRID_LO, GRID_HI, EPS_PROB, LOG_CLAMP,
+    consensus, confusion_prior, init_dirichlets, dirichlet_to_beta,
+    pi_hat_partial, pi_hat_normalize, beta_grid_pdf_cdf, hypothetical_betas,
+    mixture_entropy, pred_classes, disagreement_mask,
+    accuracy_losses, entropy_acquisition, vma_pairwise, lure_weights,
+)
+
+_ext = None
+_ext_err = None
+
+
+def _load_ext():
+    """Load the in-tree HIP extension (coda_amd/ops/_coda_hip*.so)."""
+    global _ext, _ext_err
+    if _ext is not None or _ext_err is not None:
+        return _ext
+    try:
+        from . import _coda_hip  # noqa: F401  (built by build_hip.py, in-tree)
+        _ext = _coda_hip
+    except ImportError as e:  # pragma: no cover - exercised on GPU boxes
+        _ext_err = e
+    return _ext
+
+
+def hip_available() -> bool:
+    return _load_ext() is not None
+
+
+def _want_hip(t: torch.Tensor) -> bool:
+    if not t.is_cuda:
+        return False
+    if _load_ext() is not None:
+        return True
+    if os.environ.get("CODA_AMD_ALLOW_EAGER") == "1":
+        return False
+    raise RuntimeError(
+        "coda_amd HIP extension is not built but input is on a ROCm device. "
+        "Run `python build_hip.py` (or __graft_entry__.build()) to compile "
+        f"the gfx950 kernels. Import error: {_ext_err}")
+
+
+def pbest_from_beta(alpha: torch.Tensor, beta: torch.Tensor,
+                    num_points: int = PBEST_NUM_POINTS,
+                    return_unnormalized: bool = False) -> torch.Tensor:
+    """P(model h is best) per row from diagonal Beta params: (R,H) -> (R,H)."""
+    if _want_hip(alpha) and not return_unnormalized:
+        return _ext.pbest_from_beta(alpha.contiguous(), beta.contiguous(),
+                                    int(num_points))
+    return reference.pbest_from_beta(alpha, beta, num_points,
+                                     return_unnormalized)
+
+
+def eig_chunk(alpha_cc: torch.Tensor, beta_cc: torch.Tensor,
+              chunk_classes: torch.Tensor, pbest_before: torch.Tensor,
+              pi_hat: torch.Tensor, pi_hat_xi_chunk: torch.Tensor,
+              mixture0: torch.Tensor, H_before: torch.Tensor,
+              update_weight: float = 1.0,
+              num_points: int = PBEST_NUM_POINTS) -> torch.Tensor:
+    """Fused per-chunk EIG: hypothetical Beta updates -> P(best) rows -> EIG.
+
+    alpha_cc/beta_cc: (H, C) current diagonal Betas.
+    chunk_classes: (B, H) argmax class of each model on each candidate.
+    pbest_before: (C, H); pi_hat: (C,); pi_hat_xi_chunk: (B, C);
+    mixture0: (H,); H_before: () scalar tensor.  Returns eig: (B,).
+
+    Reference semantics: coda/coda.py:261-278 (K10 + K6-K9 + K11).
+    """
+    if _want_hip(alpha_cc):
+        return _ext.eig_chunk(
+            alpha_cc.contiguous(), beta_cc.contiguous(),
+            chunk_classes.to(torch.int32).contiguous(),
+            pbest_before.contiguous(), pi_hat.contiguous(),
+            pi_hat_xi_chunk.contiguous(), mixture0.contiguous(),
+            float(H_before), float(update_weight), int(num_points))
+    a, b = reference.hypothetical_betas(alpha_cc, beta_cc, chunk_classes,
+                                        update_weight)
+    B, C, H = a.shape
+    pbest_hyp = reference.pbest_from_beta(
+        a.reshape(B * C, H), b.reshape(B * C, H), num_points).reshape(B, C, H)
+    return reference.eig_assemble(pbest_hyp, pbest_before, pi_hat,
+                                  pi_hat_xi_chunk, mixture0, H_before)
