@@ -1,0 +1,91 @@
+"""Sharded (world_size=2, gloo, CPU) vs single-process equivalence.
+
+The model axis shards across 2 ranks; the sharded CODA run must produce the
+same selections, q-values and P(best) as the single-process run (SURVEY.md
+section 4 gap (c)). Uses torch.multiprocessing spawn with a file:// init.
+"""
+import os
+import random
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from coda_amd.datasets import Dataset, make_synthetic_task
+
+
+def _single_trajectory(preds, labels, steps=4):
+    from coda_amd import CODA, Oracle
+    from coda_amd.options import LOSS_FNS
+    ds = Dataset.from_tensors(preds, labels, "cpu")
+    oracle = Oracle(ds, LOSS_FNS["acc"])
+    random.seed(0); torch.manual_seed(0)
+    sel = CODA(ds, chunk_size=64)
+    out = {"pbest0": sel.get_pbest(), "choices": []}
+    for _ in range(steps):
+        idx, q = sel.get_next_item_to_label()
+        sel.add_label(idx, oracle(idx), q)
+        out["choices"].append((int(idx), round(float(q), 5)))
+    out["pbest"] = sel.get_pbest()
+    return out
+
+
+def _worker(rank, world, init_file, preds, labels, steps, q):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank))
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=world)
+    try:
+        from coda_amd import CODA, Oracle
+        from coda_amd.options import LOSS_FNS
+        from coda_amd.parallel import Comm
+        comm = Comm(rank=rank, world=world, device=torch.device("cpu"))
+        ds = Dataset.from_tensors(preds, labels, "cpu",
+                                  shard=(rank, world))
+        oracle = Oracle(ds, LOSS_FNS["acc"])
+        random.seed(0); torch.manual_seed(0)
+        sel = CODA(ds, comm=comm, chunk_size=64)
+        pbest0 = sel.get_pbest()
+        choices = []
+        for _ in range(steps):
+            idx, qv = sel.get_next_item_to_label()
+            sel.add_label(idx, oracle(idx), qv)
+            choices.append((int(idx), round(float(qv), 5)))
+        pbest = sel.get_pbest()
+        q.put((rank, {"pbest0": pbest0.tolist(), "choices": choices,
+                      "pbest": pbest.tolist()}))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_sharded_equals_single(tmp_path):
+    preds, labels = make_synthetic_task(H=7, N=200, C=4, seed=5)
+    single = _single_trajectory(preds, labels)
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    init_file = str(tmp_path / "pg_init")
+    procs = [ctx.Process(target=_worker,
+                         args=(r, 2, init_file, preds, labels, 4, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, res = q.get(timeout=240)
+        results[rank] = res
+    for p in procs:
+        p.join(timeout=60)
+
+    for rank in (0, 1):
+        res = results[rank]
+        assert res["choices"] == single["choices"], (rank, res["choices"],
+                                                     single["choices"])
+        torch.testing.assert_close(torch.tensor(res["pbest0"]),
+                                   single["pbest0"], rtol=1e-4, atol=1e-6)
+        torch.testing.assert_close(torch.tensor(res["pbest"]),
+                                   single["pbest"], rtol=1e-4, atol=1e-6)
+    # ranks agree with each other exactly on collective-derived outputs
+    assert results[0]["pbest"] == results[1]["pbest"]

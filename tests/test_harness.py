@@ -1,0 +1,74 @@
+"""End-to-end harness: main.py CLI on a synthetic task (CPU)."""
+import os
+import sqlite3
+
+import pytest
+import torch
+
+import main as harness
+from coda_amd.datasets import write_synthetic_task
+from coda_amd import tracking
+
+
+@pytest.fixture()
+def task_dir(tmp_path):
+    write_synthetic_task(str(tmp_path / "data"), name="synthtask",
+                         H=6, N=200, C=4, seed=0)
+    return tmp_path
+
+
+def _run_cli(tmp_path, extra):
+    cwd = os.getcwd()
+    os.chdir(tmp_path)
+    try:
+        tracking.set_tracking_uri("sqlite:///coda.sqlite")
+        tracking._EXPERIMENT = None
+        tracking._RUN_STACK.clear()
+        harness.main(["--task", "synthtask", "--data-dir", "data",
+                      "--device", "cpu"] + extra)
+    finally:
+        os.chdir(cwd)
+
+
+def test_no_mlflow_smoke(task_dir):
+    _run_cli(task_dir, ["--method", "coda", "--iters", "3", "--seeds", "1",
+                        "--no-mlflow", "--chunk-size", "64"])
+    assert not (task_dir / "coda.sqlite").exists()
+
+
+def test_cli_writes_schema_and_skips_finished(task_dir, capsys):
+    _run_cli(task_dir, ["--method", "iid", "--iters", "5", "--seeds", "2"])
+    db = task_dir / "coda.sqlite"
+    assert db.exists()
+    conn = sqlite3.connect(str(db))
+    n_child = conn.execute(
+        "SELECT COUNT(*) FROM tags WHERE key='mlflow.parentRunId'"
+    ).fetchone()[0]
+    assert n_child == 2  # iid is stochastic -> both seeds ran
+    n_metrics = conn.execute(
+        "SELECT COUNT(*) FROM metrics WHERE key='regret'").fetchone()[0]
+    assert n_metrics == 10  # 5 iters x 2 seeds
+    conn.close()
+
+    # re-run: both seeds finished -> skipped
+    _run_cli(task_dir, ["--method", "iid", "--iters", "5", "--seeds", "2"])
+    out = capsys.readouterr().out
+    assert "Skipping" in out
+
+
+def test_deterministic_method_stops_after_seed0(task_dir, capsys):
+    _run_cli(task_dir, ["--method", "uncertainty", "--iters", "3",
+                        "--seeds", "3"])
+    conn = sqlite3.connect(str(task_dir / "coda.sqlite"))
+    n_child = conn.execute(
+        "SELECT COUNT(*) FROM tags WHERE key='mlflow.parentRunId'"
+    ).fetchone()[0]
+    conn.close()
+    assert n_child == 1  # deterministic -> only seed 0
+
+
+@pytest.mark.parametrize("method", ["coda", "activetesting", "vma",
+                                    "model_picker", "uncertainty"])
+def test_all_methods_run_one_iter(task_dir, method):
+    _run_cli(task_dir, ["--method", method, "--iters", "1", "--seeds", "1",
+                        "--no-mlflow", "--chunk-size", "64"])
