@@ -57,8 +57,11 @@ def test_cli_writes_schema_and_skips_finished(task_dir, capsys):
 
 
 def test_deterministic_method_stops_after_seed0(task_dir, capsys):
-    _run_cli(task_dir, ["--method", "uncertainty", "--iters", "3",
-                        "--seeds", "3"])
+    # CODA with no EIG ties is deterministic -> the harness must stop after
+    # seed 0 (main.py:164-168 in the reference). uncertainty/iid are
+    # effectively always stochastic (risk ties at step 0).
+    _run_cli(task_dir, ["--method", "coda", "--iters", "2",
+                        "--seeds", "3", "--chunk-size", "64"])
     conn = sqlite3.connect(str(task_dir / "coda.sqlite"))
     n_child = conn.execute(
         "SELECT COUNT(*) FROM tags WHERE key='mlflow.parentRunId'"
