@@ -1,0 +1,160 @@
+"""Flagship benchmark: acquisition steps/sec of the CODA engine.
+
+Measures the full active-model-selection serving step (EIG acquisition over
+the candidate pool -> oracle label -> Dirichlet posterior update -> pi_hat
+refresh -> best-model P(best)) on the ImageNet-1k-scale config named in
+BASELINE.json (config 2): H=128 candidate models x N=50k points x C=1000
+classes, synthetic prediction tensors, fp32 compute (the reference's compute
+dtype - its loader up-casts storage to fp32, coda/datasets.py:14).
+
+Scaling is STRONG: the 128-model axis shards across the N GPUs
+(RCCL/xGMI), so the whole-job metric is simply steps/sec of the one shared
+selection loop. The reference publishes no absolute throughput numbers
+ => vs_baseline = null.
+
+Contract (driver): `python bench.py --gpus N --steps K --warmup W`; for
+N > 1 the driver launches it under torch.distributed.run with one rank per
+GPU. Rank 0 prints ONE JSON line.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import math
+import os
+import random
+import time
+
+import torch
+
+from coda_amd import CODA, Oracle
+from coda_amd.datasets import Dataset
+from coda_amd.options import LOSS_FNS
+from coda_amd.parallel import init_from_env, get_comm
+
+# Headline config (overridable via env for CPU smoke testing only)
+H_TOTAL = int(os.environ.get("CODA_BENCH_H", 128))
+N_POINTS = int(os.environ.get("CODA_BENCH_N", 50_000))
+C_CLASSES = int(os.environ.get("CODA_BENCH_C", 1000))
+PREFILTER_N = int(os.environ.get("CODA_BENCH_PREFILTER", 256))
+CHUNK = int(os.environ.get("CODA_BENCH_CHUNK", 256))
+
+
+def synth_preds(model_idxs, N, C, device, seed_base=1234):
+    """Per-model deterministic synthetic predictions: rank-independent.
+
+    Model h's tensor depends only on (seed_base + h), so a sharded run sees
+    exactly the data the 1-GPU run sees for the same global model.
+    """
+    out = torch.empty(len(model_idxs), N, C, device=device,
+                      dtype=torch.float32)
+    g = torch.Generator(device=device)
+    labels_g = torch.Generator(device=device)
+    labels_g.manual_seed(seed_base - 1)
+    labels = torch.randint(0, C, (N,), generator=labels_g, device=device)
+    accs = 0.55 + 0.4 * torch.rand(
+        H_TOTAL, generator=torch.Generator().manual_seed(seed_base - 2))
+    for i, h in enumerate(model_idxs):
+        h = int(h)
+        g.manual_seed(seed_base + h)
+        logits = torch.randn(N, C, generator=g, device=device)
+        correct = torch.rand(N, generator=g, device=device) < accs[h]
+        wrong = torch.randint(1, C, (N,), generator=g, device=device)
+        target = torch.where(correct, labels, (labels + wrong) % C)
+        logits.scatter_add_(1, target.unsqueeze(1),
+                            torch.full((N, 1), 4.0, device=device))
+        out[i] = torch.softmax(logits, dim=-1)
+    return out, labels
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=3)
+    ap.add_argument("--warmup", type=int, default=1)
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    comm = init_from_env() if world > 1 else get_comm()
+    n_gpus = max(args.gpus, world)
+
+    if torch.cuda.is_available():
+        device = comm.device or torch.device("cuda", 0)
+    else:
+        device = torch.device("cpu")
+
+    shard = (comm.rank, comm.world) if comm.is_distributed else None
+    model_idxs = list(range(comm.rank, H_TOTAL, comm.world)) \
+        if shard else list(range(H_TOTAL))
+
+    preds, labels = synth_preds(model_idxs, N_POINTS, C_CLASSES, device)
+    ds = Dataset.from_tensors(preds, labels, device, shard=None)
+    ds.total_models = H_TOTAL
+    ds.shard = shard
+    oracle = Oracle(ds, LOSS_FNS["acc"])
+
+    random.seed(0)
+    torch.manual_seed(0)
+    selector = CODA(ds, comm=comm, prefilter_n=PREFILTER_N,
+                    chunk_size=CHUNK)
+
+    def one_step():
+        idx, q = selector.get_next_item_to_label()
+        y = oracle(int(idx))
+        selector.add_label(idx, y, q)
+        selector.get_best_model_prediction()
+
+    def sync():
+        comm.barrier()
+        if device.type == "cuda":
+            torch.cuda.synchronize(device)
+
+    for _ in range(args.warmup):
+        one_step()
+    sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        one_step()
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    et = torch.tensor([elapsed], device=device if device.type == "cuda"
+                      else torch.device("cpu"))
+    if comm.is_distributed:
+        torch.distributed.all_reduce(et, op=torch.distributed.ReduceOp.MAX)
+    elapsed = float(et[0])
+
+    if comm.rank == 0:
+        ms_per_step = 1000.0 * elapsed / args.steps
+        value = args.steps / elapsed
+        print(json.dumps({
+            "metric": "acquisition_steps_per_sec",
+            "value": value,
+            "unit": "steps/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "coda-eig",
+                "task": "imagenet1k-scale",
+                "H_models": H_TOTAL,
+                "N_points": N_POINTS,
+                "C_classes": C_CLASSES,
+                "prefilter_n": PREFILTER_N,
+                "chunk_size": CHUNK,
+                "global_batch": PREFILTER_N,
+                "seq_len": C_CLASSES,
+                "parallelism": f"model-shard{comm.world}",
+            },
+        }))
+
+
+if __name__ == "__main__":
+    main()
