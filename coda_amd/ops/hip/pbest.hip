@@ -6,19 +6,24 @@
 //   p_h = integral pdf_h(x) * prod_{h'!=h} cdf_{h'}(x) dx  on a P=256 grid,
 // followed by a log2-entropy EIG assembly. The reference materializes six
 // (R, H, P) fp32 tensors per chunk and runs a SEQUENTIAL Python loop over P
-// for the trapezoid CDF; these kernels keep the whole pipeline per row
-// inside one workgroup: grid point p <-> thread p, trapezoid CDF as an
-// LDS/wave inclusive scan, the H-coupling as a running per-thread register
-// (slog), and the EIG entropy fused into the epilogue. Global traffic is
-// 2*R*H floats in, R*H (or B*C) floats out - ~3 orders of magnitude less
-// than the eager formulation.
+// for the trapezoid CDF.
 //
-// Numerics: log-pdf evaluated in f64 (2 FMA per point; the f32
-// cancellation at large Beta counts is the reference's main error source),
-// exp/log in f32, the clamp ladder of the reference preserved exactly
-// (cdf clamp 1e-30, log-space clamp +-80, entropy clamp 1e-12).
+// Kernel shape (wave-per-row): one 64-lane wave owns one row; each lane
+// owns 4 consecutive grid points (P = 256 = 64 x 4). The trapezoid CDF is
+// a lane-local running sum + one wave shfl-scan - NO block barriers and no
+// LDS traffic in the hot loop (the first design used thread<->point with a
+// 3-barrier LDS scan per model per pass; this one is barrier-free after
+// staging). The H-coupling (sum_h log cdf) lives in 4 per-lane registers;
+// the EIG entropy epilogue is fused into the hypothetical-update kernel.
+// Global traffic: 2*R*H floats in, R*H (or B*C) floats out - three orders
+// of magnitude below the eager formulation.
 //
-// Workgroup = 256 threads = 4 waves (wave64); P == blockDim == 256.
+// Numerics: log-pdf evaluated in f64 (2 FMA per point; f32 cancellation at
+// large Beta counts is the reference's main error source there), exp/log
+// in f32, the reference's clamp ladder preserved exactly (cdf clamp 1e-30,
+// log-space clamp +-80, entropy clamp 1e-12).
+//
+// Workgroup = 256 threads = 4 waves = 4 independent rows.
 
 #include <torch/extension.h>
 #include <hip/hip_runtime.h>
@@ -26,6 +31,8 @@
 
 #define P_POINTS 256
 #define BLOCK 256
+#define ROWS_PER_BLOCK 4
+#define PTS_PER_LANE 4
 
 namespace {
 
@@ -34,217 +41,225 @@ constexpr double kGridHi = 1.0 - 1e-6;
 constexpr float kEps = 1e-30f;
 constexpr float kLogClamp = 80.0f;
 
-// Inclusive scan of v across the 256-thread block. Uses 4 floats of
-// scratch + two barriers. Scratch may be reused after the call returns
-// (a trailing barrier protects it).
-__device__ __forceinline__ float block_inclusive_scan(float v, float* scr) {
+__device__ __forceinline__ float wave_inclusive_scan(float v) {
     const int lane = threadIdx.x & 63;
-    const int wave = threadIdx.x >> 6;
 #pragma unroll
     for (int off = 1; off < 64; off <<= 1) {
         float n = __shfl_up(v, off, 64);
         if (lane >= off) v += n;
     }
-    if (lane == 63) scr[wave] = v;
-    __syncthreads();
-    float prefix = 0.f;
-#pragma unroll
-    for (int w = 0; w < 4; ++w) {
-        float s = scr[w];
-        if (w < wave) prefix += s;
-    }
-    v += prefix;
-    __syncthreads();
     return v;
 }
 
-// Sum of v across the block; every thread returns the total.
-__device__ __forceinline__ float block_reduce_sum(float v, float* scr) {
-    const int lane = threadIdx.x & 63;
-    const int wave = threadIdx.x >> 6;
+__device__ __forceinline__ float wave_reduce_sum(float v) {
 #pragma unroll
     for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
-    if (lane == 0) scr[wave] = v;
-    __syncthreads();
-    float total = scr[0] + scr[1] + scr[2] + scr[3];
-    __syncthreads();
-    return total;
+    return v;
 }
 
-// One Beta pdf value + its trapezoid-cumulative cdf at this thread's grid
-// point, for row Beta(a, b). lnB precomputed in f64. The pdf of the
-// PREVIOUS grid point comes through LDS (pdfbuf).
-__device__ __forceinline__ void beta_pdf_cdf(
-        double a, double b, double lnB, double lx, double l1mx, float dxf,
-        float* pdfbuf, float* scr, float& pdf, float& cdf) {
-    double t = (a - 1.0) * lx + (b - 1.0) * l1mx - lnB;
-    pdf = __expf((float)t);
-    pdfbuf[threadIdx.x] = pdf;
-    __syncthreads();
-    float prev = (threadIdx.x > 0) ? pdfbuf[threadIdx.x - 1] : 0.f;
-    float tr = (threadIdx.x > 0) ? 0.5f * (pdf + prev) * dxf : 0.f;
-    cdf = block_inclusive_scan(tr, scr);
+// Per-row P(best) core. The wave loops over models h twice:
+//   pass A accumulates slog[j] = sum_h log cdf_h(p_j) per lane point;
+//   pass B integrates pdf_h * exp(clamp(slog - log cdf_h)) and leaves the
+//   UNNORMALIZED per-model masses in s_pb[0..H).
+// Returns 1/sum_h mass (the normalizer). s_a/s_b/s_lnB/s_pb are this
+// row's staged LDS arrays.
+__device__ float pbest_row_core(const float* s_a, const float* s_b,
+                                const double* s_lnB, float* s_pb, int H) {
+    const int lane = threadIdx.x & 63;
+    const double step = (kGridHi - kGridLo) / (P_POINTS - 1);
+    const float dxf = (float)step;
+
+    double lx[PTS_PER_LANE], l1mx[PTS_PER_LANE];
+#pragma unroll
+    for (int j = 0; j < PTS_PER_LANE; ++j) {
+        double x = kGridLo + (double)(lane * PTS_PER_LANE + j) * step;
+        lx[j] = log(x);
+        l1mx[j] = log1p(-x);
+    }
+
+    float slog[PTS_PER_LANE] = {0.f, 0.f, 0.f, 0.f};
+    for (int h = 0; h < H; ++h) {
+        const double am1 = (double)s_a[h] - 1.0;
+        const double bm1 = (double)s_b[h] - 1.0;
+        const double lnB = s_lnB[h];
+        float pdf[PTS_PER_LANE];
+#pragma unroll
+        for (int j = 0; j < PTS_PER_LANE; ++j)
+            pdf[j] = __expf((float)(am1 * lx[j] + bm1 * l1mx[j] - lnB));
+        float prev3 = __shfl_up(pdf[PTS_PER_LANE - 1], 1, 64);
+        float tr0 = (lane == 0) ? 0.f : 0.5f * (pdf[0] + prev3) * dxf;
+        float tr1 = 0.5f * (pdf[1] + pdf[0]) * dxf;
+        float tr2 = 0.5f * (pdf[2] + pdf[1]) * dxf;
+        float tr3 = 0.5f * (pdf[3] + pdf[2]) * dxf;
+        float local = tr0 + tr1 + tr2 + tr3;
+        float base = wave_inclusive_scan(local) - local;
+        float c0 = base + tr0;
+        float c1 = c0 + tr1;
+        float c2 = c1 + tr2;
+        float c3 = c2 + tr3;
+        slog[0] += __logf(fmaxf(c0, kEps));
+        slog[1] += __logf(fmaxf(c1, kEps));
+        slog[2] += __logf(fmaxf(c2, kEps));
+        slog[3] += __logf(fmaxf(c3, kEps));
+    }
+
+    // trapz endpoint weights: global p == 0 (lane 0, j 0) and
+    // p == P-1 (lane 63, j 3) get 0.5
+    float w[PTS_PER_LANE] = {1.f, 1.f, 1.f, 1.f};
+    if (lane == 0) w[0] = 0.5f;
+    if (lane == 63) w[PTS_PER_LANE - 1] = 0.5f;
+
+    for (int h = 0; h < H; ++h) {
+        const double am1 = (double)s_a[h] - 1.0;
+        const double bm1 = (double)s_b[h] - 1.0;
+        const double lnB = s_lnB[h];
+        float pdf[PTS_PER_LANE];
+#pragma unroll
+        for (int j = 0; j < PTS_PER_LANE; ++j)
+            pdf[j] = __expf((float)(am1 * lx[j] + bm1 * l1mx[j] - lnB));
+        float prev3 = __shfl_up(pdf[PTS_PER_LANE - 1], 1, 64);
+        float tr0 = (lane == 0) ? 0.f : 0.5f * (pdf[0] + prev3) * dxf;
+        float tr1 = 0.5f * (pdf[1] + pdf[0]) * dxf;
+        float tr2 = 0.5f * (pdf[2] + pdf[1]) * dxf;
+        float tr3 = 0.5f * (pdf[3] + pdf[2]) * dxf;
+        float local = tr0 + tr1 + tr2 + tr3;
+        float base = wave_inclusive_scan(local) - local;
+        float cdf[PTS_PER_LANE];
+        cdf[0] = base + tr0;
+        cdf[1] = cdf[0] + tr1;
+        cdf[2] = cdf[1] + tr2;
+        cdf[3] = cdf[2] + tr3;
+        float acc = 0.f;
+#pragma unroll
+        for (int j = 0; j < PTS_PER_LANE; ++j) {
+            float lc = __logf(fmaxf(cdf[j], kEps));
+            float pe = __expf(
+                fminf(fmaxf(slog[j] - lc, -kLogClamp), kLogClamp));
+            acc += pdf[j] * pe * w[j];
+        }
+        float mass = wave_reduce_sum(acc * dxf);
+        if (lane == (h & 63)) s_pb[h] = mass;
+    }
+    // all lanes of the wave have executed the stores above in program
+    // order; wave-internal LDS visibility needs no barrier.
+    float part = 0.f;
+    for (int h = lane; h < H; h += 64) part += s_pb[h];
+    float total = wave_reduce_sum(part);
+    return 1.0f / fmaxf(total, kEps);
+}
+
+constexpr size_t lds_bytes(int H) {
+    // per row: lnB (f64 H) + a, b, pb (f32 H each)
+    return (size_t)ROWS_PER_BLOCK * ((size_t)H * 8 + 3 * (size_t)H * 4);
 }
 
 // ---------------------------------------------------------------------------
-// Kernel 1: generic P(best) over rows. alpha/beta: (R, H) -> out: (R, H).
-// One workgroup per row; two passes over H (pass A accumulates
-// slog_p = sum_h log cdf_h(p) in a register; pass B recomputes pdf/cdf and
-// integrates).  LDS: a, b (f32 H) + lnB (f64 H) + pb (f32 H) + pdfbuf(256)
-// + scratch.
+// Kernel 1: generic P(best). alpha/beta: (R, H) -> out: (R, H).
 // ---------------------------------------------------------------------------
 __global__ void __launch_bounds__(BLOCK)
 pbest_kernel(const float* __restrict__ alpha, const float* __restrict__ beta,
              float* __restrict__ out, int R, int H) {
     extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-    double* s_lnB = reinterpret_cast<double*>(smem_raw);         // H f64
-    float* s_a = reinterpret_cast<float*>(s_lnB + H);            // H
-    float* s_b = s_a + H;                                        // H
-    float* s_pb = s_b + H;                                       // H
-    float* s_pdf = s_pb + H;                                     // 256
-    float* s_scr = s_pdf + P_POINTS;                             // 8
+    double* lnB_all = reinterpret_cast<double*>(smem_raw);
+    float* f_all = reinterpret_cast<float*>(lnB_all + ROWS_PER_BLOCK * H);
 
-    const int r = blockIdx.x;
-    if (r >= R) return;
     const int tid = threadIdx.x;
+    const int row0 = blockIdx.x * ROWS_PER_BLOCK;
 
-    for (int h = tid; h < H; h += BLOCK) {
+    // cooperative staging of up to 4 rows (+ f64 lgamma, once per (r,h))
+    for (int idx = tid; idx < ROWS_PER_BLOCK * H; idx += BLOCK) {
+        const int rl = idx / H, h = idx - rl * H;
+        const int r = row0 + rl;
+        if (r >= R) continue;
         float a = alpha[(size_t)r * H + h];
         float b = beta[(size_t)r * H + h];
-        s_a[h] = a;
-        s_b[h] = b;
-        s_lnB[h] = lgamma((double)a) + lgamma((double)b)
-                 - lgamma((double)a + (double)b);
+        f_all[rl * 3 * H + h] = a;
+        f_all[rl * 3 * H + H + h] = b;
+        lnB_all[rl * H + h] = lgamma((double)a) + lgamma((double)b)
+                            - lgamma((double)a + (double)b);
     }
     __syncthreads();
 
-    const double step = (kGridHi - kGridLo) / (P_POINTS - 1);
-    const double x = kGridLo + (double)tid * step;
-    const double lx = log(x), l1mx = log1p(-x);
-    const float dxf = (float)step;
+    const int rl = tid >> 6;
+    const int r = row0 + rl;
+    if (r >= R) return;
+    float* s_a = f_all + rl * 3 * H;
+    float* s_b = s_a + H;
+    float* s_pb = s_b + H;
+    double* s_lnB = lnB_all + rl * H;
 
-    // pass A: slog_p = sum_h log cdf_h(p)
-    float slog = 0.f;
-    for (int h = 0; h < H; ++h) {
-        float pdf, cdf;
-        beta_pdf_cdf(s_a[h], s_b[h], s_lnB[h], lx, l1mx, dxf,
-                     s_pdf, s_scr, pdf, cdf);
-        slog += __logf(fmaxf(cdf, kEps));
-    }
+    float inv = pbest_row_core(s_a, s_b, s_lnB, s_pb, H);
 
-    // pass B: integrate pdf_h * exp(clamp(slog - log cdf_h))
-    const float w_trapz = (tid == 0 || tid == P_POINTS - 1) ? 0.5f : 1.0f;
-    for (int h = 0; h < H; ++h) {
-        float pdf, cdf;
-        beta_pdf_cdf(s_a[h], s_b[h], s_lnB[h], lx, l1mx, dxf,
-                     s_pdf, s_scr, pdf, cdf);
-        float lc = __logf(fmaxf(cdf, kEps));
-        float pe = __expf(fminf(fmaxf(slog - lc, -kLogClamp), kLogClamp));
-        float total = block_reduce_sum(pdf * pe * w_trapz * dxf, s_scr);
-        if (tid == 0) s_pb[h] = total;
-        __syncthreads();
-    }
-
-    // normalize over H and write out
-    float part = 0.f;
-    for (int h = tid; h < H; h += BLOCK) part += s_pb[h];
-    float total = block_reduce_sum(part, s_scr);
-    float inv = 1.0f / fmaxf(total, kEps);
-    for (int h = tid; h < H; h += BLOCK)
+    const int lane = tid & 63;
+    for (int h = lane; h < H; h += 64)
         out[(size_t)r * H + h] = s_pb[h] * inv;
 }
 
 // ---------------------------------------------------------------------------
 // Kernel 2: fused hypothetical-update P(best) + entropy epilogue for EIG.
-// One workgroup per (candidate b, hypothesized class c) row:
-//   a_h = alpha_t[c,h] + w*[cls[b,h]==c];  b_h = beta_t[c,h] + w*[!=]
-//   pb = pbest(a, b)          (normalized over H)
-//   m_h = mixture0[h] + pi_hat[c] * (pb_h - pbest_before[c,h])
-//   H_after[b,c] = -sum_h clamp(m,1e-12) log2 m
-// (reference: coda/coda.py:150-168 + :267-276). EIG itself is a trivial
-// (B,C) contraction done by the host.
+// One wave per (candidate b, hypothesized class c) row:
+//   a_h = alpha_t[c,h] + w*[cls[b,h]==c];  b_h = beta_t[c,h] + w*[else]
+//   pb = pbest(a, b)  (normalized over H)
+//   H_after[b,c] = -sum_h m log2 m,  m = clamp(mixture0[h]
+//                  + pi_hat[c]*(pb_h - pbest_before[c,h]), 1e-12)
+// (reference: coda/coda.py:150-168 + :267-276). The final EIG contraction
+// over (B, C) is done by the host.
 // ---------------------------------------------------------------------------
 __global__ void __launch_bounds__(BLOCK)
-eig_hyp_kernel(const float* __restrict__ alpha_t,   // (C, H)
-               const float* __restrict__ beta_t,    // (C, H)
-               const int* __restrict__ cls,         // (B, H)
+eig_hyp_kernel(const float* __restrict__ alpha_t,       // (C, H)
+               const float* __restrict__ beta_t,        // (C, H)
+               const int* __restrict__ cls,             // (B, H)
                const float* __restrict__ pbest_before,  // (C, H)
-               const float* __restrict__ pi_hat,    // (C,)
-               const float* __restrict__ mixture0,  // (H,)
-               float* __restrict__ h_after,         // (B, C)
+               const float* __restrict__ pi_hat,        // (C,)
+               const float* __restrict__ mixture0,      // (H,)
+               float* __restrict__ h_after,             // (B, C)
                float update_weight, int B, int C, int H) {
     extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-    double* s_lnB = reinterpret_cast<double*>(smem_raw);         // H f64
-    float* s_a = reinterpret_cast<float*>(s_lnB + H);            // H
-    float* s_b = s_a + H;                                        // H
-    float* s_pb = s_b + H;                                       // H
-    float* s_pdf = s_pb + H;                                     // 256
-    float* s_scr = s_pdf + P_POINTS;                             // 8
+    double* lnB_all = reinterpret_cast<double*>(smem_raw);
+    float* f_all = reinterpret_cast<float*>(lnB_all + ROWS_PER_BLOCK * H);
 
-    const int rid = blockIdx.x;
-    if (rid >= B * C) return;
-    const int b = rid / C;
-    const int c = rid - b * C;
     const int tid = threadIdx.x;
+    const int row0 = blockIdx.x * ROWS_PER_BLOCK;
+    const int R = B * C;
 
-    for (int h = tid; h < H; h += BLOCK) {
-        int cl = cls[(size_t)b * H + h];
-        float add = (cl == c) ? update_weight : 0.f;
+    for (int idx = tid; idx < ROWS_PER_BLOCK * H; idx += BLOCK) {
+        const int rl = idx / H, h = idx - rl * H;
+        const int r = row0 + rl;
+        if (r >= R) continue;
+        const int b = r / C, c = r - b * C;
+        const int cl = cls[(size_t)b * H + h];
+        const float add = (cl == c) ? update_weight : 0.f;
         float a = alpha_t[(size_t)c * H + h] + add;
         float bb = beta_t[(size_t)c * H + h] + (update_weight - add);
-        s_a[h] = a;
-        s_b[h] = bb;
-        s_lnB[h] = lgamma((double)a) + lgamma((double)bb)
-                 - lgamma((double)a + (double)bb);
+        f_all[rl * 3 * H + h] = a;
+        f_all[rl * 3 * H + H + h] = bb;
+        lnB_all[rl * H + h] = lgamma((double)a) + lgamma((double)bb)
+                            - lgamma((double)a + (double)bb);
     }
     __syncthreads();
 
-    const double step = (kGridHi - kGridLo) / (P_POINTS - 1);
-    const double x = kGridLo + (double)tid * step;
-    const double lx = log(x), l1mx = log1p(-x);
-    const float dxf = (float)step;
+    const int rl = tid >> 6;
+    const int r = row0 + rl;
+    if (r >= R) return;
+    const int b = r / C, c = r - b * C;
+    float* s_a = f_all + rl * 3 * H;
+    float* s_b = s_a + H;
+    float* s_pb = s_b + H;
+    double* s_lnB = lnB_all + rl * H;
 
-    float slog = 0.f;
-    for (int h = 0; h < H; ++h) {
-        float pdf, cdf;
-        beta_pdf_cdf(s_a[h], s_b[h], s_lnB[h], lx, l1mx, dxf,
-                     s_pdf, s_scr, pdf, cdf);
-        slog += __logf(fmaxf(cdf, kEps));
-    }
+    float inv = pbest_row_core(s_a, s_b, s_lnB, s_pb, H);
 
-    const float w_trapz = (tid == 0 || tid == P_POINTS - 1) ? 0.5f : 1.0f;
-    for (int h = 0; h < H; ++h) {
-        float pdf, cdf;
-        beta_pdf_cdf(s_a[h], s_b[h], s_lnB[h], lx, l1mx, dxf,
-                     s_pdf, s_scr, pdf, cdf);
-        float lc = __logf(fmaxf(cdf, kEps));
-        float pe = __expf(fminf(fmaxf(slog - lc, -kLogClamp), kLogClamp));
-        float total = block_reduce_sum(pdf * pe * w_trapz * dxf, s_scr);
-        if (tid == 0) s_pb[h] = total;
-        __syncthreads();
-    }
-
-    float part = 0.f;
-    for (int h = tid; h < H; h += BLOCK) part += s_pb[h];
-    float total = block_reduce_sum(part, s_scr);
-    float inv = 1.0f / fmaxf(total, kEps);
-
-    // entropy epilogue: -sum_h m log2 m, m = mixture0 + pi_c*(pb - before)
+    const int lane = tid & 63;
     const float pi_c = pi_hat[c];
     float ent = 0.f;
-    for (int h = tid; h < H; h += BLOCK) {
+    for (int h = lane; h < H; h += 64) {
         float pb = s_pb[h] * inv;
         float m = mixture0[h] + pi_c * (pb - pbest_before[(size_t)c * H + h]);
         m = fmaxf(m, 1e-12f);
         ent += -m * __log2f(m);
     }
-    float ent_total = block_reduce_sum(ent, s_scr);
-    if (tid == 0) h_after[rid] = ent_total;
-}
-
-size_t smem_bytes(int H) {
-    return (size_t)H * sizeof(double) + (size_t)(3 * H) * sizeof(float)
-         + (P_POINTS + 8) * sizeof(float);
+    float ent_total = wave_reduce_sum(ent);
+    if (lane == 0) h_after[r] = ent_total;
 }
 
 }  // namespace
@@ -270,10 +285,11 @@ torch::Tensor pbest_from_beta(torch::Tensor alpha, torch::Tensor beta,
     const int R = alpha.size(0), H = alpha.size(1);
     auto out = torch::empty_like(alpha);
     if (R == 0) return out;
-    size_t smem = smem_bytes(H);
+    const size_t smem = lds_bytes(H);
     TORCH_CHECK(smem <= 160 * 1024, "H too large for LDS: ", H);
+    const int blocks = (R + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
     auto stream = c10::hip::getCurrentHIPStream();
-    hipLaunchKernelGGL(pbest_kernel, dim3(R), dim3(BLOCK), smem,
+    hipLaunchKernelGGL(pbest_kernel, dim3(blocks), dim3(BLOCK), smem,
                        stream.stream(), alpha.data_ptr<float>(),
                        beta.data_ptr<float>(), out.data_ptr<float>(), R, H);
     C10_HIP_CHECK(hipGetLastError());
@@ -303,10 +319,12 @@ torch::Tensor eig_chunk(torch::Tensor alpha_cc, torch::Tensor beta_cc,
     check_f32_cuda(beta_t, "beta_cc");
 
     auto h_after = torch::empty({B, C}, alpha_t.options());
-    size_t smem = smem_bytes(H);
+    const size_t smem = lds_bytes(H);
     TORCH_CHECK(smem <= 160 * 1024, "H too large for LDS: ", H);
+    const int R = B * C;
+    const int blocks = (R + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
     auto stream = c10::hip::getCurrentHIPStream();
-    hipLaunchKernelGGL(eig_hyp_kernel, dim3(B * C), dim3(BLOCK), smem,
+    hipLaunchKernelGGL(eig_hyp_kernel, dim3(blocks), dim3(BLOCK), smem,
                        stream.stream(), alpha_t.data_ptr<float>(),
                        beta_t.data_ptr<float>(),
                        chunk_classes.data_ptr<int>(),

@@ -80,6 +80,9 @@ class CODA(ModelSelector):
         dis = (self.classes != ref_row.unsqueeze(0)).any(dim=0).to(torch.float32)
         self.comm.all_reduce_(dis)
         self._disagreement = dis > 0                      # (N,) bool
+        # host-side copy for the per-step candidate filter (indexing a GPU
+        # tensor point-by-point would be one device sync per point)
+        self._disagreement_host = self._disagreement.cpu().tolist()
 
         self.labeled_idxs, self.labels = [], []
         self.unlabeled_idxs = list(range(self.N))
@@ -103,7 +106,7 @@ class CODA(ModelSelector):
     def _prefilter(self, idxs):
         """Drop points where every model agrees (no information), optionally
         subsample to prefilter_n (reference: coda/coda.py:215-224)."""
-        mask = self._disagreement
+        mask = self._disagreement_host
         idxs = [i for i in idxs if mask[i]]
         if self.prefilter_n and len(idxs) > self.prefilter_n:
             idxs = random.sample(idxs, self.prefilter_n)
