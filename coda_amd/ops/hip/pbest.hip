@@ -39,7 +39,8 @@ namespace {
 constexpr double kGridLo = 1e-6;
 constexpr double kGridHi = 1.0 - 1e-6;
 constexpr float kEps = 1e-30f;
-constexpr float kLogClamp = 80.0f;
+constexpr float kLogClamp = 80.0f;               // natural-log clamp (reference)
+constexpr float kLog2Clamp = 115.41560327111707f; // same clamp in base 2
 
 __device__ __forceinline__ float wave_inclusive_scan(float v) {
     const int lane = threadIdx.x & 63;
@@ -58,11 +59,15 @@ __device__ __forceinline__ float wave_reduce_sum(float v) {
 }
 
 // Per-row P(best) core. The wave loops over models h twice:
-//   pass A accumulates slog[j] = sum_h log cdf_h(p_j) per lane point;
-//   pass B integrates pdf_h * exp(clamp(slog - log cdf_h)) and leaves the
-//   UNNORMALIZED per-model masses in s_pb[0..H).
-// Returns 1/sum_h mass (the normalizer). s_a/s_b/s_lnB/s_pb are this
-// row's staged LDS arrays.
+//   pass A accumulates the cdf product PI_h cdf_h(p_j) per lane point as a
+//   (mantissa, exponent) pair - one multiply + a native v_frexp per model
+//   instead of a log (base-2 log taken ONCE at the end);
+//   pass B integrates pdf_h * exp2(clamp(slog2 - log2 cdf_h)) and leaves
+//   the UNNORMALIZED per-model masses in s_pb[0..H).
+// All transcendentals are the native base-2 ops (v_exp_f32/v_log_f32);
+// log-pdf arguments are prepared in f64 as base-2 logs (s_lnB holds
+// log2 B(a,b)). Returns 1/sum_h mass. s_a/s_b/s_lnB/s_pb are this row's
+// staged LDS arrays.
 __device__ float pbest_row_core(const float* s_a, const float* s_b,
                                 const double* s_lnB, float* s_pb, int H) {
     const int lane = threadIdx.x & 63;
@@ -73,11 +78,13 @@ __device__ float pbest_row_core(const float* s_a, const float* s_b,
 #pragma unroll
     for (int j = 0; j < PTS_PER_LANE; ++j) {
         double x = kGridLo + (double)(lane * PTS_PER_LANE + j) * step;
-        lx[j] = log(x);
-        l1mx[j] = log1p(-x);
+        lx[j] = log2(x);
+        l1mx[j] = log2(1.0 - x);
     }
 
-    float slog[PTS_PER_LANE] = {0.f, 0.f, 0.f, 0.f};
+    // pass A: product of (clamped) cdfs as mantissa * 2^exp per point
+    float pm[PTS_PER_LANE] = {1.f, 1.f, 1.f, 1.f};
+    int pe_[PTS_PER_LANE] = {0, 0, 0, 0};
     for (int h = 0; h < H; ++h) {
         const double am1 = (double)s_a[h] - 1.0;
         const double bm1 = (double)s_b[h] - 1.0;
@@ -85,7 +92,7 @@ __device__ float pbest_row_core(const float* s_a, const float* s_b,
         float pdf[PTS_PER_LANE];
 #pragma unroll
         for (int j = 0; j < PTS_PER_LANE; ++j)
-            pdf[j] = __expf((float)(am1 * lx[j] + bm1 * l1mx[j] - lnB));
+            pdf[j] = exp2f((float)(am1 * lx[j] + bm1 * l1mx[j] - lnB));
         float prev3 = __shfl_up(pdf[PTS_PER_LANE - 1], 1, 64);
         float tr0 = (lane == 0) ? 0.f : 0.5f * (pdf[0] + prev3) * dxf;
         float tr1 = 0.5f * (pdf[1] + pdf[0]) * dxf;
@@ -97,11 +104,21 @@ __device__ float pbest_row_core(const float* s_a, const float* s_b,
         float c1 = c0 + tr1;
         float c2 = c1 + tr2;
         float c3 = c2 + tr3;
-        slog[0] += __logf(fmaxf(c0, kEps));
-        slog[1] += __logf(fmaxf(c1, kEps));
-        slog[2] += __logf(fmaxf(c2, kEps));
-        slog[3] += __logf(fmaxf(c3, kEps));
+        // renormalize every model: one clamped factor can be 1e-30, so a
+        // second multiply would underflow f32. v_frexp_mant/_exp are
+        // full-rate VALU ops - still cheaper than a per-model v_log.
+        float c[PTS_PER_LANE] = {c0, c1, c2, c3};
+#pragma unroll
+        for (int j = 0; j < PTS_PER_LANE; ++j) {
+            int ex;
+            pm[j] = frexpf(pm[j] * fmaxf(c[j], kEps), &ex);
+            pe_[j] += ex;
+        }
     }
+    float slog2[PTS_PER_LANE];  // log2 PI_h cdf_h per point
+#pragma unroll
+    for (int j = 0; j < PTS_PER_LANE; ++j)
+        slog2[j] = __log2f(pm[j]) + (float)pe_[j];
 
     // trapz endpoint weights: global p == 0 (lane 0, j 0) and
     // p == P-1 (lane 63, j 3) get 0.5
@@ -116,7 +133,7 @@ __device__ float pbest_row_core(const float* s_a, const float* s_b,
         float pdf[PTS_PER_LANE];
 #pragma unroll
         for (int j = 0; j < PTS_PER_LANE; ++j)
-            pdf[j] = __expf((float)(am1 * lx[j] + bm1 * l1mx[j] - lnB));
+            pdf[j] = exp2f((float)(am1 * lx[j] + bm1 * l1mx[j] - lnB));
         float prev3 = __shfl_up(pdf[PTS_PER_LANE - 1], 1, 64);
         float tr0 = (lane == 0) ? 0.f : 0.5f * (pdf[0] + prev3) * dxf;
         float tr1 = 0.5f * (pdf[1] + pdf[0]) * dxf;
@@ -132,9 +149,9 @@ __device__ float pbest_row_core(const float* s_a, const float* s_b,
         float acc = 0.f;
 #pragma unroll
         for (int j = 0; j < PTS_PER_LANE; ++j) {
-            float lc = __logf(fmaxf(cdf[j], kEps));
-            float pe = __expf(
-                fminf(fmaxf(slog[j] - lc, -kLogClamp), kLogClamp));
+            float lc = __log2f(fmaxf(cdf[j], kEps));
+            float pe = exp2f(
+                fminf(fmaxf(slog2[j] - lc, -kLog2Clamp), kLog2Clamp));
             acc += pdf[j] * pe * w[j];
         }
         float mass = wave_reduce_sum(acc * dxf);
@@ -175,8 +192,10 @@ pbest_kernel(const float* __restrict__ alpha, const float* __restrict__ beta,
         float b = beta[(size_t)r * H + h];
         f_all[rl * 3 * H + h] = a;
         f_all[rl * 3 * H + H + h] = b;
-        lnB_all[rl * H + h] = lgamma((double)a) + lgamma((double)b)
-                            - lgamma((double)a + (double)b);
+        // log2 of the Beta normalizer (the core works in base 2)
+        lnB_all[rl * H + h] = (lgamma((double)a) + lgamma((double)b)
+                            - lgamma((double)a + (double)b))
+                            * 1.4426950408889634;
     }
     __syncthreads();
 
@@ -233,8 +252,9 @@ eig_hyp_kernel(const float* __restrict__ alpha_t,       // (C, H)
         float bb = beta_t[(size_t)c * H + h] + (update_weight - add);
         f_all[rl * 3 * H + h] = a;
         f_all[rl * 3 * H + H + h] = bb;
-        lnB_all[rl * H + h] = lgamma((double)a) + lgamma((double)bb)
-                            - lgamma((double)a + (double)bb);
+        lnB_all[rl * H + h] = (lgamma((double)a) + lgamma((double)bb)
+                            - lgamma((double)a + (double)bb))
+                            * 1.4426950408889634;
     }
     __syncthreads();
 

@@ -95,18 +95,52 @@ def dirichlet_to_beta(dirichlets: torch.Tensor):
 def pi_hat_partial(dirichlets: torch.Tensor, preds: torch.Tensor,
                    chunk_h: int = 32) -> torch.Tensor:
     """Unnormalized per-item class scores summed over the LOCAL model axis:
-    sum_h preds[h] @ dirichlets[h]^T -> (N, C).
+    sum_h preds[h] @ dirichlets[h]^T -> (N, C) fp32.
 
     adjusted[h,n,c] = sum_s dirichlets[h,c,s] * preds[h,n,s]; the H sum is the
     RCCL all-reduce site when the model axis is sharded.
+
+    `preds` may be bf16 (the MI355X fast path: bf16 MFMA at ~16x the f32
+    rate; per-GEMM results are accumulated into fp32 across models, so only
+    the per-model contraction rounds at bf16). The fp32 path matches the
+    reference bit-for-bit semantics (coda/coda.py:227).
     """
     H = preds.shape[0]
+    dirichlets = dirichlets.to(preds.dtype)
     out = None
     for h0 in range(0, H, chunk_h):
         h1 = min(h0 + chunk_h, H)
-        part = torch.bmm(preds[h0:h1], dirichlets[h0:h1].transpose(1, 2)).sum(0)
+        part = torch.bmm(preds[h0:h1], dirichlets[h0:h1].transpose(1, 2))
+        part = part.float().sum(0)
         out = part if out is None else out + part
     return out
+
+
+def pi_hat_pack(preds: torch.Tensor) -> torch.Tensor:
+    """Pack (H, N, C) predictions as a (N, H*C) bf16 GEMM operand.
+
+    pi_hat's contraction adjusted[n,c] = sum_h sum_s preds[h,n,s]*D[h,c,s]
+    is a SINGLE GEMM of shape (N) x (C) x (K=H*C): A[n, h*C+s] =
+    preds[h,n,s], B[h*C+s, c] = D[h,c,s]. Packing A once at selector init
+    turns the per-step pi_hat update into one bf16 MFMA GEMM (f32
+    accumulate over the whole K, which folds the model-axis sum into the
+    GEMM - no (H,N,C) intermediate, ~16x the f32 matrix rate).
+
+    Note: torch.bmm over the model batch is both slower (f32) and broken
+    (hipBLASLt bf16 batched GEMM faults at batch=32, N=50k, C=1000 on
+    ROCm 7.0) - hence this packed single-GEMM formulation.
+    """
+    H, N, C = preds.shape
+    return preds.permute(1, 0, 2).reshape(N, H * C).contiguous().to(
+        torch.bfloat16)
+
+
+def pi_hat_partial_packed(dirichlets: torch.Tensor,
+                          packed: torch.Tensor) -> torch.Tensor:
+    """(H,C,C) Dirichlets + packed (N, H*C) bf16 preds -> (N, C) fp32."""
+    H, C, _ = dirichlets.shape
+    B = dirichlets.transpose(1, 2).reshape(H * C, C).to(packed.dtype)
+    return (packed @ B).float()
 
 
 def pi_hat_normalize(adjusted_sum: torch.Tensor):

@@ -79,6 +79,10 @@ def parse_args(argv=None):
     # MI355X additions
     parser.add_argument("--device", default=None, help="cpu / cuda")
     parser.add_argument("--chunk-size", type=int, default=100)
+    parser.add_argument("--pi-hat-precision", default="auto",
+                        choices=["auto", "fp32", "bf16"],
+                        help="pi_hat contraction dtype (bf16 = MFMA fast "
+                             "path on GPU, fp32 accumulate)")
     parser.add_argument("--sharded", action="store_true",
                         help="Shard the model axis across torchrun ranks.")
     return parser.parse_args(argv)

@@ -119,7 +119,7 @@ class TestEndToEnd:
             ds = Dataset.from_tensors(preds, labels, device)
             oracle = Oracle(ds, LOSS_FNS["acc"])
             random.seed(0); torch.manual_seed(0)
-            sel = CODA(ds, chunk_size=64)
+            sel = CODA(ds, chunk_size=64, pi_hat_precision="fp32")
             traj = []
             for _ in range(5):
                 idx, q = sel.get_next_item_to_label()

@@ -246,3 +246,17 @@ class TestShardedEquivalence:
         m0s, H0s = mixture_entropy_sharded(pb, pi, comm)
         got = S.eig_chunk_sharded(a0, b0, cls, pb, pi, pi_xi, m0s, H0s, comm)
         torch.testing.assert_close(got, want, rtol=1e-5, atol=1e-6)
+
+
+class TestPiHatPacked:
+    def test_packed_matches_fp32(self):
+        g = torch.Generator().manual_seed(21)
+        H, N, C = 6, 40, 5
+        preds = torch.softmax(torch.randn(H, N, C, generator=g), -1)
+        D = torch.rand(H, C, C, generator=g) + 0.1
+        want = ops.pi_hat_partial(D, preds)
+        packed = ops.pi_hat_pack(preds)
+        assert packed.shape == (N, H * C) and packed.dtype == torch.bfloat16
+        got = ops.pi_hat_partial_packed(D, packed)
+        # bf16 inputs: ~0.4% relative tolerance
+        torch.testing.assert_close(got, want, rtol=2e-2, atol=1e-3)
