@@ -31,6 +31,23 @@
 
 #define P_POINTS 256
 #define BLOCK 256
+
+// Tuning hooks (scripts/build_variants.py): CODA_MIN_WAVES forces the
+// register allocator toward higher occupancy; CODA_UNROLL_H software-
+// pipelines the model loop (independent pdf math overlaps the scan's
+// dependent shfl chain).
+#ifdef CODA_MIN_WAVES
+#define CODA_LB __launch_bounds__(BLOCK, CODA_MIN_WAVES)
+#else
+#define CODA_LB __launch_bounds__(BLOCK)
+#endif
+#define CODA_STR2(x) #x
+#define CODA_STR(x) CODA_STR2(x)
+#ifdef CODA_UNROLL_H
+#define CODA_HLOOP _Pragma(CODA_STR(unroll CODA_UNROLL_H))
+#else
+#define CODA_HLOOP
+#endif
 #define ROWS_PER_BLOCK 4
 #define PTS_PER_LANE 4
 
@@ -42,6 +59,34 @@ constexpr float kEps = 1e-30f;
 constexpr float kLogClamp = 80.0f;               // natural-log clamp (reference)
 constexpr float kLog2Clamp = 115.41560327111707f; // same clamp in base 2
 
+#ifdef CODA_DPP_SCAN
+// DPP cross-lane add: moved-in value (0 where no source / masked), added.
+// dpp_ctrl: row_shr:N = 0x110|N, row_bcast15 = 0x142, row_bcast31 = 0x143.
+template <int CTRL, int ROW_MASK>
+__device__ __forceinline__ float dpp_add(float x) {
+    int moved = __builtin_amdgcn_update_dpp(0, __float_as_int(x), CTRL,
+                                            ROW_MASK, 0xf, true);
+    return x + __int_as_float(moved);
+}
+
+// Wave64 inclusive add-scan in 6 DPP VALU ops (1-2 cycle dependent
+// latency each) instead of 6 dependent ds_bpermute rounds (~50 cycles
+// each): the classic GCN row_shr/row_bcast ladder.
+__device__ __forceinline__ float wave_inclusive_scan(float v) {
+    v = dpp_add<0x111, 0xf>(v);  // row_shr:1
+    v = dpp_add<0x112, 0xf>(v);  // row_shr:2
+    v = dpp_add<0x114, 0xf>(v);  // row_shr:4
+    v = dpp_add<0x118, 0xf>(v);  // row_shr:8  (rows of 16 complete)
+    v = dpp_add<0x142, 0xa>(v);  // row_bcast:15 -> rows 1, 3
+    v = dpp_add<0x143, 0xc>(v);  // row_bcast:31 -> rows 2, 3
+    return v;
+}
+
+__device__ __forceinline__ float wave_reduce_sum(float v) {
+    return __int_as_float(__builtin_amdgcn_readlane(
+        __float_as_int(wave_inclusive_scan(v)), 63));
+}
+#else
 __device__ __forceinline__ float wave_inclusive_scan(float v) {
     const int lane = threadIdx.x & 63;
 #pragma unroll
@@ -57,6 +102,7 @@ __device__ __forceinline__ float wave_reduce_sum(float v) {
     for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
     return v;
 }
+#endif
 
 // Per-row P(best) core. The wave loops over models h twice:
 //   pass A accumulates the cdf product PI_h cdf_h(p_j) per lane point as a
@@ -74,25 +120,40 @@ __device__ float pbest_row_core(const float* s_a, const float* s_b,
     const double step = (kGridHi - kGridLo) / (P_POINTS - 1);
     const float dxf = (float)step;
 
-    double lx[PTS_PER_LANE], l1mx[PTS_PER_LANE];
+    // Delta-centered log-pdf: the large, cancellation-prone part
+    //   K = am1*lx0 + bm1*l1mx0 - lnB
+    // is computed in f64 ONCE per (model, lane) at the lane's first grid
+    // point; the per-point remainder uses small f32 deltas
+    // (|delta| <= 3*step/x, so f32 FMA keeps ~1e-5 absolute accuracy in
+    // the base-2 exponent at any Beta count). This removes 6 of the 8
+    // f64 FMAs per model and ~8 VGPRs of f64 state vs the all-f64 form.
+    double lx0, l1mx0;
+    float dlx[PTS_PER_LANE], dl1mx[PTS_PER_LANE];
+    {
+        double x0 = kGridLo + (double)(lane * PTS_PER_LANE) * step;
+        lx0 = log2(x0);
+        l1mx0 = log2(1.0 - x0);
 #pragma unroll
-    for (int j = 0; j < PTS_PER_LANE; ++j) {
-        double x = kGridLo + (double)(lane * PTS_PER_LANE + j) * step;
-        lx[j] = log2(x);
-        l1mx[j] = log2(1.0 - x);
+        for (int j = 0; j < PTS_PER_LANE; ++j) {
+            double x = kGridLo + (double)(lane * PTS_PER_LANE + j) * step;
+            dlx[j] = (float)(log2(x) - lx0);
+            dl1mx[j] = (float)(log2(1.0 - x) - l1mx0);
+        }
     }
 
     // pass A: product of (clamped) cdfs as mantissa * 2^exp per point
     float pm[PTS_PER_LANE] = {1.f, 1.f, 1.f, 1.f};
     int pe_[PTS_PER_LANE] = {0, 0, 0, 0};
+    CODA_HLOOP
     for (int h = 0; h < H; ++h) {
-        const double am1 = (double)s_a[h] - 1.0;
-        const double bm1 = (double)s_b[h] - 1.0;
-        const double lnB = s_lnB[h];
+        const float am1 = s_a[h] - 1.0f;
+        const float bm1 = s_b[h] - 1.0f;
+        const float K = (float)((double)am1 * lx0 + (double)bm1 * l1mx0
+                                - s_lnB[h]);
         float pdf[PTS_PER_LANE];
 #pragma unroll
         for (int j = 0; j < PTS_PER_LANE; ++j)
-            pdf[j] = exp2f((float)(am1 * lx[j] + bm1 * l1mx[j] - lnB));
+            pdf[j] = exp2f(fmaf(am1, dlx[j], fmaf(bm1, dl1mx[j], K)));
         float prev3 = __shfl_up(pdf[PTS_PER_LANE - 1], 1, 64);
         float tr0 = (lane == 0) ? 0.f : 0.5f * (pdf[0] + prev3) * dxf;
         float tr1 = 0.5f * (pdf[1] + pdf[0]) * dxf;
@@ -126,14 +187,16 @@ __device__ float pbest_row_core(const float* s_a, const float* s_b,
     if (lane == 0) w[0] = 0.5f;
     if (lane == 63) w[PTS_PER_LANE - 1] = 0.5f;
 
+    CODA_HLOOP
     for (int h = 0; h < H; ++h) {
-        const double am1 = (double)s_a[h] - 1.0;
-        const double bm1 = (double)s_b[h] - 1.0;
-        const double lnB = s_lnB[h];
+        const float am1 = s_a[h] - 1.0f;
+        const float bm1 = s_b[h] - 1.0f;
+        const float K = (float)((double)am1 * lx0 + (double)bm1 * l1mx0
+                                - s_lnB[h]);
         float pdf[PTS_PER_LANE];
 #pragma unroll
         for (int j = 0; j < PTS_PER_LANE; ++j)
-            pdf[j] = exp2f((float)(am1 * lx[j] + bm1 * l1mx[j] - lnB));
+            pdf[j] = exp2f(fmaf(am1, dlx[j], fmaf(bm1, dl1mx[j], K)));
         float prev3 = __shfl_up(pdf[PTS_PER_LANE - 1], 1, 64);
         float tr0 = (lane == 0) ? 0.f : 0.5f * (pdf[0] + prev3) * dxf;
         float tr1 = 0.5f * (pdf[1] + pdf[0]) * dxf;
@@ -173,7 +236,7 @@ constexpr size_t lds_bytes(int H) {
 // ---------------------------------------------------------------------------
 // Kernel 1: generic P(best). alpha/beta: (R, H) -> out: (R, H).
 // ---------------------------------------------------------------------------
-__global__ void __launch_bounds__(BLOCK)
+__global__ void CODA_LB
 pbest_kernel(const float* __restrict__ alpha, const float* __restrict__ beta,
              float* __restrict__ out, int R, int H) {
     extern __shared__ __attribute__((aligned(16))) char smem_raw[];
@@ -224,7 +287,7 @@ pbest_kernel(const float* __restrict__ alpha, const float* __restrict__ beta,
 // (reference: coda/coda.py:150-168 + :267-276). The final EIG contraction
 // over (B, C) is done by the host.
 // ---------------------------------------------------------------------------
-__global__ void __launch_bounds__(BLOCK)
+__global__ void CODA_LB
 eig_hyp_kernel(const float* __restrict__ alpha_t,       // (C, H)
                const float* __restrict__ beta_t,        // (C, H)
                const int* __restrict__ cls,             // (B, H)
