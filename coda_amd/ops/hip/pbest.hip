@@ -28,6 +28,7 @@
 #include <torch/extension.h>
 #include <hip/hip_runtime.h>
 #include <c10/hip/HIPStream.h>
+#include <vector>
 
 #define P_POINTS 256
 #define BLOCK 256
@@ -114,22 +115,15 @@ __device__ __forceinline__ float wave_reduce_sum(float v) {
 // log-pdf arguments are prepared in f64 as base-2 logs (s_lnB holds
 // log2 B(a,b)). Returns 1/sum_h mass. s_a/s_b/s_lnB/s_pb are this row's
 // staged LDS arrays.
-__device__ float pbest_row_core(const float* s_a, const float* s_b,
-                                const double* s_lnB, float* s_pb, int H) {
-    const int lane = threadIdx.x & 63;
-    const double step = (kGridHi - kGridLo) / (P_POINTS - 1);
-    const float dxf = (float)step;
-
-    // Delta-centered log-pdf: the large, cancellation-prone part
-    //   K = am1*lx0 + bm1*l1mx0 - lnB
-    // is computed in f64 ONCE per (model, lane) at the lane's first grid
-    // point; the per-point remainder uses small f32 deltas
-    // (|delta| <= 3*step/x, so f32 FMA keeps ~1e-5 absolute accuracy in
-    // the base-2 exponent at any Beta count). This removes 6 of the 8
-    // f64 FMAs per model and ~8 VGPRs of f64 state vs the all-f64 form.
+// Shared per-lane grid state: base-2 logs of this lane's 4 grid points,
+// delta-centered (see pass A comment below).
+struct LaneGrid {
     double lx0, l1mx0;
     float dlx[PTS_PER_LANE], dl1mx[PTS_PER_LANE];
-    {
+    float dxf;
+    __device__ void init(int lane) {
+        const double step = (kGridHi - kGridLo) / (P_POINTS - 1);
+        dxf = (float)step;
         double x0 = kGridLo + (double)(lane * PTS_PER_LANE) * step;
         lx0 = log2(x0);
         l1mx0 = log2(1.0 - x0);
@@ -140,63 +134,25 @@ __device__ float pbest_row_core(const float* s_a, const float* s_b,
             dl1mx[j] = (float)(log2(1.0 - x) - l1mx0);
         }
     }
-
-    // pass A: product of (clamped) cdfs as mantissa * 2^exp per point
-    float pm[PTS_PER_LANE] = {1.f, 1.f, 1.f, 1.f};
-    int pe_[PTS_PER_LANE] = {0, 0, 0, 0};
-    CODA_HLOOP
-    for (int h = 0; h < H; ++h) {
-        const float am1 = s_a[h] - 1.0f;
-        const float bm1 = s_b[h] - 1.0f;
+    // Delta-centered log-pdf: the large, cancellation-prone anchor
+    //   K = am1*lx0 + bm1*l1mx0 - log2B
+    // is computed in f64 once per (model, lane); per-point remainders are
+    // small f32 deltas (~1e-5 absolute accuracy in the base-2 exponent at
+    // any Beta count).
+    __device__ __forceinline__ void pdf4(float a, float b, double lnB,
+                                         float (&pdf)[PTS_PER_LANE]) const {
+        const float am1 = a - 1.0f;
+        const float bm1 = b - 1.0f;
         const float K = (float)((double)am1 * lx0 + (double)bm1 * l1mx0
-                                - s_lnB[h]);
-        float pdf[PTS_PER_LANE];
+                                - lnB);
 #pragma unroll
         for (int j = 0; j < PTS_PER_LANE; ++j)
             pdf[j] = exp2f(fmaf(am1, dlx[j], fmaf(bm1, dl1mx[j], K)));
-        float prev3 = __shfl_up(pdf[PTS_PER_LANE - 1], 1, 64);
-        float tr0 = (lane == 0) ? 0.f : 0.5f * (pdf[0] + prev3) * dxf;
-        float tr1 = 0.5f * (pdf[1] + pdf[0]) * dxf;
-        float tr2 = 0.5f * (pdf[2] + pdf[1]) * dxf;
-        float tr3 = 0.5f * (pdf[3] + pdf[2]) * dxf;
-        float local = tr0 + tr1 + tr2 + tr3;
-        float base = wave_inclusive_scan(local) - local;
-        float c0 = base + tr0;
-        float c1 = c0 + tr1;
-        float c2 = c1 + tr2;
-        float c3 = c2 + tr3;
-        // renormalize every model: one clamped factor can be 1e-30, so a
-        // second multiply would underflow f32. v_frexp_mant/_exp are
-        // full-rate VALU ops - still cheaper than a per-model v_log.
-        float c[PTS_PER_LANE] = {c0, c1, c2, c3};
-#pragma unroll
-        for (int j = 0; j < PTS_PER_LANE; ++j) {
-            int ex;
-            pm[j] = frexpf(pm[j] * fmaxf(c[j], kEps), &ex);
-            pe_[j] += ex;
-        }
     }
-    float slog2[PTS_PER_LANE];  // log2 PI_h cdf_h per point
-#pragma unroll
-    for (int j = 0; j < PTS_PER_LANE; ++j)
-        slog2[j] = __log2f(pm[j]) + (float)pe_[j];
-
-    // trapz endpoint weights: global p == 0 (lane 0, j 0) and
-    // p == P-1 (lane 63, j 3) get 0.5
-    float w[PTS_PER_LANE] = {1.f, 1.f, 1.f, 1.f};
-    if (lane == 0) w[0] = 0.5f;
-    if (lane == 63) w[PTS_PER_LANE - 1] = 0.5f;
-
-    CODA_HLOOP
-    for (int h = 0; h < H; ++h) {
-        const float am1 = s_a[h] - 1.0f;
-        const float bm1 = s_b[h] - 1.0f;
-        const float K = (float)((double)am1 * lx0 + (double)bm1 * l1mx0
-                                - s_lnB[h]);
-        float pdf[PTS_PER_LANE];
-#pragma unroll
-        for (int j = 0; j < PTS_PER_LANE; ++j)
-            pdf[j] = exp2f(fmaf(am1, dlx[j], fmaf(bm1, dl1mx[j], K)));
+    // Trapezoid-cumulative cdf at this lane's 4 points from the wave's pdf.
+    __device__ __forceinline__ void cdf4(const float (&pdf)[PTS_PER_LANE],
+                                         int lane,
+                                         float (&cdf)[PTS_PER_LANE]) const {
         float prev3 = __shfl_up(pdf[PTS_PER_LANE - 1], 1, 64);
         float tr0 = (lane == 0) ? 0.f : 0.5f * (pdf[0] + prev3) * dxf;
         float tr1 = 0.5f * (pdf[1] + pdf[0]) * dxf;
@@ -204,11 +160,57 @@ __device__ float pbest_row_core(const float* s_a, const float* s_b,
         float tr3 = 0.5f * (pdf[3] + pdf[2]) * dxf;
         float local = tr0 + tr1 + tr2 + tr3;
         float base = wave_inclusive_scan(local) - local;
-        float cdf[PTS_PER_LANE];
         cdf[0] = base + tr0;
         cdf[1] = cdf[0] + tr1;
         cdf[2] = cdf[1] + tr2;
         cdf[3] = cdf[2] + tr3;
+    }
+};
+
+// Pass A: slog2[j] = log2 PROD_h clamp(cdf_h(p_j), 1e-30), accumulated as
+// a (mantissa, exponent) pair - one multiply + native v_frexp per model
+// instead of a per-model v_log (base-2 log taken once at the end).
+__device__ void pbest_pass_a(const float* s_a, const float* s_b,
+                             const double* s_lnB, int H,
+                             const LaneGrid& g, int lane,
+                             float (&slog2)[PTS_PER_LANE]) {
+    float pm[PTS_PER_LANE] = {1.f, 1.f, 1.f, 1.f};
+    int pe_[PTS_PER_LANE] = {0, 0, 0, 0};
+    CODA_HLOOP
+    for (int h = 0; h < H; ++h) {
+        float pdf[PTS_PER_LANE], cdf[PTS_PER_LANE];
+        g.pdf4(s_a[h], s_b[h], s_lnB[h], pdf);
+        g.cdf4(pdf, lane, cdf);
+        // renormalize every model: one clamped factor can be 1e-30, so a
+        // second multiply would underflow f32.
+#pragma unroll
+        for (int j = 0; j < PTS_PER_LANE; ++j) {
+            int ex;
+            pm[j] = frexpf(pm[j] * fmaxf(cdf[j], kEps), &ex);
+            pe_[j] += ex;
+        }
+    }
+#pragma unroll
+    for (int j = 0; j < PTS_PER_LANE; ++j)
+        slog2[j] = __log2f(pm[j]) + (float)pe_[j];
+}
+
+// Pass B: unnormalized per-model masses
+//   pb_h = trapz pdf_h * exp2(clamp(slog2 - log2 cdf_h, +-115.4))
+// left in s_pb[0..H); returns the wave-local sum over H (the normalizer
+// partial - globally reduced by the caller when H is sharded).
+__device__ float pbest_pass_b(const float* s_a, const float* s_b,
+                              const double* s_lnB, float* s_pb, int H,
+                              const LaneGrid& g, int lane,
+                              const float (&slog2)[PTS_PER_LANE]) {
+    float w[PTS_PER_LANE] = {1.f, 1.f, 1.f, 1.f};
+    if (lane == 0) w[0] = 0.5f;                 // trapz endpoint p == 0
+    if (lane == 63) w[PTS_PER_LANE - 1] = 0.5f;  // and p == P-1
+    CODA_HLOOP
+    for (int h = 0; h < H; ++h) {
+        float pdf[PTS_PER_LANE], cdf[PTS_PER_LANE];
+        g.pdf4(s_a[h], s_b[h], s_lnB[h], pdf);
+        g.cdf4(pdf, lane, cdf);
         float acc = 0.f;
 #pragma unroll
         for (int j = 0; j < PTS_PER_LANE; ++j) {
@@ -217,15 +219,70 @@ __device__ float pbest_row_core(const float* s_a, const float* s_b,
                 fminf(fmaxf(slog2[j] - lc, -kLog2Clamp), kLog2Clamp));
             acc += pdf[j] * pe * w[j];
         }
-        float mass = wave_reduce_sum(acc * dxf);
+        float mass = wave_reduce_sum(acc * g.dxf);
         if (lane == (h & 63)) s_pb[h] = mass;
     }
-    // all lanes of the wave have executed the stores above in program
-    // order; wave-internal LDS visibility needs no barrier.
+    // all lanes of the wave executed the stores above in program order;
+    // wave-internal LDS visibility needs no barrier.
     float part = 0.f;
     for (int h = lane; h < H; h += 64) part += s_pb[h];
-    float total = wave_reduce_sum(part);
+    return wave_reduce_sum(part);
+}
+
+// Fused single-device core: both passes; returns 1/total.
+__device__ float pbest_row_core(const float* s_a, const float* s_b,
+                                const double* s_lnB, float* s_pb, int H) {
+    const int lane = threadIdx.x & 63;
+    LaneGrid g;
+    g.init(lane);
+    float slog2[PTS_PER_LANE];
+    pbest_pass_a(s_a, s_b, s_lnB, H, g, lane, slog2);
+    float total = pbest_pass_b(s_a, s_b, s_lnB, s_pb, H, g, lane, slog2);
     return 1.0f / fmaxf(total, kEps);
+}
+
+// Cooperative LDS staging of up to ROWS_PER_BLOCK rows of Beta params
+// (+ the f64 log2-Beta-normalizer, computed once per (row, model)).
+__device__ void stage_plain(const float* alpha, const float* beta, int R,
+                            int H, int row0, double* lnB_all, float* f_all) {
+    for (int idx = threadIdx.x; idx < ROWS_PER_BLOCK * H; idx += BLOCK) {
+        const int rl = idx / H, h = idx - rl * H;
+        const int r = row0 + rl;
+        if (r >= R) continue;
+        float a = alpha[(size_t)r * H + h];
+        float b = beta[(size_t)r * H + h];
+        f_all[rl * 3 * H + h] = a;
+        f_all[rl * 3 * H + H + h] = b;
+        lnB_all[rl * H + h] = (lgamma((double)a) + lgamma((double)b)
+                            - lgamma((double)a + (double)b))
+                            * 1.4426950408889634;
+    }
+    __syncthreads();
+}
+
+// Staging with the hypothetical update applied on the fly:
+// row r = (candidate b, hypothesized class c); model h gets alpha+w if its
+// argmax class on b equals c, else beta+w (reference coda/coda.py:150-168).
+__device__ void stage_hyp(const float* alpha_t, const float* beta_t,
+                          const int* cls, float update_weight, int B, int C,
+                          int H, int row0, double* lnB_all, float* f_all) {
+    const int R = B * C;
+    for (int idx = threadIdx.x; idx < ROWS_PER_BLOCK * H; idx += BLOCK) {
+        const int rl = idx / H, h = idx - rl * H;
+        const int r = row0 + rl;
+        if (r >= R) continue;
+        const int b = r / C, c = r - b * C;
+        const int cl = cls[(size_t)b * H + h];
+        const float add = (cl == c) ? update_weight : 0.f;
+        float a = alpha_t[(size_t)c * H + h] + add;
+        float bb = beta_t[(size_t)c * H + h] + (update_weight - add);
+        f_all[rl * 3 * H + h] = a;
+        f_all[rl * 3 * H + H + h] = bb;
+        lnB_all[rl * H + h] = (lgamma((double)a) + lgamma((double)bb)
+                            - lgamma((double)a + (double)bb))
+                            * 1.4426950408889634;
+    }
+    __syncthreads();
 }
 
 constexpr size_t lds_bytes(int H) {
@@ -245,22 +302,7 @@ pbest_kernel(const float* __restrict__ alpha, const float* __restrict__ beta,
 
     const int tid = threadIdx.x;
     const int row0 = blockIdx.x * ROWS_PER_BLOCK;
-
-    // cooperative staging of up to 4 rows (+ f64 lgamma, once per (r,h))
-    for (int idx = tid; idx < ROWS_PER_BLOCK * H; idx += BLOCK) {
-        const int rl = idx / H, h = idx - rl * H;
-        const int r = row0 + rl;
-        if (r >= R) continue;
-        float a = alpha[(size_t)r * H + h];
-        float b = beta[(size_t)r * H + h];
-        f_all[rl * 3 * H + h] = a;
-        f_all[rl * 3 * H + H + h] = b;
-        // log2 of the Beta normalizer (the core works in base 2)
-        lnB_all[rl * H + h] = (lgamma((double)a) + lgamma((double)b)
-                            - lgamma((double)a + (double)b))
-                            * 1.4426950408889634;
-    }
-    __syncthreads();
+    stage_plain(alpha, beta, R, H, row0, lnB_all, f_all);
 
     const int rl = tid >> 6;
     const int r = row0 + rl;
@@ -303,23 +345,8 @@ eig_hyp_kernel(const float* __restrict__ alpha_t,       // (C, H)
     const int tid = threadIdx.x;
     const int row0 = blockIdx.x * ROWS_PER_BLOCK;
     const int R = B * C;
-
-    for (int idx = tid; idx < ROWS_PER_BLOCK * H; idx += BLOCK) {
-        const int rl = idx / H, h = idx - rl * H;
-        const int r = row0 + rl;
-        if (r >= R) continue;
-        const int b = r / C, c = r - b * C;
-        const int cl = cls[(size_t)b * H + h];
-        const float add = (cl == c) ? update_weight : 0.f;
-        float a = alpha_t[(size_t)c * H + h] + add;
-        float bb = beta_t[(size_t)c * H + h] + (update_weight - add);
-        f_all[rl * 3 * H + h] = a;
-        f_all[rl * 3 * H + H + h] = bb;
-        lnB_all[rl * H + h] = (lgamma((double)a) + lgamma((double)bb)
-                            - lgamma((double)a + (double)bb))
-                            * 1.4426950408889634;
-    }
-    __syncthreads();
+    stage_hyp(alpha_t, beta_t, cls, update_weight, B, C, H, row0,
+              lnB_all, f_all);
 
     const int rl = tid >> 6;
     const int r = row0 + rl;
@@ -343,6 +370,128 @@ eig_hyp_kernel(const float* __restrict__ alpha_t,       // (C, H)
     }
     float ent_total = wave_reduce_sum(ent);
     if (lane == 0) h_after[r] = ent_total;
+}
+
+
+// ---------------------------------------------------------------------------
+// Two-phase (model-axis-sharded) kernels. When H shards across ranks, the
+// coupling term sum_h log2 cdf_h(p) is all-reduced between pass A and
+// pass B (SURVEY.md section 2.4 - the latency-critical collective of the
+// EIG loop). Phase 1 writes each row's slog2 partial (R, P) coalesced
+// (lane l owns points 4l..4l+3); after the all-reduce, phase 2 reads the
+// GLOBAL slog2, integrates, and emits UNNORMALIZED per-model masses plus
+// the local normalizer partial (second, tiny all-reduce done by the host).
+// ---------------------------------------------------------------------------
+__global__ void CODA_LB
+pbest_phase1_kernel(const float* __restrict__ alpha,
+                    const float* __restrict__ beta,
+                    float* __restrict__ slog2_out,  // (R, P)
+                    int R, int H) {
+    extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+    double* lnB_all = reinterpret_cast<double*>(smem_raw);
+    float* f_all = reinterpret_cast<float*>(lnB_all + ROWS_PER_BLOCK * H);
+    const int row0 = blockIdx.x * ROWS_PER_BLOCK;
+    stage_plain(alpha, beta, R, H, row0, lnB_all, f_all);
+    const int rl = threadIdx.x >> 6;
+    const int r = row0 + rl;
+    if (r >= R) return;
+    const int lane = threadIdx.x & 63;
+    LaneGrid g;
+    g.init(lane);
+    float slog2[PTS_PER_LANE];
+    pbest_pass_a(f_all + rl * 3 * H, f_all + rl * 3 * H + H,
+                 lnB_all + rl * H, H, g, lane, slog2);
+#pragma unroll
+    for (int j = 0; j < PTS_PER_LANE; ++j)
+        slog2_out[(size_t)r * P_POINTS + lane * PTS_PER_LANE + j] = slog2[j];
+}
+
+__global__ void CODA_LB
+pbest_phase2_kernel(const float* __restrict__ alpha,
+                    const float* __restrict__ beta,
+                    const float* __restrict__ slog2_in,  // (R, P) global
+                    float* __restrict__ pb_out,          // (R, H) unnorm
+                    float* __restrict__ tot_out,         // (R,) partial
+                    int R, int H) {
+    extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+    double* lnB_all = reinterpret_cast<double*>(smem_raw);
+    float* f_all = reinterpret_cast<float*>(lnB_all + ROWS_PER_BLOCK * H);
+    const int row0 = blockIdx.x * ROWS_PER_BLOCK;
+    stage_plain(alpha, beta, R, H, row0, lnB_all, f_all);
+    const int rl = threadIdx.x >> 6;
+    const int r = row0 + rl;
+    if (r >= R) return;
+    const int lane = threadIdx.x & 63;
+    LaneGrid g;
+    g.init(lane);
+    float slog2[PTS_PER_LANE];
+#pragma unroll
+    for (int j = 0; j < PTS_PER_LANE; ++j)
+        slog2[j] = slog2_in[(size_t)r * P_POINTS + lane * PTS_PER_LANE + j];
+    float* s_pb = f_all + rl * 3 * H + 2 * H;
+    float total = pbest_pass_b(f_all + rl * 3 * H, f_all + rl * 3 * H + H,
+                               lnB_all + rl * H, s_pb, H, g, lane, slog2);
+    for (int h = lane; h < H; h += 64)
+        pb_out[(size_t)r * H + h] = s_pb[h];
+    if (lane == 0) tot_out[r] = total;
+}
+
+__global__ void CODA_LB
+eig_phase1_kernel(const float* __restrict__ alpha_t,
+                  const float* __restrict__ beta_t,
+                  const int* __restrict__ cls,
+                  float* __restrict__ slog2_out,  // (B*C, P)
+                  float update_weight, int B, int C, int H) {
+    extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+    double* lnB_all = reinterpret_cast<double*>(smem_raw);
+    float* f_all = reinterpret_cast<float*>(lnB_all + ROWS_PER_BLOCK * H);
+    const int row0 = blockIdx.x * ROWS_PER_BLOCK;
+    stage_hyp(alpha_t, beta_t, cls, update_weight, B, C, H, row0,
+              lnB_all, f_all);
+    const int rl = threadIdx.x >> 6;
+    const int r = row0 + rl;
+    if (r >= B * C) return;
+    const int lane = threadIdx.x & 63;
+    LaneGrid g;
+    g.init(lane);
+    float slog2[PTS_PER_LANE];
+    pbest_pass_a(f_all + rl * 3 * H, f_all + rl * 3 * H + H,
+                 lnB_all + rl * H, H, g, lane, slog2);
+#pragma unroll
+    for (int j = 0; j < PTS_PER_LANE; ++j)
+        slog2_out[(size_t)r * P_POINTS + lane * PTS_PER_LANE + j] = slog2[j];
+}
+
+__global__ void CODA_LB
+eig_phase2_kernel(const float* __restrict__ alpha_t,
+                  const float* __restrict__ beta_t,
+                  const int* __restrict__ cls,
+                  const float* __restrict__ slog2_in,  // (B*C, P) global
+                  float* __restrict__ pb_out,          // (B*C, H) unnorm
+                  float* __restrict__ tot_out,         // (B*C,) partial
+                  float update_weight, int B, int C, int H) {
+    extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+    double* lnB_all = reinterpret_cast<double*>(smem_raw);
+    float* f_all = reinterpret_cast<float*>(lnB_all + ROWS_PER_BLOCK * H);
+    const int row0 = blockIdx.x * ROWS_PER_BLOCK;
+    stage_hyp(alpha_t, beta_t, cls, update_weight, B, C, H, row0,
+              lnB_all, f_all);
+    const int rl = threadIdx.x >> 6;
+    const int r = row0 + rl;
+    if (r >= B * C) return;
+    const int lane = threadIdx.x & 63;
+    LaneGrid g;
+    g.init(lane);
+    float slog2[PTS_PER_LANE];
+#pragma unroll
+    for (int j = 0; j < PTS_PER_LANE; ++j)
+        slog2[j] = slog2_in[(size_t)r * P_POINTS + lane * PTS_PER_LANE + j];
+    float* s_pb = f_all + rl * 3 * H + 2 * H;
+    float total = pbest_pass_b(f_all + rl * 3 * H, f_all + rl * 3 * H + H,
+                               lnB_all + rl * H, s_pb, H, g, lane, slog2);
+    for (int h = lane; h < H; h += 64)
+        pb_out[(size_t)r * H + h] = s_pb[h];
+    if (lane == 0) tot_out[r] = total;
 }
 
 }  // namespace
@@ -420,10 +569,113 @@ torch::Tensor eig_chunk(torch::Tensor alpha_cc, torch::Tensor beta_cc,
     return h_before - (pi_hat_xi * h_after).sum(-1);
 }
 
+
+// ---- Two-phase (sharded) host bindings ----
+
+torch::Tensor pbest_phase1(torch::Tensor alpha, torch::Tensor beta) {
+    check_f32_cuda(alpha, "alpha");
+    check_f32_cuda(beta, "beta");
+    const int R = alpha.size(0), H = alpha.size(1);
+    auto slog2 = torch::empty({R, P_POINTS}, alpha.options());
+    if (R == 0) return slog2;
+    const size_t smem = lds_bytes(H);
+    TORCH_CHECK(smem <= 160 * 1024, "H too large for LDS: ", H);
+    const int blocks = (R + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
+    auto stream = c10::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(pbest_phase1_kernel, dim3(blocks), dim3(BLOCK), smem,
+                       stream.stream(), alpha.data_ptr<float>(),
+                       beta.data_ptr<float>(), slog2.data_ptr<float>(),
+                       R, H);
+    C10_HIP_CHECK(hipGetLastError());
+    return slog2;
+}
+
+std::vector<torch::Tensor> pbest_phase2(torch::Tensor alpha,
+                                        torch::Tensor beta,
+                                        torch::Tensor slog2) {
+    check_f32_cuda(alpha, "alpha");
+    check_f32_cuda(beta, "beta");
+    check_f32_cuda(slog2, "slog2");
+    const int R = alpha.size(0), H = alpha.size(1);
+    auto pb = torch::empty({R, H}, alpha.options());
+    auto tot = torch::empty({R}, alpha.options());
+    if (R == 0) return {pb, tot};
+    const size_t smem = lds_bytes(H);
+    const int blocks = (R + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
+    auto stream = c10::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(pbest_phase2_kernel, dim3(blocks), dim3(BLOCK), smem,
+                       stream.stream(), alpha.data_ptr<float>(),
+                       beta.data_ptr<float>(), slog2.data_ptr<float>(),
+                       pb.data_ptr<float>(), tot.data_ptr<float>(), R, H);
+    C10_HIP_CHECK(hipGetLastError());
+    return {pb, tot};
+}
+
+torch::Tensor eig_phase1(torch::Tensor alpha_cc, torch::Tensor beta_cc,
+                         torch::Tensor chunk_classes, double update_weight) {
+    const int H = alpha_cc.size(0), C = alpha_cc.size(1);
+    const int B = chunk_classes.size(0);
+    auto alpha_t = alpha_cc.t().contiguous();
+    auto beta_t = beta_cc.t().contiguous();
+    check_f32_cuda(alpha_t, "alpha_cc");
+    TORCH_CHECK(chunk_classes.scalar_type() == torch::kInt32,
+                "chunk_classes must be int32");
+    auto slog2 = torch::empty({(int64_t)B * C, P_POINTS}, alpha_t.options());
+    const size_t smem = lds_bytes(H);
+    TORCH_CHECK(smem <= 160 * 1024, "H too large for LDS: ", H);
+    const int R = B * C;
+    const int blocks = (R + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
+    auto stream = c10::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(eig_phase1_kernel, dim3(blocks), dim3(BLOCK), smem,
+                       stream.stream(), alpha_t.data_ptr<float>(),
+                       beta_t.data_ptr<float>(),
+                       chunk_classes.data_ptr<int>(),
+                       slog2.data_ptr<float>(), (float)update_weight,
+                       B, C, H);
+    C10_HIP_CHECK(hipGetLastError());
+    return slog2;
+}
+
+std::vector<torch::Tensor> eig_phase2(torch::Tensor alpha_cc,
+                                      torch::Tensor beta_cc,
+                                      torch::Tensor chunk_classes,
+                                      torch::Tensor slog2,
+                                      double update_weight) {
+    const int H = alpha_cc.size(0), C = alpha_cc.size(1);
+    const int B = chunk_classes.size(0);
+    auto alpha_t = alpha_cc.t().contiguous();
+    auto beta_t = beta_cc.t().contiguous();
+    check_f32_cuda(alpha_t, "alpha_cc");
+    check_f32_cuda(slog2, "slog2");
+    auto pb = torch::empty({(int64_t)B * C, H}, alpha_t.options());
+    auto tot = torch::empty({(int64_t)B * C}, alpha_t.options());
+    const size_t smem = lds_bytes(H);
+    const int R = B * C;
+    const int blocks = (R + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
+    auto stream = c10::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(eig_phase2_kernel, dim3(blocks), dim3(BLOCK), smem,
+                       stream.stream(), alpha_t.data_ptr<float>(),
+                       beta_t.data_ptr<float>(),
+                       chunk_classes.data_ptr<int>(),
+                       slog2.data_ptr<float>(), pb.data_ptr<float>(),
+                       tot.data_ptr<float>(), (float)update_weight,
+                       B, C, H);
+    C10_HIP_CHECK(hipGetLastError());
+    return {pb, tot};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "coda_amd fused gfx950 kernels";
     m.def("pbest_from_beta", &pbest_from_beta,
           "Beta-grid P(best) per row: (R,H),(R,H) -> (R,H)");
     m.def("eig_chunk", &eig_chunk,
           "Fused hypothetical P(best) + entropy EIG for a candidate chunk");
+    m.def("pbest_phase1", &pbest_phase1,
+          "Sharded pass A: local sum_h log2 cdf partials (R, P)");
+    m.def("pbest_phase2", &pbest_phase2,
+          "Sharded pass B: unnormalized masses + normalizer partial");
+    m.def("eig_phase1", &eig_phase1,
+          "Sharded hypothetical pass A: slog2 partials (B*C, P)");
+    m.def("eig_phase2", &eig_phase2,
+          "Sharded hypothetical pass B: unnorm masses + totals");
 }

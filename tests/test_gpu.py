@@ -146,3 +146,39 @@ class TestEndToEnd:
         a, b = _rand_betas(2, 4, device=dev)
         with pytest.raises(RuntimeError, match="HIP extension"):
             ops.pbest_from_beta(a, b)
+
+
+class TestShardedPhases:
+    """Single-GPU equivalence: the two-phase sharded kernels composed with
+    a no-op Comm must match the fused kernels (this chains with the CPU
+    gloo test: eager-sharded == eager, phases == fused, so
+    HIP-sharded-on-N-ranks inherits correctness)."""
+
+    def test_pbest_phases_equal_fused(self, dev):
+        from coda_amd import ops
+        from coda_amd.ops import sharded as S
+        from coda_amd.parallel import Comm
+        a, b = _rand_betas(32, 48, seed=21, device=dev)
+        got = S.pbest_from_beta_sharded(a, b, Comm())
+        want = ops.pbest_from_beta(a, b)
+        torch.testing.assert_close(got, want, rtol=1e-4, atol=1e-6)
+
+    def test_eig_phases_equal_fused(self, dev):
+        from coda_amd import ops
+        from coda_amd.ops import sharded as S
+        from coda_amd.parallel import Comm
+        g = torch.Generator().manual_seed(22)
+        H, C, B = 16, 10, 24
+        a0 = (torch.rand(H, C, generator=g) * 20 + 1).to(dev)
+        b0 = (torch.rand(H, C, generator=g) * 20 + 1).to(dev)
+        cls = torch.randint(0, C, (B, H), generator=g).to(dev)
+        pi = torch.softmax(torch.rand(C, generator=g), 0).to(dev)
+        pi_xi = torch.softmax(torch.rand(B, C, generator=g), -1).to(dev)
+        pb = ops.pbest_from_beta(a0.t().contiguous(), b0.t().contiguous())
+        m0, H0 = ops.mixture_entropy(pb, pi)
+        want = ops.eig_chunk(a0, b0, cls, pb, pi, pi_xi, m0, H0)
+        comm = Comm()
+        m0s, H0s = S.mixture_entropy_sharded(pb, pi, comm)
+        got = S.eig_chunk_sharded(a0, b0, cls, pb, pi, pi_xi, m0s, H0s,
+                                  comm)
+        torch.testing.assert_close(got, want, rtol=1e-4, atol=1e-5)
