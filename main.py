@@ -19,6 +19,7 @@ from __future__ import annotations
 import argparse
 import os
 import random
+import time
 
 import numpy as np
 import torch
@@ -85,6 +86,10 @@ def parse_args(argv=None):
                              "path on GPU, fp32 accumulate)")
     parser.add_argument("--sharded", action="store_true",
                         help="Shard the model axis across torchrun ranks.")
+    parser.add_argument("--checkpoint-every", type=int, default=0,
+                        help="Save selector state every K steps (0 = off); "
+                             "an interrupted seed resumes mid-run.")
+    parser.add_argument("--checkpoint-dir", default="checkpoints")
     return parser.parse_args(argv)
 
 
@@ -110,6 +115,7 @@ def build_selector(dataset, args, loss_fn, comm=None):
 
 def do_model_selection_experiment(dataset, oracle, args, loss_fn, seed=0,
                                   comm=None, log=True):
+    from coda_amd import checkpoint as ckpt
     comm = comm or get_comm()
     seed_all(seed)
     true_losses = oracle.true_losses(dataset.preds)
@@ -130,13 +136,34 @@ def do_model_selection_experiment(dataset, oracle, args, loss_fn, seed=0,
     if is_main:
         print("Regret at 0:", float(regret_loss))
 
-    cumulative_regret_loss = 0.0
-    iterator = tqdm(range(args.iters)) if is_main else range(args.iters)
+    # mid-run checkpoint/resume (new capability)
+    ckpt_path = None
+    start_m, cumulative_regret_loss = 0, 0.0
+    if args.checkpoint_every:
+        os.makedirs(args.checkpoint_dir, exist_ok=True)
+        ckpt_path = os.path.join(
+            args.checkpoint_dir, f"{args.task}-{args.method}-{seed}.pt")
+        if ckpt.exists(ckpt_path, comm.rank, comm.world):
+            blob = torch.load(
+                ckpt_path + (f".rank{comm.rank}" if comm.world > 1 else ""),
+                map_location="cpu", weights_only=False)
+            ckpt.load_state_dict(selector, blob["selector"])
+            start_m = blob["m"]
+            cumulative_regret_loss = blob["cumulative"]
+            random.setstate(blob["py_rng"])
+            torch.set_rng_state(blob["torch_rng"])
+            if is_main:
+                print(f"Resumed {ckpt_path} at step {start_m}")
+
+    iterator = tqdm(range(start_m, args.iters)) if is_main \
+        else range(start_m, args.iters)
     for m in iterator:
+        t0 = time.perf_counter()
         chosen_idx, selection_prob = selector.get_next_item_to_label()
         true_class = oracle(chosen_idx)
         selector.add_label(chosen_idx, true_class, selection_prob)
         best_model_idx_pred = selector.get_best_model_prediction()
+        step_s = time.perf_counter() - t0
 
         regret_loss = true_losses[best_model_idx_pred] - best_loss
         cumulative_regret_loss += float(regret_loss)
@@ -144,6 +171,22 @@ def do_model_selection_experiment(dataset, oracle, args, loss_fn, seed=0,
             tracking.log_metric("regret", float(regret_loss), step=m + 1)
             tracking.log_metric("cumulative regret",
                                 float(cumulative_regret_loss), step=m + 1)
+            tracking.log_metric("step_seconds", step_s, step=m + 1)
+
+        if ckpt_path and (m + 1) % args.checkpoint_every == 0:
+            blob = {"selector": ckpt.state_dict(selector), "m": m + 1,
+                    "cumulative": cumulative_regret_loss,
+                    "py_rng": random.getstate(),
+                    "torch_rng": torch.get_rng_state()}
+            path = ckpt_path + (f".rank{comm.rank}" if comm.world > 1
+                                else "")
+            torch.save(blob, path + ".tmp")
+            os.replace(path + ".tmp", path)
+
+    if ckpt_path:
+        path = ckpt_path + (f".rank{comm.rank}" if comm.world > 1 else "")
+        if os.path.exists(path):
+            os.remove(path)
 
     return selector.stochastic
 
