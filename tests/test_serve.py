@@ -1,0 +1,82 @@
+"""Serving layer: the interactive step-wise protocol over FastAPI."""
+import pytest
+
+fastapi = pytest.importorskip("fastapi")
+from fastapi.testclient import TestClient  # noqa: E402
+
+from coda_amd.datasets import Dataset, make_synthetic_task  # noqa: E402
+from coda_amd.oracle import Oracle  # noqa: E402
+from coda_amd.options import LOSS_FNS  # noqa: E402
+from coda_amd.serve import create_app  # noqa: E402
+
+
+@pytest.fixture(scope="module")
+def client():
+    preds, labels = make_synthetic_task(H=5, N=150, C=4, seed=3)
+    ds = Dataset.from_tensors(preds, labels, "cpu")
+    oracle = Oracle(ds, LOSS_FNS["acc"])
+    app = create_app(ds, method="coda", oracle=oracle, chunk_size=64)
+    return TestClient(app), oracle
+
+
+def test_full_interactive_loop(client):
+    c, oracle = client
+    st = c.get("/state").json()
+    assert st["step"] == 0
+    for _ in range(3):
+        nxt = c.get("/next").json()
+        idx = nxt["index"]
+        assert 0 <= idx < 150 and nxt["prob"] >= 0
+        label = oracle(idx)
+        st = c.post("/answer", json={"index": idx, "label": label}).json()
+        assert "best_model" in st and "regret" in st
+    assert st["step"] == 3
+    pb = c.get("/pbest").json()["pbest"]
+    assert len(pb) == 5
+    assert abs(sum(pb) - 1.0) < 1e-3
+
+
+def test_skip_removes_point(client):
+    c, _ = client
+    nxt = c.get("/next").json()
+    before = c.get("/state").json()["step"]
+    st = c.post("/skip", json={"index": nxt["index"]}).json()
+    assert st["step"] == before  # no label consumed
+    # the skipped point is never proposed again
+    seen = {c.get("/next").json()["index"] for _ in range(5)}
+    assert nxt["index"] not in seen
+
+
+def test_restart_resets(client):
+    c, _ = client
+    st = c.post("/start", json={"method": "iid"}).json()
+    assert st["step"] == 0 and st["method"] == "iid"
+    nxt = c.get("/next").json()
+    st = c.post("/answer", json={"index": nxt["index"], "label": 0}).json()
+    assert st["step"] == 1
+    # switch back to coda
+    st = c.post("/start", json={"method": "coda"}).json()
+    assert st["step"] == 0
+
+
+def test_html_page(client):
+    c, _ = client
+    r = c.get("/")
+    assert r.status_code == 200 and "coda_amd" in r.text
+
+
+def test_synthetic_builder(tmp_path):
+    import subprocess
+    import sys
+    import torch
+    out = tmp_path / "demo.pt"
+    r = subprocess.run(
+        [sys.executable, "-m", "coda_amd.serve.build_predictions",
+         "--synthetic", "4", "--n", "60", "--classes-n", "5",
+         "--out", str(out)],
+        capture_output=True, text=True, timeout=120,
+        cwd="/root/repo")
+    assert r.returncode == 0, r.stderr
+    t = torch.load(str(out), weights_only=True)
+    assert t.shape == (4, 60, 5)
+    assert (tmp_path / "demo_labels.pt").exists()
