@@ -22,10 +22,19 @@ from typing import Optional, Tuple
 import torch
 
 
+STORAGE_DTYPES = {
+    "fp32": torch.float32,
+    "bf16": torch.bfloat16,
+    "fp8": torch.float8_e4m3fn,
+}
+
+
 class Dataset:
     def __init__(self, filepath: str, device,
                  shard: Optional[Tuple[int, int]] = None,
-                 pin: bool = False):
+                 pin: bool = False,
+                 storage_dtype: str = "fp32",
+                 stream_chunk_mb: int = 0):
         """Load a prediction tensor.
 
         Args:
@@ -34,6 +43,15 @@ class Dataset:
             shard: optional (rank, world_size); keeps models h with
                 h % world_size == rank.
             pin: stage through pinned host memory with an async copy.
+            storage_dtype: on-device storage format - "fp32" (reference
+                parity), "bf16" or "fp8" (e4m3). Compute always up-casts
+                to fp32 (the reference's own fp16-on-disk -> .float()
+                pattern, coda/datasets.py:14, generalized). Halving /
+                quartering storage is what fits large model pools into
+                the 288 GB of HBM3E.
+            stream_chunk_mb: >0 streams the tensor host->device in chunks
+                of this size through pinned buffers on a side stream
+                (dtype conversion on device), instead of one blocking copy.
         """
         self.device = torch.device(device)
         preds = torch.load(filepath, map_location="cpu", weights_only=True)
@@ -46,7 +64,13 @@ class Dataset:
             preds = preds[self.model_idxs]
         else:
             self.model_idxs = torch.arange(self.total_models)
-        self.preds = _to_device(preds, self.device, pin)
+        self.storage_dtype = storage_dtype
+        dt = STORAGE_DTYPES[storage_dtype]
+        if stream_chunk_mb and self.device.type == "cuda":
+            self.preds = _stream_to_device(preds, self.device, dt,
+                                           stream_chunk_mb)
+        else:
+            self.preds = _to_device(preds, self.device, pin).to(dt)
 
         self.labels = None
         label_p = filepath.replace(".pt", "_labels.pt")
@@ -72,6 +96,34 @@ class Dataset:
         self.preds = preds.to(self.device)
         self.labels = labels.to(self.device) if labels is not None else None
         return self
+
+
+def _stream_to_device(t: torch.Tensor, device: torch.device,
+                      dtype: torch.dtype, chunk_mb: int) -> torch.Tensor:
+    """Pinned double-buffered host->HBM streaming with on-device dtype
+    conversion (fp32 on the host wire here; a bf16/fp8 source file would
+    halve wire bytes too). Chunks along the model axis."""
+    H = t.shape[0]
+    out = torch.empty(t.shape, dtype=dtype, device=device)
+    bytes_per_model = t[0].numel() * t.element_size()
+    rows = max(1, (chunk_mb << 20) // max(1, bytes_per_model))
+    side = torch.cuda.Stream(device=device)
+    bufs = [torch.empty((min(rows, H),) + t.shape[1:],
+                        dtype=t.dtype).pin_memory() for _ in range(2)]
+    events = [torch.cuda.Event(), torch.cuda.Event()]
+    for e in events:
+        e.record(side)
+    for i, h0 in enumerate(range(0, H, rows)):
+        h1 = min(h0 + rows, H)
+        buf = bufs[i % 2]
+        events[i % 2].synchronize()  # buffer free to reuse
+        buf[:h1 - h0].copy_(t[h0:h1])
+        with torch.cuda.stream(side):
+            staged = buf[:h1 - h0].to(device, non_blocking=True)
+            out[h0:h1].copy_(staged.to(dtype))
+            events[i % 2].record(side)
+    torch.cuda.current_stream(device).wait_stream(side)
+    return out
 
 
 def _to_device(t: torch.Tensor, device: torch.device, pin: bool) -> torch.Tensor:

@@ -43,8 +43,24 @@ LOG_CLAMP = 80.0
 # ---------------------------------------------------------------------------
 
 def consensus(preds: torch.Tensor) -> torch.Tensor:
-    """Mean over the model axis: (H, N, C) -> (N, C)."""
-    return preds.mean(dim=0)
+    """Mean over the model axis: (H, N, C) -> (N, C), fp32."""
+    return init_model_stats(preds)[1] / preds.shape[0]
+
+
+def init_model_stats(preds: torch.Tensor, chunk_h: int = 16):
+    """One chunked fp32-upcast pass over the model axis at selector init:
+    cached argmax classes (H, N) + the consensus sum (N, C) fp32.
+    Storage may be fp32 / bf16 / fp8 (coda_amd.datasets.STORAGE_DTYPES);
+    compute is always fp32."""
+    H, N, C = preds.shape
+    classes = torch.empty(H, N, dtype=torch.long, device=preds.device)
+    ens = torch.zeros(N, C, dtype=torch.float32, device=preds.device)
+    for h0 in range(0, H, chunk_h):
+        h1 = min(h0 + chunk_h, H)
+        ch = preds[h0:h1].float()
+        classes[h0:h1] = ch.argmax(-1)
+        ens += ch.sum(0)
+    return classes, ens
 
 
 def confusion_prior(pseudo_labels: torch.Tensor, preds: torch.Tensor) -> torch.Tensor:
@@ -55,11 +71,16 @@ def confusion_prior(pseudo_labels: torch.Tensor, preds: torch.Tensor) -> torch.T
     the one-hot einsum, which never materializes the (N, C) one-hot.
     """
     H, N, C = preds.shape
-    conf = preds.new_zeros(H, C, C)
-    # Flatten (H, C) target rows: row index = h*C + label[n]
-    idx = (pseudo_labels.unsqueeze(0) +
-           torch.arange(H, device=preds.device).unsqueeze(1) * C)  # (H, N)
-    conf.view(H * C, C).index_add_(0, idx.reshape(-1), preds.reshape(-1, C))
+    conf = torch.zeros(H, C, C, dtype=torch.float32, device=preds.device)
+    # Flatten (H, C) target rows: row index = h*C + label[n]; chunk over H
+    # with fp32 upcast so bf16/fp8 storage accumulates exactly.
+    chunk_h = max(1, min(H, (1 << 28) // max(1, N * C)))
+    for h0 in range(0, H, chunk_h):
+        h1 = min(h0 + chunk_h, H)
+        idx = (pseudo_labels.unsqueeze(0) +
+               torch.arange(h0, h1, device=preds.device).unsqueeze(1) * C)
+        conf.view(H * C, C).index_add_(0, idx.reshape(-1),
+                                       preds[h0:h1].float().reshape(-1, C))
     return conf / conf.sum(-1, keepdim=True).clamp_min(1e-6)
 
 
@@ -106,14 +127,26 @@ def pi_hat_partial(dirichlets: torch.Tensor, preds: torch.Tensor,
     reference bit-for-bit semantics (coda/coda.py:227).
     """
     H = preds.shape[0]
-    dirichlets = dirichlets.to(preds.dtype)
     out = None
     for h0 in range(0, H, chunk_h):
         h1 = min(h0 + chunk_h, H)
-        part = torch.bmm(preds[h0:h1], dirichlets[h0:h1].transpose(1, 2))
-        part = part.float().sum(0)
+        part = torch.bmm(preds[h0:h1].float(),
+                         dirichlets[h0:h1].float().transpose(1, 2)).sum(0)
         out = part if out is None else out + part
     return out
+
+
+def pi_hat_pack_chunked(preds: torch.Tensor, chunk_h: int = 16):
+    """pi_hat_pack for bf16/fp8 storage: fills the (N, H*C) bf16 operand
+    model-by-model without materializing a permuted fp32 copy."""
+    H, N, C = preds.shape
+    packed = torch.empty(N, H * C, dtype=torch.bfloat16,
+                         device=preds.device)
+    view = packed.view(N, H, C)
+    for h0 in range(0, H, chunk_h):
+        h1 = min(h0 + chunk_h, H)
+        view[:, h0:h1] = preds[h0:h1].to(torch.bfloat16).permute(1, 0, 2)
+    return packed
 
 
 def pi_hat_pack(preds: torch.Tensor) -> torch.Tensor:
@@ -246,8 +279,12 @@ def eig_assemble(pbest_hyp: torch.Tensor, pbest_before: torch.Tensor,
 # ---------------------------------------------------------------------------
 
 def pred_classes(preds: torch.Tensor) -> torch.Tensor:
-    """Argmax over classes: (H, N, C) -> (H, N) int64. Cached by callers."""
-    return preds.argmax(dim=-1)
+    """Argmax over classes: (H, N, C) -> (H, N) int64. Cached by callers.
+    Up-casts low-precision storage dtypes that lack an argmax kernel."""
+    try:
+        return preds.argmax(dim=-1)
+    except RuntimeError:
+        return init_model_stats(preds)[0]
 
 
 def disagreement_mask(classes: torch.Tensor) -> torch.Tensor:

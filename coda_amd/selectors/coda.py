@@ -60,19 +60,21 @@ class CODA(ModelSelector):
         self.update_strength = learning_rate
 
         preds = dataset.preds
-        # cached argmax classes (Hl, N) - never change, computed once
-        self.classes = ops.pred_classes(preds)
+        # one chunked fp32-upcast pass: cached argmax classes (Hl, N) +
+        # consensus sum (storage may be fp32/bf16/fp8)
+        self.classes, ens_sum = ops.init_model_stats(preds)
 
         # pi_hat compute dtype: bf16 MFMA on GPU (one packed (N, H*C) GEMM,
         # ~16x the f32 matrix rate, f32 accumulation), fp32 elsewhere.
         if pi_hat_precision == "auto":
             pi_hat_precision = "bf16" if preds.is_cuda else "fp32"
+        if preds.dtype == torch.float8_e4m3fn and pi_hat_precision == "fp32":
+            pi_hat_precision = "bf16"  # fp8 storage has no fp32 bmm path
         self.pi_hat_precision = pi_hat_precision
-        self._pi_packed = ops.pi_hat_pack(preds) \
+        self._pi_packed = ops.pi_hat_pack_chunked(preds) \
             if pi_hat_precision == "bf16" else None
 
         # consensus prior: global mean over H (all-reduce site K1)
-        ens_sum = preds.sum(dim=0)
         self.comm.all_reduce_(ens_sum)
         pseudo = (ens_sum / self.H).argmax(-1)            # (N,) global pseudo-labels
         soft_conf = ops.confusion_prior(pseudo, preds)    # (Hl, C, C)

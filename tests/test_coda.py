@@ -114,3 +114,31 @@ def test_chunk_size_invariance(synthetic_dataset):
     e2, c2 = s2.eig_batched()
     assert c1 == c2
     torch.testing.assert_close(e1, e2, rtol=1e-6, atol=1e-7)
+
+
+@pytest.mark.parametrize("dtype", ["bf16", "fp8"])
+def test_low_precision_storage(synthetic_task, dtype, tmp_path):
+    """CODA runs on bf16/fp8 storage; trajectories match fp32 on a
+    well-separated task (compute always up-casts to fp32)."""
+    from coda_amd.datasets import write_synthetic_task
+    path = write_synthetic_task(str(tmp_path), name="lp", H=6, N=200, C=4,
+                                seed=9, best_acc=0.95, worst_acc=0.4)
+
+    def run(storage):
+        ds = Dataset(path, "cpu", storage_dtype=storage)
+        oracle = Oracle(ds, LOSS_FNS["acc"])
+        random.seed(0); torch.manual_seed(0)
+        sel = CODA(ds, chunk_size=64)
+        traj = []
+        for _ in range(3):
+            idx, q = sel.get_next_item_to_label()
+            sel.add_label(idx, oracle(int(idx)), q)
+            traj.append(int(idx))
+        return traj, int(sel.get_best_model_prediction())
+
+    t32, b32 = run("fp32")
+    tlp, blp = run(dtype)
+    assert blp == b32
+    # bf16 rounds probabilities to ~3 decimal digits; selections can
+    if dtype == "bf16":
+        assert tlp == t32
