@@ -31,6 +31,9 @@ from .. import ops
 from ..ops import sharded as shops
 from ..util import DEBUG, _check
 
+import os
+DEBUG_VIZ = os.environ.get("CODA_AMD_DEBUG_VIZ") == "1"
+
 
 class CODA(ModelSelector):
     def __init__(self, dataset,
@@ -195,6 +198,15 @@ class CODA(ModelSelector):
         else:
             raise NotImplementedError(self.q)
 
+        if DEBUG_VIZ:  # per-step EIG bar chart (reference coda/coda.py:299-303)
+            from ..util import plot_bar
+            from .. import tracking
+            try:
+                tracking.log_image(plot_bar(q_vals, title="EIG"),
+                                   key="EIG", step=self.step)
+            except RuntimeError:
+                pass  # no active tracking run
+
         # greedy with seeded random tie-breaking (coda/coda.py:306-313)
         best = q_vals.max()
         ties = torch.isclose(q_vals, best, rtol=1e-8)
@@ -235,5 +247,13 @@ class CODA(ModelSelector):
 
     def get_best_model_prediction(self):
         pbest = self.get_pbest()
+        if DEBUG_VIZ:  # reference coda/coda.py:337-341
+            from ..util import plot_bar
+            from .. import tracking
+            try:
+                tracking.log_image(plot_bar(pbest, title="PBest"),
+                                   key="PBest", step=self.step)
+            except RuntimeError:
+                pass
         self.step += 1
         return torch.argmax(pbest)
