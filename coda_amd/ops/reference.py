@@ -176,6 +176,30 @@ def pi_hat_partial_packed(dirichlets: torch.Tensor,
     return (packed @ B).float()
 
 
+def pi_hat_delta(preds: torch.Tensor, point_classes: torch.Tensor,
+                 chunk_h: int = 32) -> torch.Tensor:
+    """Incremental pi_hat update term for one labeled point.
+
+    add_label moves ONLY row `true_class` of each model's Dirichlet
+    (coda/coda.py:316-317: D[h, y, s] += lr*[s == argmax_h(idx)]), so the
+    posterior marginal's change is rank-1:
+        adjusted[n, y] += lr * sum_h preds[h, n, cls_h]
+    with cls_h = model h's argmax class at the labeled point. This op
+    returns sum_h preds[h, :, cls_h] -> (N,) fp32: an O(H*N) gather-sum
+    replacing the O(H*N*C^2) full contraction per step (exact in exact
+    arithmetic; fp32 += drift is ~1e-7/step).
+
+    preds: (H, N, C) any storage dtype; point_classes: (H,) int64.
+    """
+    H, N, C = preds.shape
+    out = torch.zeros(N, dtype=torch.float32, device=preds.device)
+    idx = point_classes.view(H, 1, 1).expand(H, N, 1)
+    for h0 in range(0, H, chunk_h):
+        h1 = min(h0 + chunk_h, H)
+        out += preds[h0:h1].float().gather(2, idx[h0:h1]).squeeze(-1).sum(0)
+    return out
+
+
 def pi_hat_normalize(adjusted_sum: torch.Tensor):
     """(N, C) unnormalized -> (pi_hat_xi (N,C), pi_hat (C,))."""
     pi_xi = adjusted_sum / adjusted_sum.sum(dim=-1, keepdim=True).clamp_min(1e-12)

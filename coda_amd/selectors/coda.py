@@ -76,6 +76,7 @@ class CODA(ModelSelector):
         self.pi_hat_precision = pi_hat_precision
         self._pi_packed = ops.pi_hat_pack_chunked(preds) \
             if pi_hat_precision == "bf16" else None
+        self._adjusted = None  # maintained incrementally after init
 
         # consensus prior: global mean over H (all-reduce site K1)
         self.comm.all_reduce_(ens_sum)
@@ -130,13 +131,23 @@ class CODA(ModelSelector):
         return idxs
 
     def update_pi_hat(self):
-        """Confusion-adjusted class marginals (K5; all-reduce over H)."""
+        """Confusion-adjusted class marginals (K5; all-reduce over H).
+
+        Runs the FULL contraction (one packed bf16 MFMA GEMM on GPU).
+        Called once at init and by checkpoint restore; per-label updates
+        go through the exact rank-1 incremental path in add_label
+        (ops.pi_hat_delta), which is O(H*N) instead of O(H*N*C^2).
+        """
         if self._pi_packed is not None:
             adjusted = ops.pi_hat_partial_packed(self.dirichlets,
                                                  self._pi_packed)
+            # the packed operand is only needed for full refreshes; drop it
+            # to reclaim the (N, H*C) bf16 buffer once initialized
+            self._pi_packed = None
         else:
             adjusted = ops.pi_hat_partial(self.dirichlets, self.dataset.preds)
         self.comm.all_reduce_(adjusted)
+        self._adjusted = adjusted
         self.pi_hat_xi, self.pi_hat = ops.pi_hat_normalize(adjusted)
 
     # ------------------------------------------------------------------
@@ -220,12 +231,20 @@ class CODA(ModelSelector):
         return cand[idx_local], float(q_vals[idx_local])
 
     def add_label(self, idx, true_class, selection_prob):
-        """Posterior update (K13) + pi_hat refresh (coda/coda.py:315-323)."""
+        """Posterior update (K13) + incremental pi_hat refresh.
+
+        Reference semantics (coda/coda.py:315-323); the pi_hat refresh is
+        the exact rank-1 increment (only Dirichlet row `true_class`
+        moved), all-reduced over shards.
+        """
         idx = int(idx)
         onehot = torch.nn.functional.one_hot(
             self.classes[:, idx], self.C).to(self.dirichlets.dtype)  # (Hl, C)
         self.dirichlets[:, int(true_class)] += self.update_strength * onehot
-        self.update_pi_hat()
+        delta = ops.pi_hat_delta(self.dataset.preds, self.classes[:, idx])
+        self.comm.all_reduce_(delta)
+        self._adjusted[:, int(true_class)] += self.update_strength * delta
+        self.pi_hat_xi, self.pi_hat = ops.pi_hat_normalize(self._adjusted)
         self.labeled_idxs.append(idx)
         self.labels.append(int(true_class))
         self.q_vals.append(selection_prob)

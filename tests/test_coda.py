@@ -142,3 +142,21 @@ def test_low_precision_storage(synthetic_task, dtype, tmp_path):
     # bf16 rounds probabilities to ~3 decimal digits; selections can
     if dtype == "bf16":
         assert tlp == t32
+
+
+def test_incremental_pi_hat_matches_full():
+    """The rank-1 incremental pi_hat maintained by add_label equals a full
+    recomputation from the updated Dirichlets (to fp32 roundoff)."""
+    from coda_amd import ops
+    preds, labels = make_synthetic_task(H=6, N=250, C=5, seed=13)
+    ds = Dataset.from_tensors(preds, labels, "cpu")
+    oracle = Oracle(ds, LOSS_FNS["acc"])
+    random.seed(0); torch.manual_seed(0)
+    sel = CODA(ds, chunk_size=64)
+    for _ in range(5):
+        i, q = sel.get_next_item_to_label()
+        sel.add_label(i, oracle(int(i)), q)
+    full = ops.pi_hat_partial(sel.dirichlets, ds.preds)
+    torch.testing.assert_close(sel._adjusted, full, rtol=1e-4, atol=1e-5)
+    xi_full, pi_full = ops.pi_hat_normalize(full)
+    torch.testing.assert_close(sel.pi_hat, pi_full, rtol=1e-4, atol=1e-6)

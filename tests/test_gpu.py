@@ -182,3 +182,34 @@ class TestShardedPhases:
         got = S.eig_chunk_sharded(a0, b0, cls, pb, pi, pi_xi, m0s, H0s,
                                   comm)
         torch.testing.assert_close(got, want, rtol=1e-4, atol=1e-5)
+
+
+class TestStorageGPU:
+    def test_streamed_ingest_matches_direct(self, dev, tmp_path):
+        from coda_amd.datasets import Dataset, write_synthetic_task
+        path = write_synthetic_task(str(tmp_path), name="st", H=6, N=400,
+                                    C=8, seed=11)
+        direct = Dataset(path, dev, storage_dtype="bf16")
+        streamed = Dataset(path, dev, storage_dtype="bf16",
+                           stream_chunk_mb=1)
+        torch.cuda.synchronize()
+        assert torch.equal(direct.preds, streamed.preds)
+
+    def test_fp8_storage_coda_runs(self, dev, tmp_path):
+        from coda_amd import CODA, Oracle
+        from coda_amd.datasets import Dataset, write_synthetic_task
+        from coda_amd.options import LOSS_FNS
+        path = write_synthetic_task(str(tmp_path), name="f8", H=6, N=300,
+                                    C=5, seed=12, best_acc=0.95,
+                                    worst_acc=0.4)
+        ds = Dataset(path, dev, storage_dtype="fp8")
+        assert ds.preds.dtype == torch.float8_e4m3fn
+        oracle = Oracle(ds, LOSS_FNS["acc"])
+        random.seed(0); torch.manual_seed(0)
+        sel = CODA(ds, chunk_size=64)
+        for _ in range(3):
+            i, q = sel.get_next_item_to_label()
+            sel.add_label(i, oracle(int(i)), q)
+        p = sel.get_pbest()
+        assert torch.isfinite(p).all()
+        assert abs(float(p.sum()) - 1.0) < 1e-3
