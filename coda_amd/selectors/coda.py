@@ -46,7 +46,8 @@ class CODA(ModelSelector):
                  comm: Comm = None,
                  chunk_size: int = 100,
                  num_points: int = 256,
-                 pi_hat_precision: str = "auto"):
+                 pi_hat_precision: str = "auto",
+                 eig_impl: str = "auto"):
         self.dataset = dataset
         self.device = dataset.preds.device
         self.comm = comm or get_comm()
@@ -57,6 +58,10 @@ class CODA(ModelSelector):
         self.q = q
         self.chunk_size = chunk_size
         self.num_points = num_points
+        # EIG implementation: 'fused' = the wave-per-row HIP kernel (GPU) /
+        # eager (CPU); 'table' = the factored per-step curve tables + MFMA
+        # GEMM (ops/table.py). 'auto' resolves per device at eig time.
+        self.eig_impl = eig_impl
 
         # hyperparams (reference names: coda/coda.py:189-190)
         self.prior_strength = 1.0 - alpha
@@ -173,6 +178,16 @@ class CODA(ModelSelector):
         else:
             mixture0, H_before = ops.mixture_entropy(pbest_before, self.pi_hat)
 
+        impl = self.eig_impl
+        if impl == "auto":
+            impl = "table" if (self.device.type == "cuda"
+                               and not self.comm.is_distributed) else "fused"
+        tables = None
+        if impl == "table" and not self.comm.is_distributed:
+            from ..ops import table as tops
+            tables = tops.table_precompute(alpha_cc, beta_cc,
+                                           num_points=self.num_points)
+
         eig_chunks = []
         for s in range(0, cand.numel(), self.chunk_size):
             ids = cand[s:s + self.chunk_size]
@@ -183,6 +198,11 @@ class CODA(ModelSelector):
                     alpha_cc, beta_cc, chunk_classes, pbest_before,
                     self.pi_hat, pi_xi, mixture0, H_before, self.comm,
                     num_points=self.num_points)
+            elif tables is not None:
+                from ..ops import table as tops
+                eig = tops.eig_chunk_table(
+                    tables, chunk_classes, pbest_before, self.pi_hat,
+                    pi_xi, mixture0, H_before)
             else:
                 eig = ops.eig_chunk(
                     alpha_cc, beta_cc, chunk_classes, pbest_before,

@@ -260,3 +260,66 @@ class TestPiHatPacked:
         got = ops.pi_hat_partial_packed(D, packed)
         # bf16 inputs: ~0.4% relative tolerance
         torch.testing.assert_close(got, want, rtol=2e-2, atol=1e-3)
+
+
+class TestTableEig:
+    """v2 (table-factored) EIG vs v1 (fused composition) equivalence."""
+
+    def _setup(self, seed=31, H=7, C=5, B=11):
+        g = torch.Generator().manual_seed(seed)
+        a0 = torch.rand(H, C, generator=g) * 10 + 1
+        b0 = torch.rand(H, C, generator=g) * 10 + 1
+        cls = torch.randint(0, C, (B, H), generator=g)
+        pi = torch.softmax(torch.rand(C, generator=g), 0)
+        pi_xi = torch.softmax(torch.rand(B, C, generator=g), -1)
+        pb = ops.pbest_from_beta(a0.t().contiguous(), b0.t().contiguous())
+        m0, H0 = ops.mixture_entropy(pb, pi)
+        return a0, b0, cls, pi, pi_xi, pb, m0, H0
+
+    def test_pbest_hyp_matches_v1(self):
+        from coda_amd.ops import table as T
+        a0, b0, cls, pi, pi_xi, pb, m0, H0 = self._setup()
+        B, H = cls.shape
+        C = a0.shape[1]
+        tables = T.table_precompute(a0, b0)
+        got = T.pbest_hyp_table(tables, cls)
+        ah, bh = ops.hypothetical_betas(a0, b0, cls, 1.0)
+        want = ops.pbest_from_beta(ah.reshape(B * C, H),
+                                   bh.reshape(B * C, H)).reshape(B, C, H)
+        torch.testing.assert_close(got, want, rtol=2e-3, atol=1e-5)
+
+    def test_eig_matches_v1(self):
+        from coda_amd.ops import table as T
+        a0, b0, cls, pi, pi_xi, pb, m0, H0 = self._setup(seed=32)
+        tables = T.table_precompute(a0, b0)
+        got = T.eig_chunk_table(tables, cls, pb, pi, pi_xi, m0, H0)
+        want = ops.eig_chunk(a0, b0, cls, pb, pi, pi_xi, m0, H0)
+        torch.testing.assert_close(got, want, rtol=5e-3, atol=1e-5)
+
+    def test_concentrated_betas_finite(self):
+        from coda_amd.ops import table as T
+        g = torch.Generator().manual_seed(33)
+        H, C, B = 5, 4, 6
+        a0 = torch.rand(H, C, generator=g) * 3000 + 100
+        b0 = torch.rand(H, C, generator=g) * 3000 + 100
+        cls = torch.randint(0, C, (B, H), generator=g)
+        tables = T.table_precompute(a0, b0)
+        pb = T.pbest_hyp_table(tables, cls)
+        assert torch.isfinite(pb).all()
+        torch.testing.assert_close(pb.sum(-1), torch.ones(B, C), atol=2e-3,
+                                   rtol=0)
+
+    def test_selector_table_impl_matches_fused(self):
+        import random
+        from coda_amd import CODA
+        from coda_amd.datasets import Dataset, make_synthetic_task
+        preds, labels = make_synthetic_task(H=6, N=200, C=4, seed=34)
+        ds = Dataset.from_tensors(preds, labels, "cpu")
+        random.seed(0); torch.manual_seed(0)
+        s1 = CODA(ds, chunk_size=64, eig_impl="fused")
+        e1, c1 = s1.eig_batched()
+        random.seed(0); torch.manual_seed(0)
+        s2 = CODA(ds, chunk_size=64, eig_impl="table")
+        e2, c2 = s2.eig_batched()
+        assert c1 == c2
+        torch.testing.assert_close(e1, e2, rtol=5e-3, atol=1e-5)
