@@ -67,6 +67,8 @@ def load_state_dict(selector, state: Dict[str, Any]):
         setattr(selector, f, v)
     # derived state that depends on the posterior
     if cls == "CODA":
+        selector._tables = None       # force a fresh v2 table build
+        selector._tables_dirty = set()
         selector.update_pi_hat()
     return selector
 

@@ -121,3 +121,23 @@ def eig_chunk_table(tables: EigTables, chunk_classes: torch.Tensor,
     pb = pbest_hyp_table(tables, chunk_classes)
     return eig_assemble(pb, pbest_before, pi_hat, pi_hat_xi_chunk,
                         mixture0, H_before)
+
+
+def table_update_rows(tables: EigTables, alpha_cc: torch.Tensor,
+                      beta_cc: torch.Tensor, rows,
+                      update_weight: float = 1.0) -> EigTables:
+    """Recompute the table slices for the given class rows in place.
+
+    add_label moves only Dirichlet row `true_class` of each model
+    (coda/coda.py:316-317), so between steps only that class's H*2
+    curves change - an O(1/C) refresh instead of a full rebuild.
+    """
+    rows = torch.as_tensor(rows, device=alpha_cc.device, dtype=torch.long)
+    if rows.numel() == 0:
+        return tables
+    sub = table_precompute(alpha_cc[:, rows], beta_cc[:, rows],
+                           update_weight, tables.EG.shape[-1])
+    tables.EG[rows] = sub.EG
+    tables.delta[rows] = sub.delta
+    tables.s_base[rows] = sub.s_base
+    return tables

@@ -62,6 +62,8 @@ class CODA(ModelSelector):
         # eager (CPU); 'table' = the factored per-step curve tables + MFMA
         # GEMM (ops/table.py). 'auto' resolves per device at eig time.
         self.eig_impl = eig_impl
+        self._tables = None          # persistent v2 curve tables
+        self._tables_dirty = set()   # class rows touched since last build
 
         # hyperparams (reference names: coda/coda.py:189-190)
         self.prior_strength = 1.0 - alpha
@@ -185,8 +187,14 @@ class CODA(ModelSelector):
         tables = None
         if impl == "table" and not self.comm.is_distributed:
             from ..ops import table as tops
-            tables = tops.table_precompute(alpha_cc, beta_cc,
-                                           num_points=self.num_points)
+            if self._tables is None:
+                self._tables = tops.table_precompute(
+                    alpha_cc, beta_cc, num_points=self.num_points)
+            elif self._tables_dirty:
+                tops.table_update_rows(self._tables, alpha_cc, beta_cc,
+                                       sorted(self._tables_dirty))
+            self._tables_dirty.clear()
+            tables = self._tables
 
         eig_chunks = []
         for s in range(0, cand.numel(), self.chunk_size):
@@ -261,6 +269,7 @@ class CODA(ModelSelector):
         onehot = torch.nn.functional.one_hot(
             self.classes[:, idx], self.C).to(self.dirichlets.dtype)  # (Hl, C)
         self.dirichlets[:, int(true_class)] += self.update_strength * onehot
+        self._tables_dirty.add(int(true_class))
         delta = ops.pi_hat_delta(self.dataset.preds, self.classes[:, idx])
         self.comm.all_reduce_(delta)
         self._adjusted[:, int(true_class)] += self.update_strength * delta

@@ -323,3 +323,18 @@ class TestTableEig:
         e2, c2 = s2.eig_batched()
         assert c1 == c2
         torch.testing.assert_close(e1, e2, rtol=5e-3, atol=1e-5)
+
+    def test_incremental_table_update_matches_rebuild(self):
+        from coda_amd.ops import table as T
+        g = torch.Generator().manual_seed(35)
+        H, C = 6, 5
+        a0 = torch.rand(H, C, generator=g) * 10 + 1
+        b0 = torch.rand(H, C, generator=g) * 10 + 1
+        tables = T.table_precompute(a0, b0)
+        # perturb class rows 1 and 3 (what add_label does)
+        a0[:, 1] += 0.01; b0[:, 3] += 0.02
+        T.table_update_rows(tables, a0, b0, [1, 3])
+        fresh = T.table_precompute(a0, b0)
+        torch.testing.assert_close(tables.EG, fresh.EG)
+        torch.testing.assert_close(tables.delta, fresh.delta)
+        torch.testing.assert_close(tables.s_base, fresh.s_base)
