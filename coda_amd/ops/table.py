@@ -141,3 +141,60 @@ def table_update_rows(tables: EigTables, alpha_cc: torch.Tensor,
     tables.delta[rows] = sub.delta
     tables.s_base[rows] = sub.s_base
     return tables
+
+
+# ---------------------------------------------------------------------------
+# Sharded v2: the model axis is split across ranks; each rank holds table
+# slices for its local models. The cross-rank coupling is carried by an
+# all-gather of the SELECTED delta curves - (B, H, P) floats per chunk,
+# world-times smaller than all-reducing the dense (B, C, P) slog - plus
+# a once-per-step (C, P) all-reduce for the base term and two tiny (B, C)
+# all-reduces (normalizer, entropy partials).
+# ---------------------------------------------------------------------------
+
+def s_base_global(tables: EigTables, comm) -> torch.Tensor:
+    """sum over ALL models of log2 cdf_v0: all-reduced (C, P)."""
+    out = tables.s_base.clone()
+    comm.all_reduce_(out)
+    return out
+
+
+def eig_chunk_table_sharded(tables: EigTables, s_base_all: torch.Tensor,
+                            cls_local: torch.Tensor,
+                            pbest_before_local: torch.Tensor,
+                            pi_hat: torch.Tensor,
+                            pi_hat_xi_chunk: torch.Tensor,
+                            mixture0_local: torch.Tensor,
+                            H_before: torch.Tensor, comm) -> torch.Tensor:
+    """(B,) EIG with local tables; identical on every rank."""
+    EG, delta, _, w = tables
+    C, Hl, _, P = EG.shape
+    B = cls_local.shape[0]
+    cls_l = cls_local.long()
+
+    flat = delta.permute(1, 0, 2).reshape(Hl * C, P)
+    sel = flat[(torch.arange(Hl, device=cls_l.device) * C).unsqueeze(0)
+               + cls_l]                                      # (B, Hl, P)
+    sel_all = comm.all_gather_cat(sel, dim=1)                # (B, H, P)
+    cls_all = comm.all_gather_cat(cls_l, dim=1)              # (B, H)
+    Hg = cls_all.shape[1]
+
+    slog = s_base_all.unsqueeze(0).repeat(B, 1, 1)           # (B, C, P)
+    slog.scatter_add_(1, cls_all.unsqueeze(-1).expand(B, Hg, P), sel_all)
+    ES = torch.exp2(slog) * w
+
+    M = torch.bmm(EG.reshape(C, Hl * 2, P),
+                  ES.permute(1, 2, 0).contiguous())          # (C, 2Hl, B)
+    Mp = M.view(C, Hl, 2, B).permute(3, 0, 1, 2)             # (B, C, Hl, 2)
+    eq = (cls_l.unsqueeze(1) ==
+          torch.arange(C, device=cls_l.device).view(1, C, 1)).long()
+    pb = Mp.gather(3, eq.unsqueeze(-1)).squeeze(-1)          # (B, C, Hl)
+    tot = pb.sum(-1)                                         # (B, C)
+    comm.all_reduce_(tot)
+    pb = pb / tot.clamp_min(EPS_PROB).unsqueeze(-1)
+
+    d = pi_hat.view(1, C, 1) * (pb - pbest_before_local.unsqueeze(0))
+    m = (mixture0_local.view(1, 1, Hl) + d).clamp_min(1e-12)
+    H_after = -(m * m.log2()).sum(-1)                        # (B, C) partial
+    comm.all_reduce_(H_after)
+    return H_before - (pi_hat_xi_chunk * H_after).sum(-1)

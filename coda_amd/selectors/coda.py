@@ -182,10 +182,9 @@ class CODA(ModelSelector):
 
         impl = self.eig_impl
         if impl == "auto":
-            impl = "table" if (self.device.type == "cuda"
-                               and not self.comm.is_distributed) else "fused"
-        tables = None
-        if impl == "table" and not self.comm.is_distributed:
+            impl = "table" if self.device.type == "cuda" else "fused"
+        tables = s_base_all = None
+        if impl == "table":
             from ..ops import table as tops
             if self._tables is None:
                 self._tables = tops.table_precompute(
@@ -195,13 +194,20 @@ class CODA(ModelSelector):
                                        sorted(self._tables_dirty))
             self._tables_dirty.clear()
             tables = self._tables
+            if self.comm.is_distributed:
+                s_base_all = tops.s_base_global(tables, self.comm)
 
         eig_chunks = []
         for s in range(0, cand.numel(), self.chunk_size):
             ids = cand[s:s + self.chunk_size]
             chunk_classes = self.classes[:, ids].t().contiguous()  # (B, Hl)
             pi_xi = self.pi_hat_xi[ids]                            # (B, C)
-            if self.comm.is_distributed:
+            if self.comm.is_distributed and tables is not None:
+                from ..ops import table as tops
+                eig = tops.eig_chunk_table_sharded(
+                    tables, s_base_all, chunk_classes, pbest_before,
+                    self.pi_hat, pi_xi, mixture0, H_before, self.comm)
+            elif self.comm.is_distributed:
                 eig = shops.eig_chunk_sharded(
                     alpha_cc, beta_cc, chunk_classes, pbest_before,
                     self.pi_hat, pi_xi, mixture0, H_before, self.comm,

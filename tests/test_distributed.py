@@ -89,3 +89,62 @@ def test_sharded_equals_single(tmp_path):
                                    single["pbest"], rtol=1e-4, atol=1e-6)
     # ranks agree with each other exactly on collective-derived outputs
     assert results[0]["pbest"] == results[1]["pbest"]
+
+
+def _worker_table(rank, world, init_file, preds, labels, steps, q):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank))
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=world)
+    try:
+        from coda_amd import CODA, Oracle
+        from coda_amd.options import LOSS_FNS
+        from coda_amd.parallel import Comm
+        comm = Comm(rank=rank, world=world, device=torch.device("cpu"))
+        ds = Dataset.from_tensors(preds, labels, "cpu", shard=(rank, world))
+        oracle = Oracle(ds, LOSS_FNS["acc"])
+        random.seed(0); torch.manual_seed(0)
+        sel = CODA(ds, comm=comm, chunk_size=64, eig_impl="table")
+        choices = []
+        for _ in range(steps):
+            idx, qv = sel.get_next_item_to_label()
+            sel.add_label(idx, oracle(idx), qv)
+            choices.append((int(idx), round(float(qv), 5)))
+        q.put((rank, choices))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_sharded_table_impl_equals_single(tmp_path):
+    """The table-factored (v2) sharded path: same selections as the
+    single-process table path."""
+    preds, labels = make_synthetic_task(H=7, N=200, C=4, seed=5)
+
+    from coda_amd import CODA, Oracle
+    from coda_amd.options import LOSS_FNS
+    ds = Dataset.from_tensors(preds, labels, "cpu")
+    oracle = Oracle(ds, LOSS_FNS["acc"])
+    random.seed(0); torch.manual_seed(0)
+    sel = CODA(ds, chunk_size=64, eig_impl="table")
+    single = []
+    for _ in range(4):
+        idx, qv = sel.get_next_item_to_label()
+        sel.add_label(idx, oracle(idx), qv)
+        single.append((int(idx), round(float(qv), 5)))
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    init_file = str(tmp_path / "pg_init_tbl")
+    procs = [ctx.Process(target=_worker_table,
+                         args=(r, 2, init_file, preds, labels, 4, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, res = q.get(timeout=240)
+        results[rank] = res
+    for p in procs:
+        p.join(timeout=60)
+    assert results[0] == single and results[1] == single
