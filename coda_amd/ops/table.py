@@ -104,12 +104,20 @@ def pbest_hyp_table(tables: EigTables,
     slog.scatter_add_(1, cls.unsqueeze(-1).expand(B, H, P), sel)
 
     ES = torch.exp2(slog) * w                               # (B, C, P)
+    # stride-aware bmm: (P, B) columns are contiguous in p, so rocBLAS
+    # consumes the permuted view directly (no 262 MB transposed copy)
     M = torch.bmm(EG.reshape(C, H * 2, P),
-                  ES.permute(1, 2, 0).contiguous())         # (C, 2H, B)
-    Mp = M.view(C, H, 2, B).permute(3, 0, 1, 2)             # (B, C, H, 2)
-    eq = (cls.unsqueeze(1) ==
-          torch.arange(C, device=cls.device).view(1, C, 1)).long()
-    pb = Mp.gather(3, eq.unsqueeze(-1)).squeeze(-1)         # (B, C, H)
+                  ES.permute(1, 2, 0))                      # (C, 2H, B)
+    Mv = M.view(C, H, 2, B)
+    # v-selection: v=1 only where cls(b,h) == c - exactly H of the C*H
+    # positions per candidate. Start from the v=0 plane and scatter the
+    # H hit values per candidate (tiny) instead of a dense (B,C,H,2)
+    # gather.
+    pb = Mv[:, :, 0, :].permute(2, 0, 1).contiguous()       # (B, C, H)
+    ar_h = torch.arange(H, device=cls.device)
+    ar_b = torch.arange(B, device=cls.device)
+    hit_vals = Mv[cls, ar_h.view(1, H), 1, ar_b.view(B, 1)]  # (B, H)
+    pb[ar_b.view(B, 1), cls, ar_h.view(1, H)] = hit_vals
     return pb / pb.sum(-1, keepdim=True).clamp_min(EPS_PROB)
 
 
@@ -184,11 +192,13 @@ def eig_chunk_table_sharded(tables: EigTables, s_base_all: torch.Tensor,
     ES = torch.exp2(slog) * w
 
     M = torch.bmm(EG.reshape(C, Hl * 2, P),
-                  ES.permute(1, 2, 0).contiguous())          # (C, 2Hl, B)
-    Mp = M.view(C, Hl, 2, B).permute(3, 0, 1, 2)             # (B, C, Hl, 2)
-    eq = (cls_l.unsqueeze(1) ==
-          torch.arange(C, device=cls_l.device).view(1, C, 1)).long()
-    pb = Mp.gather(3, eq.unsqueeze(-1)).squeeze(-1)          # (B, C, Hl)
+                  ES.permute(1, 2, 0))                       # (C, 2Hl, B)
+    Mv = M.view(C, Hl, 2, B)
+    pb = Mv[:, :, 0, :].permute(2, 0, 1).contiguous()        # (B, C, Hl)
+    ar_h = torch.arange(Hl, device=cls_l.device)
+    ar_b = torch.arange(B, device=cls_l.device)
+    hit_vals = Mv[cls_l, ar_h.view(1, Hl), 1, ar_b.view(B, 1)]  # (B, Hl)
+    pb[ar_b.view(B, 1), cls_l, ar_h.view(1, Hl)] = hit_vals
     tot = pb.sum(-1)                                         # (B, C)
     comm.all_reduce_(tot)
     pb = pb / tot.clamp_min(EPS_PROB).unsqueeze(-1)

@@ -105,6 +105,11 @@ class CODA(ModelSelector):
         # host-side copy for the per-step candidate filter (indexing a GPU
         # tensor point-by-point would be one device sync per point)
         self._disagreement_host = self._disagreement.cpu().tolist()
+        # persistently maintained unlabeled-and-disagreeing candidate list
+        # (the mask is static; add_label removes one entry) - avoids an
+        # O(N) Python filter pass every step
+        self._active_candidates = [
+            i for i in range(self.N) if self._disagreement_host[i]]
 
         self.labeled_idxs, self.labels = [], []
         self.unlabeled_idxs = list(range(self.N))
@@ -130,8 +135,11 @@ class CODA(ModelSelector):
     def _prefilter(self, idxs):
         """Drop points where every model agrees (no information), optionally
         subsample to prefilter_n (reference: coda/coda.py:215-224)."""
-        mask = self._disagreement_host
-        idxs = [i for i in idxs if mask[i]]
+        if idxs is self.unlabeled_idxs:
+            idxs = self._active_candidates  # incrementally maintained
+        else:
+            mask = self._disagreement_host
+            idxs = [i for i in idxs if mask[i]]
         if self.prefilter_n and len(idxs) > self.prefilter_n:
             idxs = random.sample(idxs, self.prefilter_n)
             self.stochastic = True
@@ -284,6 +292,21 @@ class CODA(ModelSelector):
         self.labels.append(int(true_class))
         self.q_vals.append(selection_prob)
         self.unlabeled_idxs.remove(idx)
+        try:
+            self._active_candidates.remove(idx)
+        except ValueError:
+            pass  # labeled point was not a disagreeing candidate
+
+    def skip(self, idx):
+        """Drop a point without labeling it (the demo's "I don't know"
+        path, reference demo/app.py:188-189)."""
+        idx = int(idx)
+        if idx in self.unlabeled_idxs:
+            self.unlabeled_idxs.remove(idx)
+        try:
+            self._active_candidates.remove(idx)
+        except ValueError:
+            pass
 
     # ------------------------------------------------------------------
     def get_pbest(self):

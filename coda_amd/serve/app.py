@@ -83,11 +83,14 @@ class SelectorSession:
     def skip(self, index: int):
         """'I don't know': drop the point without labeling it."""
         with self._lock:
-            unl = getattr(self.selector, "unlabeled_idxs", None)
-            if unl is None:
-                unl = self.selector.d_u_idxs
-            if int(index) in unl:
-                unl.remove(int(index))
+            if hasattr(self.selector, "skip"):
+                self.selector.skip(int(index))
+            else:
+                unl = getattr(self.selector, "unlabeled_idxs", None)
+                if unl is None:
+                    unl = self.selector.d_u_idxs
+                if int(index) in unl:
+                    unl.remove(int(index))
             self._pending = None
             return self._state_locked()
 

@@ -1,0 +1,61 @@
+"""Sub-step timing of the table-factored EIG path (GPU box)."""
+import os, sys, time
+sys.path.insert(0, ".")
+import torch
+from coda_amd.ops import table as T
+from coda_amd import ops
+
+dev = torch.device("cuda")
+H, C, B, P = 128, 1000, 256, 256
+g = torch.Generator().manual_seed(0)
+a0 = (torch.rand(H, C, generator=g) * 20 + 1).to(dev)
+b0 = (torch.rand(H, C, generator=g) * 20 + 1).to(dev)
+cls = torch.randint(0, C, (B, H), generator=g).to(dev)
+pi = torch.softmax(torch.rand(C, generator=g), 0).to(dev)
+pixi = torch.softmax(torch.rand(B, C, generator=g), -1).to(dev)
+pb0 = ops.pbest_from_beta(a0.t().contiguous(), b0.t().contiguous())
+m0, H0 = ops.mixture_entropy(pb0, pi)
+
+def timed(fn, reps=5):
+    fn(); torch.cuda.synchronize()
+    t = time.perf_counter()
+    for _ in range(reps): fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t) / reps * 1000
+
+print(f"precompute        {timed(lambda: T.table_precompute(a0, b0), 3):7.2f} ms")
+tables = T.table_precompute(a0, b0)
+print(f"update_rows(1)    {timed(lambda: T.table_update_rows(tables, a0, b0, [3])):7.2f} ms")
+
+EG, delta, s_base, w = tables
+clsl = cls.long()
+flat = delta.permute(1, 0, 2).reshape(H * C, P)
+idx = (torch.arange(H, device=dev) * C).unsqueeze(0) + clsl
+def f_sel(): return flat[idx]
+sel = f_sel()
+def f_slog():
+    slog = s_base.unsqueeze(0).repeat(B, 1, 1)
+    slog.scatter_add_(1, clsl.unsqueeze(-1).expand(B, H, P), sel)
+    return slog
+slog = f_slog()
+def f_es(): return torch.exp2(slog) * w
+ES = f_es()
+def f_perm(): return ES.permute(1, 2, 0).contiguous()
+ESm = f_perm()
+EGm = EG.reshape(C, H * 2, P)
+def f_bmm(): return torch.bmm(EGm, ESm)
+M = f_bmm()
+def f_gather():
+    Mp = M.view(C, H, 2, B).permute(3, 0, 1, 2)
+    eq = (clsl.unsqueeze(1) == torch.arange(C, device=dev).view(1, C, 1)).long()
+    pb = Mp.gather(3, eq.unsqueeze(-1)).squeeze(-1)
+    return pb / pb.sum(-1, keepdim=True).clamp_min(1e-30)
+pb = f_gather()
+def f_eig(): return ops.reference.eig_assemble(pb, pb0, pi, pixi, m0, H0)
+
+for name, fn in [("sel gather", f_sel), ("slog build", f_slog),
+                 ("exp2*w", f_es), ("permute copy", f_perm),
+                 ("bmm", f_bmm), ("M gather+norm", f_gather),
+                 ("eig_assemble", f_eig)]:
+    print(f"{name:17s} {timed(fn):7.2f} ms")
+print(f"whole chunk       {timed(lambda: T.eig_chunk_table(tables, cls, pb0, pi, pixi, m0, H0)):7.2f} ms")
