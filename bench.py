@@ -40,6 +40,9 @@ N_POINTS = int(os.environ.get("CODA_BENCH_N", 50_000))
 C_CLASSES = int(os.environ.get("CODA_BENCH_C", 1000))
 PREFILTER_N = int(os.environ.get("CODA_BENCH_PREFILTER", 256))
 CHUNK = int(os.environ.get("CODA_BENCH_CHUNK", 256))
+# storage dtype for the prediction pool (fp32 | bf16 | fp8); compute is
+# always fp32 (coda_amd/datasets.py STORAGE_DTYPES)
+STORAGE = os.environ.get("CODA_BENCH_STORAGE", "fp32")
 
 
 def synth_preds(model_idxs, N, C, device, seed_base=1234):
@@ -91,6 +94,9 @@ def main():
         if shard else list(range(H_TOTAL))
 
     preds, labels = synth_preds(model_idxs, N_POINTS, C_CLASSES, device)
+    if STORAGE != "fp32":
+        from coda_amd.datasets import STORAGE_DTYPES
+        preds = preds.to(STORAGE_DTYPES[STORAGE])
     ds = Dataset.from_tensors(preds, labels, device, shard=None)
     ds.total_models = H_TOTAL
     ds.shard = shard
@@ -142,7 +148,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "fp32",
+            "dtype": "fp32" if STORAGE == "fp32" else f"fp32-compute/{STORAGE}-storage",
             "data": "synthetic",
             "config": {
                 "model": "coda-eig",
