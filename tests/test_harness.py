@@ -77,48 +77,57 @@ def test_all_methods_run_one_iter(task_dir, method):
                         "--no-mlflow", "--chunk-size", "64"])
 
 
-def test_checkpoint_resume(task_dir):
-    """Interrupted seed resumes mid-run and finishes with the same
-    trajectory as an uninterrupted run (RNG state is checkpointed too)."""
-    import torch
+def test_checkpoint_resume(task_dir, monkeypatch):
+    """A seed killed mid-run resumes from its checkpoint and produces the
+    SAME metrics as an uninterrupted run (RNG state checkpointed too)."""
     import sqlite3 as sq
+    import main as harness_mod
 
-    def regrets(d):
-        conn = sq.connect(str(d / "coda.sqlite"))
+    def regrets(exp):
+        conn = sq.connect(str(task_dir / "coda.sqlite"))
         rows = conn.execute(
-            "SELECT step, value FROM metrics WHERE key='regret' "
-            "ORDER BY step").fetchall()
+            "SELECT m.step, m.value FROM metrics m "
+            "JOIN runs r ON m.run_uuid=r.run_uuid "
+            "JOIN experiments e ON r.experiment_id=e.experiment_id "
+            "WHERE e.name=? AND m.key='regret' ORDER BY m.step",
+            (exp,)).fetchall()
         conn.close()
         return rows
 
     # uninterrupted reference run
     _run_cli(task_dir, ["--method", "iid", "--iters", "6", "--seeds", "1",
                         "--experiment-name", "full"])
-    want = regrets(task_dir)
-
-    # interrupted run: monkey-kill after step 3 by running iters=3 with
-    # checkpointing, then continuing to 6 from the checkpoint
-    _run_cli(task_dir, ["--method", "iid", "--iters", "3", "--seeds", "1",
-                        "--experiment-name", "ck", "--checkpoint-every", "1",
-                        "--checkpoint-dir", str(task_dir / "ck")])
-    # iters=3 finished -> checkpoint file removed; recreate the interrupted
-    # state by running again with iters=3 but keeping the checkpoint
-    # (simulate a crash: copy the saved state before completion)
-    # Simpler: run a fresh experiment with iters=6 + checkpointing, then
-    # verify a resumed second invocation skips (already finished).
-    _run_cli(task_dir, ["--method", "iid", "--iters", "6", "--seeds", "1",
-                        "--experiment-name", "ck6", "--checkpoint-every",
-                        "2", "--checkpoint-dir", str(task_dir / "ck6"),
-                        "--force-rerun"])
-    conn = sq.connect(str(task_dir / "coda.sqlite"))
-    n = conn.execute(
-        "SELECT COUNT(*) FROM metrics m JOIN runs r ON m.run_uuid=r.run_uuid "
-        "JOIN experiments e ON r.experiment_id=e.experiment_id "
-        "WHERE e.name='ck6' AND m.key='regret'").fetchone()[0]
-    conn.close()
-    assert n == 6
-    assert not list((task_dir / "ck6").glob("*.pt")), "checkpoint not cleaned"
+    want = regrets("full")
     assert len(want) == 6
+
+    # crash after step 3: patch the oracle to explode on the 4th call
+    calls = {"n": 0}
+    orig = harness_mod.Oracle.__call__
+
+    def exploding(self, idx):
+        calls["n"] += 1
+        if calls["n"] > 3:
+            raise RuntimeError("simulated crash")
+        return orig(self, idx)
+
+    monkeypatch.setattr(harness_mod.Oracle, "__call__", exploding)
+    try:
+        _run_cli(task_dir, ["--method", "iid", "--iters", "6", "--seeds",
+                            "1", "--experiment-name", "ck",
+                            "--checkpoint-every", "1",
+                            "--checkpoint-dir", str(task_dir / "ck")])
+    except RuntimeError:
+        pass
+    monkeypatch.setattr(harness_mod.Oracle, "__call__", orig)
+    assert list((task_dir / "ck").glob("*.pt")), "no checkpoint left"
+
+    # resume: run id exists with FAILED status -> re-runs, resuming at 3
+    _run_cli(task_dir, ["--method", "iid", "--iters", "6", "--seeds", "1",
+                        "--experiment-name", "ck", "--checkpoint-every",
+                        "1", "--checkpoint-dir", str(task_dir / "ck")])
+    got = regrets("ck")
+    assert got == want, (got, want)
+    assert not list((task_dir / "ck").glob("*.pt")), "checkpoint not cleaned"
 
 
 def test_checkpoint_state_roundtrip(task_dir):

@@ -100,6 +100,19 @@ def test_aggregate_and_analysis(pipeline_dir):
     assert r.returncode == 0, r.stderr
     assert (d / "fig5.png").exists()
 
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "analysis", "fig4.py")],
+        cwd=str(d), env=env, capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0, r.stderr
+    assert "failure rate" in r.stdout
+
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "analysis", "fig3.py"),
+         "--data-dir", "data"],
+        cwd=str(d), env=env, capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0, r.stderr
+    assert "by class count" in r.stdout
+
 
 def test_clear_db_selected(pipeline_dir):
     d, env = pipeline_dir
