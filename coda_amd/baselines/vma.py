@@ -14,8 +14,6 @@ from __future__ import annotations
 
 import random
 
-import torch
-
 from .activetesting import ActiveTesting
 from .. import ops
 

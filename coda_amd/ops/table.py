@@ -32,7 +32,6 @@ per-candidate transcendentals at all.
 """
 from __future__ import annotations
 
-import math
 from typing import NamedTuple
 
 import torch

@@ -1,5 +1,5 @@
 """Sub-step timing of the table-factored EIG path (GPU box)."""
-import os, sys, time
+import sys, time
 sys.path.insert(0, ".")
 import torch
 from coda_amd.ops import table as T

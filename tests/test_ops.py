@@ -6,7 +6,6 @@ broadcasts for the structured ops). The HIP kernels are checked against
 these same eager ops in tests/test_gpu.py.
 """
 import numpy as np
-import pytest
 import scipy.stats
 import torch
 
