@@ -494,6 +494,91 @@ eig_phase2_kernel(const float* __restrict__ alpha_t,
     if (lane == 0) tot_out[r] = total;
 }
 
+
+// ---------------------------------------------------------------------------
+// Table-path (v2) fusion kernels. The table-factored EIG (coda_amd/ops/
+// table.py) pairs per-step curve tables with per-chunk candidate state;
+// these two kernels fuse its elementwise/scatter glue:
+//   es_build: ES[c,b,p] = w[p] * 2^( s_base[c,p]
+//                + sum_{h: cls(b,h)==c} delta[c,h,p] )
+//     - one wave per (b,c) row; the h loop is a wave-uniform compare with
+//       a rare (avg H/C) delta-row accumulate; replaces a repeat +
+//       scatter-add + exp2 + mul chain over (B,C,P).
+//   eig_assemble_k: given M[c,b,2h+v] (the GEMM output), select
+//       v = [cls(b,h)==c], normalize over h, and emit the log2-entropy
+//       H_after[b,c] of the updated P(best) mixture.
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(BLOCK)
+es_build_kernel(const float* __restrict__ s_base,   // (C, P)
+                const float* __restrict__ delta,    // (C, H, P)
+                const int* __restrict__ cls,        // (B, H)
+                const float* __restrict__ w,        // (P,)
+                float* __restrict__ es,             // (C, B, P)
+                int B, int C, int H) {
+    const int r = blockIdx.x * ROWS_PER_BLOCK + (threadIdx.x >> 6);
+    if (r >= B * C) return;
+    const int b = r / C, c = r - b * C;
+    const int lane = threadIdx.x & 63;
+    const int p0 = lane * PTS_PER_LANE;
+
+    float acc[PTS_PER_LANE];
+    const float4 sb = *reinterpret_cast<const float4*>(
+        s_base + (size_t)c * P_POINTS + p0);
+    acc[0] = sb.x; acc[1] = sb.y; acc[2] = sb.z; acc[3] = sb.w;
+    const size_t dbase = (size_t)c * H * P_POINTS + p0;
+    for (int h = 0; h < H; ++h) {
+        if (cls[(size_t)b * H + h] == c) {
+            const float4 d = *reinterpret_cast<const float4*>(
+                delta + dbase + (size_t)h * P_POINTS);
+            acc[0] += d.x; acc[1] += d.y; acc[2] += d.z; acc[3] += d.w;
+        }
+    }
+    const float4 wv = *reinterpret_cast<const float4*>(w + p0);
+    float4 out;
+    out.x = exp2f(acc[0]) * wv.x;
+    out.y = exp2f(acc[1]) * wv.y;
+    out.z = exp2f(acc[2]) * wv.z;
+    out.w = exp2f(acc[3]) * wv.w;
+    *reinterpret_cast<float4*>(
+        es + ((size_t)c * B + b) * P_POINTS + p0) = out;
+}
+
+__global__ void __launch_bounds__(BLOCK)
+eig_assemble_kernel(const float* __restrict__ m,            // (C, B, 2H)
+                    const int* __restrict__ cls,            // (B, H)
+                    const float* __restrict__ pi_hat,       // (C,)
+                    const float* __restrict__ pbest_before, // (C, H)
+                    const float* __restrict__ mixture0,     // (H,)
+                    float* __restrict__ h_after,            // (B, C)
+                    int B, int C, int H) {
+    const int r = blockIdx.x * ROWS_PER_BLOCK + (threadIdx.x >> 6);
+    if (r >= B * C) return;
+    const int b = r / C, c = r - b * C;
+    const int lane = threadIdx.x & 63;
+    const float* row = m + ((size_t)c * B + b) * (2 * H);
+
+    float total = 0.f;
+    for (int h = lane; h < H; h += 64) {
+        const int v = (cls[(size_t)b * H + h] == c) ? 1 : 0;
+        total += row[2 * h + v];
+    }
+    total = wave_reduce_sum(total);
+    const float inv = 1.0f / fmaxf(total, kEps);
+
+    const float pi_c = pi_hat[c];
+    float ent = 0.f;
+    for (int h = lane; h < H; h += 64) {
+        const int v = (cls[(size_t)b * H + h] == c) ? 1 : 0;
+        const float pb = row[2 * h + v] * inv;
+        float mm = mixture0[h]
+                 + pi_c * (pb - pbest_before[(size_t)c * H + h]);
+        mm = fmaxf(mm, 1e-12f);
+        ent += -mm * __log2f(mm);
+    }
+    ent = wave_reduce_sum(ent);
+    if (lane == 0) h_after[r] = ent;
+}
+
 }  // namespace
 
 // ---------------------------------------------------------------------------
@@ -664,6 +749,55 @@ std::vector<torch::Tensor> eig_phase2(torch::Tensor alpha_cc,
     return {pb, tot};
 }
 
+
+// ---- Table-path (v2) fusion host bindings ----
+
+torch::Tensor es_build(torch::Tensor s_base, torch::Tensor delta,
+                       torch::Tensor cls, torch::Tensor w) {
+    check_f32_cuda(s_base, "s_base");
+    check_f32_cuda(delta, "delta");
+    check_f32_cuda(w, "w");
+    TORCH_CHECK(cls.scalar_type() == torch::kInt32, "cls must be int32");
+    TORCH_CHECK(s_base.size(1) == P_POINTS, "P must be 256");
+    const int C = s_base.size(0), H = delta.size(1);
+    const int B = cls.size(0);
+    auto es = torch::empty({C, B, P_POINTS}, s_base.options());
+    const int R = B * C;
+    const int blocks = (R + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
+    auto stream = c10::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(es_build_kernel, dim3(blocks), dim3(BLOCK), 0,
+                       stream.stream(), s_base.data_ptr<float>(),
+                       delta.data_ptr<float>(), cls.data_ptr<int>(),
+                       w.data_ptr<float>(), es.data_ptr<float>(), B, C, H);
+    C10_HIP_CHECK(hipGetLastError());
+    return es;
+}
+
+torch::Tensor eig_assemble_k(torch::Tensor m, torch::Tensor cls,
+                             torch::Tensor pi_hat,
+                             torch::Tensor pbest_before,
+                             torch::Tensor mixture0) {
+    check_f32_cuda(m, "m");
+    check_f32_cuda(pi_hat, "pi_hat");
+    check_f32_cuda(pbest_before, "pbest_before");
+    check_f32_cuda(mixture0, "mixture0");
+    TORCH_CHECK(cls.scalar_type() == torch::kInt32, "cls must be int32");
+    const int C = m.size(0), B = m.size(1);
+    const int H = m.size(2) / 2;
+    auto h_after = torch::empty({B, C}, m.options());
+    const int R = B * C;
+    const int blocks = (R + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
+    auto stream = c10::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(eig_assemble_kernel, dim3(blocks), dim3(BLOCK), 0,
+                       stream.stream(), m.data_ptr<float>(),
+                       cls.data_ptr<int>(), pi_hat.data_ptr<float>(),
+                       pbest_before.data_ptr<float>(),
+                       mixture0.data_ptr<float>(),
+                       h_after.data_ptr<float>(), B, C, H);
+    C10_HIP_CHECK(hipGetLastError());
+    return h_after;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "coda_amd fused gfx950 kernels";
     m.def("pbest_from_beta", &pbest_from_beta,
@@ -678,4 +812,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "Sharded hypothetical pass A: slog2 partials (B*C, P)");
     m.def("eig_phase2", &eig_phase2,
           "Sharded hypothetical pass B: unnorm masses + totals");
+    m.def("es_build", &es_build,
+          "v2: fused slog scatter + exp2 + trapz weights -> ES (C,B,P)");
+    m.def("eig_assemble_k", &eig_assemble_k,
+          "v2: v-select + normalize + log2-entropy -> H_after (B,C)");
 }

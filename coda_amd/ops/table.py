@@ -124,7 +124,22 @@ def eig_chunk_table(tables: EigTables, chunk_classes: torch.Tensor,
                     pbest_before: torch.Tensor, pi_hat: torch.Tensor,
                     pi_hat_xi_chunk: torch.Tensor, mixture0: torch.Tensor,
                     H_before: torch.Tensor) -> torch.Tensor:
-    """(B,) EIG for a candidate chunk through the v2 tables."""
+    """(B,) EIG for a candidate chunk through the v2 tables.
+
+    GPU path: the es_build / eig_assemble_k fusion kernels around the
+    batched GEMM (see pbest.hip); CPU path: the torch composition above.
+    """
+    EG, delta, s_base, w = tables
+    C, H, _, P = EG.shape
+    import coda_amd.ops as O
+    if EG.is_cuda and P == PBEST_NUM_POINTS and O._want_hip(EG):
+        cls32 = chunk_classes.to(torch.int32).contiguous()
+        ES = O._ext.es_build(s_base, delta, cls32, w)        # (C, B, P)
+        M = torch.bmm(ES, EG.reshape(C, 2 * H, P).transpose(1, 2))
+        h_after = O._ext.eig_assemble_k(M, cls32, pi_hat.contiguous(),
+                                        pbest_before.contiguous(),
+                                        mixture0.contiguous())  # (B, C)
+        return H_before - (pi_hat_xi_chunk * h_after).sum(-1)
     pb = pbest_hyp_table(tables, chunk_classes)
     return eig_assemble(pb, pbest_before, pi_hat, pi_hat_xi_chunk,
                         mixture0, H_before)
