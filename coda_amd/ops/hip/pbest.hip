@@ -579,6 +579,95 @@ eig_assemble_kernel(const float* __restrict__ m,            // (C, B, 2H)
     if (lane == 0) h_after[r] = ent;
 }
 
+
+// Sharded-v2 variants: the cross-rank coupling arrives as the gathered
+// selected-delta curves sel_all (B, Hg, P) + gathered classes (B, Hg);
+// normalization needs a GLOBAL total, so the assemble step splits into a
+// totals kernel (partial sums, all-reduced by the host) and an entropy
+// kernel consuming the reduced totals.
+__global__ void __launch_bounds__(BLOCK)
+es_build_gathered_kernel(const float* __restrict__ s_base_all,  // (C, P)
+                         const float* __restrict__ sel_all,  // (B, Hg, P)
+                         const int* __restrict__ cls_all,    // (B, Hg)
+                         const float* __restrict__ w,        // (P,)
+                         float* __restrict__ es,             // (C, B, P)
+                         int B, int C, int Hg) {
+    const int r = blockIdx.x * ROWS_PER_BLOCK + (threadIdx.x >> 6);
+    if (r >= B * C) return;
+    const int b = r / C, c = r - b * C;
+    const int lane = threadIdx.x & 63;
+    const int p0 = lane * PTS_PER_LANE;
+
+    float acc[PTS_PER_LANE];
+    const float4 sb = *reinterpret_cast<const float4*>(
+        s_base_all + (size_t)c * P_POINTS + p0);
+    acc[0] = sb.x; acc[1] = sb.y; acc[2] = sb.z; acc[3] = sb.w;
+    const size_t sbase = (size_t)b * Hg * P_POINTS + p0;
+    for (int h = 0; h < Hg; ++h) {
+        if (cls_all[(size_t)b * Hg + h] == c) {
+            const float4 d = *reinterpret_cast<const float4*>(
+                sel_all + sbase + (size_t)h * P_POINTS);
+            acc[0] += d.x; acc[1] += d.y; acc[2] += d.z; acc[3] += d.w;
+        }
+    }
+    const float4 wv = *reinterpret_cast<const float4*>(w + p0);
+    float4 out;
+    out.x = exp2f(acc[0]) * wv.x;
+    out.y = exp2f(acc[1]) * wv.y;
+    out.z = exp2f(acc[2]) * wv.z;
+    out.w = exp2f(acc[3]) * wv.w;
+    *reinterpret_cast<float4*>(
+        es + ((size_t)c * B + b) * P_POINTS + p0) = out;
+}
+
+__global__ void __launch_bounds__(BLOCK)
+eig_totals_kernel(const float* __restrict__ m,    // (C, B, 2H)
+                  const int* __restrict__ cls,    // (B, H) local
+                  float* __restrict__ totals,     // (B, C) partial
+                  int B, int C, int H) {
+    const int r = blockIdx.x * ROWS_PER_BLOCK + (threadIdx.x >> 6);
+    if (r >= B * C) return;
+    const int b = r / C, c = r - b * C;
+    const int lane = threadIdx.x & 63;
+    const float* row = m + ((size_t)c * B + b) * (2 * H);
+    float total = 0.f;
+    for (int h = lane; h < H; h += 64) {
+        const int v = (cls[(size_t)b * H + h] == c) ? 1 : 0;
+        total += row[2 * h + v];
+    }
+    total = wave_reduce_sum(total);
+    if (lane == 0) totals[r] = total;
+}
+
+__global__ void __launch_bounds__(BLOCK)
+eig_entropy_kernel(const float* __restrict__ m,            // (C, B, 2H)
+                   const int* __restrict__ cls,            // (B, H) local
+                   const float* __restrict__ totals,       // (B, C) GLOBAL
+                   const float* __restrict__ pi_hat,       // (C,)
+                   const float* __restrict__ pbest_before, // (C, H) local
+                   const float* __restrict__ mixture0,     // (H,) local
+                   float* __restrict__ h_after,            // (B, C) partial
+                   int B, int C, int H) {
+    const int r = blockIdx.x * ROWS_PER_BLOCK + (threadIdx.x >> 6);
+    if (r >= B * C) return;
+    const int b = r / C, c = r - b * C;
+    const int lane = threadIdx.x & 63;
+    const float* row = m + ((size_t)c * B + b) * (2 * H);
+    const float inv = 1.0f / fmaxf(totals[r], kEps);
+    const float pi_c = pi_hat[c];
+    float ent = 0.f;
+    for (int h = lane; h < H; h += 64) {
+        const int v = (cls[(size_t)b * H + h] == c) ? 1 : 0;
+        const float pb = row[2 * h + v] * inv;
+        float mm = mixture0[h]
+                 + pi_c * (pb - pbest_before[(size_t)c * H + h]);
+        mm = fmaxf(mm, 1e-12f);
+        ent += -mm * __log2f(mm);
+    }
+    ent = wave_reduce_sum(ent);
+    if (lane == 0) h_after[r] = ent;
+}
+
 }  // namespace
 
 // ---------------------------------------------------------------------------
@@ -798,6 +887,66 @@ torch::Tensor eig_assemble_k(torch::Tensor m, torch::Tensor cls,
     return h_after;
 }
 
+
+torch::Tensor es_build_gathered(torch::Tensor s_base_all,
+                                torch::Tensor sel_all,
+                                torch::Tensor cls_all, torch::Tensor w) {
+    check_f32_cuda(s_base_all, "s_base_all");
+    check_f32_cuda(sel_all, "sel_all");
+    check_f32_cuda(w, "w");
+    TORCH_CHECK(cls_all.scalar_type() == torch::kInt32,
+                "cls_all must be int32");
+    const int C = s_base_all.size(0);
+    const int B = cls_all.size(0), Hg = cls_all.size(1);
+    auto es = torch::empty({C, B, P_POINTS}, s_base_all.options());
+    const int R = B * C;
+    const int blocks = (R + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
+    auto stream = c10::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(es_build_gathered_kernel, dim3(blocks), dim3(BLOCK),
+                       0, stream.stream(), s_base_all.data_ptr<float>(),
+                       sel_all.data_ptr<float>(), cls_all.data_ptr<int>(),
+                       w.data_ptr<float>(), es.data_ptr<float>(), B, C, Hg);
+    C10_HIP_CHECK(hipGetLastError());
+    return es;
+}
+
+torch::Tensor eig_totals(torch::Tensor m, torch::Tensor cls) {
+    check_f32_cuda(m, "m");
+    const int C = m.size(0), B = m.size(1), H = m.size(2) / 2;
+    auto totals = torch::empty({B, C}, m.options());
+    const int R = B * C;
+    const int blocks = (R + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
+    auto stream = c10::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(eig_totals_kernel, dim3(blocks), dim3(BLOCK), 0,
+                       stream.stream(), m.data_ptr<float>(),
+                       cls.data_ptr<int>(), totals.data_ptr<float>(),
+                       B, C, H);
+    C10_HIP_CHECK(hipGetLastError());
+    return totals;
+}
+
+torch::Tensor eig_entropy(torch::Tensor m, torch::Tensor cls,
+                          torch::Tensor totals, torch::Tensor pi_hat,
+                          torch::Tensor pbest_before,
+                          torch::Tensor mixture0) {
+    check_f32_cuda(m, "m");
+    check_f32_cuda(totals, "totals");
+    const int C = m.size(0), B = m.size(1), H = m.size(2) / 2;
+    auto h_after = torch::empty({B, C}, m.options());
+    const int R = B * C;
+    const int blocks = (R + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
+    auto stream = c10::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(eig_entropy_kernel, dim3(blocks), dim3(BLOCK), 0,
+                       stream.stream(), m.data_ptr<float>(),
+                       cls.data_ptr<int>(), totals.data_ptr<float>(),
+                       pi_hat.data_ptr<float>(),
+                       pbest_before.data_ptr<float>(),
+                       mixture0.data_ptr<float>(),
+                       h_after.data_ptr<float>(), B, C, H);
+    C10_HIP_CHECK(hipGetLastError());
+    return h_after;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "coda_amd fused gfx950 kernels";
     m.def("pbest_from_beta", &pbest_from_beta,
@@ -816,4 +965,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "v2: fused slog scatter + exp2 + trapz weights -> ES (C,B,P)");
     m.def("eig_assemble_k", &eig_assemble_k,
           "v2: v-select + normalize + log2-entropy -> H_after (B,C)");
+    m.def("es_build_gathered", &es_build_gathered,
+          "sharded v2: ES from gathered selected-delta curves");
+    m.def("eig_totals", &eig_totals,
+          "sharded v2: local normalizer partials (B,C)");
+    m.def("eig_entropy", &eig_entropy,
+          "sharded v2: entropy partials from globally-reduced totals");
 }

@@ -196,10 +196,26 @@ def eig_chunk_table_sharded(tables: EigTables, s_base_all: torch.Tensor,
 
     flat = delta.permute(1, 0, 2).reshape(Hl * C, P)
     sel = flat[(torch.arange(Hl, device=cls_l.device) * C).unsqueeze(0)
-               + cls_l]                                      # (B, Hl, P)
-    sel_all = comm.all_gather_cat(sel, dim=1)                # (B, H, P)
+               + cls_l].contiguous()                         # (B, Hl, P)
+    sel_all = comm.all_gather_cat(sel, dim=1).contiguous()   # (B, H, P)
     cls_all = comm.all_gather_cat(cls_l, dim=1)              # (B, H)
     Hg = cls_all.shape[1]
+
+    import coda_amd.ops as O
+    if EG.is_cuda and P == PBEST_NUM_POINTS and O._want_hip(EG):
+        cls_all32 = cls_all.to(torch.int32).contiguous()
+        cls_l32 = cls_l.to(torch.int32).contiguous()
+        ES = O._ext.es_build_gathered(s_base_all.contiguous(), sel_all,
+                                      cls_all32, w)          # (C, B, P)
+        M = torch.bmm(ES, EG.reshape(C, 2 * Hl, P).transpose(1, 2))
+        tot = O._ext.eig_totals(M, cls_l32)                  # (B, C) partial
+        comm.all_reduce_(tot)
+        h_after = O._ext.eig_entropy(M, cls_l32, tot.contiguous(),
+                                     pi_hat.contiguous(),
+                                     pbest_before_local.contiguous(),
+                                     mixture0_local.contiguous())
+        comm.all_reduce_(h_after)
+        return H_before - (pi_hat_xi_chunk * h_after).sum(-1)
 
     slog = s_base_all.unsqueeze(0).repeat(B, 1, 1)           # (B, C, P)
     slog.scatter_add_(1, cls_all.unsqueeze(-1).expand(B, Hg, P), sel_all)

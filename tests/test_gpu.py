@@ -240,3 +240,26 @@ class TestTableFusionKernels:
                                 pi.to(dev), pi_xi.to(dev), m0.to(dev),
                                 H0.to(dev)).cpu()
         torch.testing.assert_close(got, want, rtol=5e-3, atol=1e-5)
+
+    def test_sharded_table_gpu_path_single_rank(self, dev):
+        """es_build_gathered + eig_totals/eig_entropy with a no-op Comm
+        must match the single-device fused path."""
+        from coda_amd import ops
+        from coda_amd.ops import table as T
+        from coda_amd.parallel import Comm
+        g = torch.Generator().manual_seed(42)
+        H, C, B = 16, 10, 24
+        a0 = (torch.rand(H, C, generator=g) * 20 + 1).to(dev)
+        b0 = (torch.rand(H, C, generator=g) * 20 + 1).to(dev)
+        cls = torch.randint(0, C, (B, H), generator=g).to(dev)
+        pi = torch.softmax(torch.rand(C, generator=g), 0).to(dev)
+        pi_xi = torch.softmax(torch.rand(B, C, generator=g), -1).to(dev)
+        pb0 = ops.pbest_from_beta(a0.t().contiguous(), b0.t().contiguous())
+        m0, H0 = ops.mixture_entropy(pb0, pi)
+        tables = T.table_precompute(a0, b0)
+        want = T.eig_chunk_table(tables, cls, pb0, pi, pi_xi, m0, H0)
+        comm = Comm()
+        sba = T.s_base_global(tables, comm)
+        got = T.eig_chunk_table_sharded(tables, sba, cls, pb0, pi, pi_xi,
+                                        m0, H0, comm)
+        torch.testing.assert_close(got, want, rtol=1e-4, atol=1e-6)
