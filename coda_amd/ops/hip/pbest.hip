@@ -511,7 +511,8 @@ eig_phase2_kernel(const float* __restrict__ alpha_t,
 __global__ void __launch_bounds__(BLOCK)
 es_build_kernel(const float* __restrict__ s_base,   // (C, P)
                 const float* __restrict__ delta,    // (C, H, P)
-                const int* __restrict__ cls,        // (B, H)
+                const int* __restrict__ hvals,      // (B, H) h sorted by class
+                const int* __restrict__ offsets,    // (B, C+1) CSR
                 const float* __restrict__ w,        // (P,)
                 float* __restrict__ es,             // (C, B, P)
                 int B, int C, int H) {
@@ -526,12 +527,15 @@ es_build_kernel(const float* __restrict__ s_base,   // (C, P)
         s_base + (size_t)c * P_POINTS + p0);
     acc[0] = sb.x; acc[1] = sb.y; acc[2] = sb.z; acc[3] = sb.w;
     const size_t dbase = (size_t)c * H * P_POINTS + p0;
-    for (int h = 0; h < H; ++h) {
-        if (cls[(size_t)b * H + h] == c) {
-            const float4 d = *reinterpret_cast<const float4*>(
-                delta + dbase + (size_t)h * P_POINTS);
-            acc[0] += d.x; acc[1] += d.y; acc[2] += d.z; acc[3] += d.w;
-        }
+    // CSR over "models whose argmax class on candidate b is c": avg H/C
+    // iterations instead of an H-long wave-uniform load chain
+    const int k0 = offsets[(size_t)b * (C + 1) + c];
+    const int k1 = offsets[(size_t)b * (C + 1) + c + 1];
+    for (int k = k0; k < k1; ++k) {
+        const int h = hvals[(size_t)b * H + k];
+        const float4 d = *reinterpret_cast<const float4*>(
+            delta + dbase + (size_t)h * P_POINTS);
+        acc[0] += d.x; acc[1] += d.y; acc[2] += d.z; acc[3] += d.w;
     }
     const float4 wv = *reinterpret_cast<const float4*>(w + p0);
     float4 out;
@@ -588,7 +592,8 @@ eig_assemble_kernel(const float* __restrict__ m,            // (C, B, 2H)
 __global__ void __launch_bounds__(BLOCK)
 es_build_gathered_kernel(const float* __restrict__ s_base_all,  // (C, P)
                          const float* __restrict__ sel_all,  // (B, Hg, P)
-                         const int* __restrict__ cls_all,    // (B, Hg)
+                         const int* __restrict__ hvals,      // (B, Hg) CSR
+                         const int* __restrict__ offsets,    // (B, C+1)
                          const float* __restrict__ w,        // (P,)
                          float* __restrict__ es,             // (C, B, P)
                          int B, int C, int Hg) {
@@ -603,12 +608,13 @@ es_build_gathered_kernel(const float* __restrict__ s_base_all,  // (C, P)
         s_base_all + (size_t)c * P_POINTS + p0);
     acc[0] = sb.x; acc[1] = sb.y; acc[2] = sb.z; acc[3] = sb.w;
     const size_t sbase = (size_t)b * Hg * P_POINTS + p0;
-    for (int h = 0; h < Hg; ++h) {
-        if (cls_all[(size_t)b * Hg + h] == c) {
-            const float4 d = *reinterpret_cast<const float4*>(
-                sel_all + sbase + (size_t)h * P_POINTS);
-            acc[0] += d.x; acc[1] += d.y; acc[2] += d.z; acc[3] += d.w;
-        }
+    const int k0 = offsets[(size_t)b * (C + 1) + c];
+    const int k1 = offsets[(size_t)b * (C + 1) + c + 1];
+    for (int k = k0; k < k1; ++k) {
+        const int h = hvals[(size_t)b * Hg + k];
+        const float4 d = *reinterpret_cast<const float4*>(
+            sel_all + sbase + (size_t)h * P_POINTS);
+        acc[0] += d.x; acc[1] += d.y; acc[2] += d.z; acc[3] += d.w;
     }
     const float4 wv = *reinterpret_cast<const float4*>(w + p0);
     float4 out;
@@ -842,21 +848,25 @@ std::vector<torch::Tensor> eig_phase2(torch::Tensor alpha_cc,
 // ---- Table-path (v2) fusion host bindings ----
 
 torch::Tensor es_build(torch::Tensor s_base, torch::Tensor delta,
-                       torch::Tensor cls, torch::Tensor w) {
+                       torch::Tensor hvals, torch::Tensor offsets,
+                       torch::Tensor w) {
     check_f32_cuda(s_base, "s_base");
     check_f32_cuda(delta, "delta");
     check_f32_cuda(w, "w");
-    TORCH_CHECK(cls.scalar_type() == torch::kInt32, "cls must be int32");
+    TORCH_CHECK(hvals.scalar_type() == torch::kInt32 &&
+                offsets.scalar_type() == torch::kInt32,
+                "hvals/offsets must be int32");
     TORCH_CHECK(s_base.size(1) == P_POINTS, "P must be 256");
     const int C = s_base.size(0), H = delta.size(1);
-    const int B = cls.size(0);
+    const int B = hvals.size(0);
     auto es = torch::empty({C, B, P_POINTS}, s_base.options());
     const int R = B * C;
     const int blocks = (R + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
     auto stream = c10::hip::getCurrentHIPStream();
     hipLaunchKernelGGL(es_build_kernel, dim3(blocks), dim3(BLOCK), 0,
                        stream.stream(), s_base.data_ptr<float>(),
-                       delta.data_ptr<float>(), cls.data_ptr<int>(),
+                       delta.data_ptr<float>(), hvals.data_ptr<int>(),
+                       offsets.data_ptr<int>(),
                        w.data_ptr<float>(), es.data_ptr<float>(), B, C, H);
     C10_HIP_CHECK(hipGetLastError());
     return es;
@@ -890,21 +900,24 @@ torch::Tensor eig_assemble_k(torch::Tensor m, torch::Tensor cls,
 
 torch::Tensor es_build_gathered(torch::Tensor s_base_all,
                                 torch::Tensor sel_all,
-                                torch::Tensor cls_all, torch::Tensor w) {
+                                torch::Tensor hvals, torch::Tensor offsets,
+                                torch::Tensor w) {
     check_f32_cuda(s_base_all, "s_base_all");
     check_f32_cuda(sel_all, "sel_all");
     check_f32_cuda(w, "w");
-    TORCH_CHECK(cls_all.scalar_type() == torch::kInt32,
-                "cls_all must be int32");
+    TORCH_CHECK(hvals.scalar_type() == torch::kInt32 &&
+                offsets.scalar_type() == torch::kInt32,
+                "hvals/offsets must be int32");
     const int C = s_base_all.size(0);
-    const int B = cls_all.size(0), Hg = cls_all.size(1);
+    const int B = hvals.size(0), Hg = hvals.size(1);
     auto es = torch::empty({C, B, P_POINTS}, s_base_all.options());
     const int R = B * C;
     const int blocks = (R + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
     auto stream = c10::hip::getCurrentHIPStream();
     hipLaunchKernelGGL(es_build_gathered_kernel, dim3(blocks), dim3(BLOCK),
                        0, stream.stream(), s_base_all.data_ptr<float>(),
-                       sel_all.data_ptr<float>(), cls_all.data_ptr<int>(),
+                       sel_all.data_ptr<float>(), hvals.data_ptr<int>(),
+                       offsets.data_ptr<int>(),
                        w.data_ptr<float>(), es.data_ptr<float>(), B, C, Hg);
     C10_HIP_CHECK(hipGetLastError());
     return es;

@@ -42,6 +42,21 @@ from .reference import (EPS_PROB, GRID_HI, GRID_LO, PBEST_NUM_POINTS,
 _LOG2E = 1.4426950408889634
 
 
+def _class_csr(cls: torch.Tensor, C: int):
+    """Sort each candidate's models by predicted class -> CSR buckets.
+
+    cls: (B, H) long. Returns (hvals (B,H) int32 - model indices sorted by
+    class; offsets (B, C+1) int32).
+    """
+    B, H = cls.shape
+    sorted_cls, order = torch.sort(cls, dim=1, stable=True)
+    counts = torch.zeros(B, C, dtype=torch.int32, device=cls.device)
+    counts.scatter_add_(1, cls, torch.ones_like(cls, dtype=torch.int32))
+    offsets = torch.zeros(B, C + 1, dtype=torch.int32, device=cls.device)
+    offsets[:, 1:] = counts.cumsum(1)
+    return order.to(torch.int32).contiguous(), offsets.contiguous()
+
+
 class EigTables(NamedTuple):
     EG: torch.Tensor       # (C, H, 2, P)
     delta: torch.Tensor    # (C, H, P)
@@ -134,7 +149,8 @@ def eig_chunk_table(tables: EigTables, chunk_classes: torch.Tensor,
     import coda_amd.ops as O
     if EG.is_cuda and P == PBEST_NUM_POINTS and O._want_hip(EG):
         cls32 = chunk_classes.to(torch.int32).contiguous()
-        ES = O._ext.es_build(s_base, delta, cls32, w)        # (C, B, P)
+        hvals, offsets = _class_csr(chunk_classes.long(), C)
+        ES = O._ext.es_build(s_base, delta, hvals, offsets, w)  # (C, B, P)
         M = torch.bmm(ES, EG.reshape(C, 2 * H, P).transpose(1, 2))
         h_after = O._ext.eig_assemble_k(M, cls32, pi_hat.contiguous(),
                                         pbest_before.contiguous(),
@@ -203,10 +219,10 @@ def eig_chunk_table_sharded(tables: EigTables, s_base_all: torch.Tensor,
 
     import coda_amd.ops as O
     if EG.is_cuda and P == PBEST_NUM_POINTS and O._want_hip(EG):
-        cls_all32 = cls_all.to(torch.int32).contiguous()
         cls_l32 = cls_l.to(torch.int32).contiguous()
+        hvals, offsets = _class_csr(cls_all, C)
         ES = O._ext.es_build_gathered(s_base_all.contiguous(), sel_all,
-                                      cls_all32, w)          # (C, B, P)
+                                      hvals, offsets, w)     # (C, B, P)
         M = torch.bmm(ES, EG.reshape(C, 2 * Hl, P).transpose(1, 2))
         tot = O._ext.eig_totals(M, cls_l32)                  # (B, C) partial
         comm.all_reduce_(tot)
