@@ -18,6 +18,7 @@ import os
 import torch
 
 from . import reference
+from ..util import DEBUG, _check
 from .reference import (  # re-export cheap ops + constants
     PBEST_NUM_POINTS, GRID_LO, GRID_HI, EPS_PROB, LOG_CLAMP,
     consensus, confusion_prior, init_dirichlets, dirichlet_to_beta,
@@ -66,10 +67,19 @@ def _want_hip(t: torch.Tensor) -> bool:
 def pbest_from_beta(alpha: torch.Tensor, beta: torch.Tensor,
                     num_points: int = PBEST_NUM_POINTS,
                     return_unnormalized: bool = False) -> torch.Tensor:
-    """P(model h is best) per row from diagonal Beta params: (R,H) -> (R,H)."""
-    if _want_hip(alpha) and not return_unnormalized:
-        return _ext.pbest_from_beta(alpha.contiguous(), beta.contiguous(),
-                                    int(num_points))
+    """P(model h is best) per row from diagonal Beta params: (R,H) -> (R,H).
+
+    The HIP kernel is compiled for the reference's P=256 grid; other grid
+    sizes run the eager formulation (on GPU too - that is a shape choice,
+    not a silent fallback).
+    """
+    if (num_points == PBEST_NUM_POINTS and not return_unnormalized
+            and _want_hip(alpha)):
+        out = _ext.pbest_from_beta(alpha.contiguous(), beta.contiguous(),
+                                   int(num_points))
+        if DEBUG:
+            _check(out, "pbest(kernel)")
+        return out
     return reference.pbest_from_beta(alpha, beta, num_points,
                                      return_unnormalized)
 
@@ -89,13 +99,16 @@ def eig_chunk(alpha_cc: torch.Tensor, beta_cc: torch.Tensor,
 
     Reference semantics: coda/coda.py:261-278 (K10 + K6-K9 + K11).
     """
-    if _want_hip(alpha_cc):
-        return _ext.eig_chunk(
+    if num_points == PBEST_NUM_POINTS and _want_hip(alpha_cc):
+        out = _ext.eig_chunk(
             alpha_cc.contiguous(), beta_cc.contiguous(),
             chunk_classes.to(torch.int32).contiguous(),
             pbest_before.contiguous(), pi_hat.contiguous(),
             pi_hat_xi_chunk.contiguous(), mixture0.contiguous(),
             float(H_before), float(update_weight), int(num_points))
+        if DEBUG:
+            _check(out, "eig(kernel)")
+        return out
     a, b = reference.hypothetical_betas(alpha_cc, beta_cc, chunk_classes,
                                         update_weight)
     B, C, H = a.shape

@@ -160,3 +160,53 @@ def test_incremental_pi_hat_matches_full():
     torch.testing.assert_close(sel._adjusted, full, rtol=1e-4, atol=1e-5)
     xi_full, pi_full = ops.pi_hat_normalize(full)
     torch.testing.assert_close(sel.pi_hat, pi_full, rtol=1e-4, atol=1e-6)
+
+
+class TestEdgeShapes:
+    """Degenerate and boundary shapes the engine must survive."""
+
+    @pytest.mark.parametrize("H,C", [(2, 2), (2, 5), (3, 2), (9, 3)])
+    def test_small_pools_and_binary_tasks(self, H, C):
+        preds, labels = make_synthetic_task(H=H, N=80, C=C, seed=40)
+        ds = Dataset.from_tensors(preds, labels, "cpu")
+        oracle = Oracle(ds, LOSS_FNS["acc"])
+        random.seed(0); torch.manual_seed(0)
+        sel = CODA(ds, chunk_size=32)
+        for _ in range(3):
+            idx, q = sel.get_next_item_to_label()
+            sel.add_label(idx, oracle(int(idx)), q)
+        p = sel.get_pbest()
+        assert torch.isfinite(p).all() and p.shape == (H,)
+
+    def test_all_models_agree_everywhere(self):
+        """Unanimous pool: the prefilter drops everything and the
+        fallback-to-all-unlabeled path (coda/coda.py:239 `or`) engages."""
+        preds = torch.zeros(4, 50, 3)
+        preds[:, :, 1] = 1.0
+        labels = torch.ones(50, dtype=torch.long)
+        ds = Dataset.from_tensors(preds, labels, "cpu")
+        oracle = Oracle(ds, LOSS_FNS["acc"])
+        random.seed(0); torch.manual_seed(0)
+        sel = CODA(ds, chunk_size=16)
+        idx, q = sel.get_next_item_to_label()
+        sel.add_label(idx, oracle(int(idx)), q)
+        assert torch.isfinite(sel.get_pbest()).all()
+
+    def test_nearly_exhausted_pool(self):
+        preds, labels = make_synthetic_task(H=4, N=12, C=3, seed=41)
+        ds = Dataset.from_tensors(preds, labels, "cpu")
+        oracle = Oracle(ds, LOSS_FNS["acc"])
+        random.seed(0); torch.manual_seed(0)
+        sel = CODA(ds, chunk_size=8)
+        for _ in range(10):
+            idx, q = sel.get_next_item_to_label()
+            sel.add_label(idx, oracle(int(idx)), q)
+        assert len(sel.unlabeled_idxs) == 2
+
+    def test_nonstandard_num_points(self):
+        """num_points != 256 runs the eager grid (GPU included)."""
+        preds, labels = make_synthetic_task(H=4, N=60, C=3, seed=42)
+        ds = Dataset.from_tensors(preds, labels, "cpu")
+        sel = CODA(ds, chunk_size=16, num_points=128, eig_impl="fused")
+        e, c = sel.eig_batched()
+        assert torch.isfinite(e).all()
