@@ -105,7 +105,19 @@ def build_selector(dataset, args, loss_fn, comm=None):
     if args.method == "vma":
         return VMA(dataset, loss_fn)
     if args.method == "model_picker":
+        # per-task epsilon: the built-in table, overridden by a tuned
+        # best_epsilons.json (scripts/modelpicker_eps_gridsearch.py)
         eps = TASK_EPS.get(args.task)
+        import json
+        for cand in ("best_epsilons.json",
+                     os.path.join(args.data_dir, "best_epsilons.json")):
+            if os.path.exists(cand):
+                try:
+                    tuned = json.load(open(cand))
+                    if args.task in tuned:
+                        eps = float(tuned[args.task])
+                except (ValueError, OSError):
+                    pass
         if eps is None:
             print(args.task, "not in TASK_EPS; using default")
             return ModelPicker(dataset)

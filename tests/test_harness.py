@@ -163,3 +163,24 @@ def test_checkpoint_state_roundtrip(task_dir):
     rnd.seed(42); torch.manual_seed(42)
     i2, q2 = sel2.get_next_item_to_label()
     assert int(i1) == int(i2) and abs(q1 - q2) < 1e-6
+
+
+def test_model_picker_reads_tuned_epsilon(task_dir, capsys):
+    """best_epsilons.json (from the grid search) overrides TASK_EPS."""
+    import json
+    import main as harness_mod
+    (task_dir / "best_epsilons.json").write_text(
+        json.dumps({"synthtask": 0.41}))
+    cwd = os.getcwd()
+    os.chdir(task_dir)
+    try:
+        args = harness_mod.parse_args(
+            ["--task", "synthtask", "--data-dir", "data",
+             "--method", "model_picker", "--device", "cpu"])
+        from coda_amd.datasets import Dataset
+        ds = Dataset(str(task_dir / "data" / "synthtask.pt"), "cpu")
+        from coda_amd.options import LOSS_FNS
+        sel = harness_mod.build_selector(ds, args, LOSS_FNS["acc"])
+        assert abs(sel.epsilon - 0.41) < 1e-9
+    finally:
+        os.chdir(cwd)
