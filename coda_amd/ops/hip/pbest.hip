@@ -149,6 +149,21 @@ struct LaneGrid {
         for (int j = 0; j < PTS_PER_LANE; ++j)
             pdf[j] = exp2f(fmaf(am1, dlx[j], fmaf(bm1, dl1mx[j], K)));
     }
+    // pdf4 variant that also returns the base-2 log-pdf values (t2).
+    __device__ __forceinline__ void pdf4_log(float a, float b, double lnB,
+                                             float (&pdf)[PTS_PER_LANE],
+                                             float (&t2)[PTS_PER_LANE]) const {
+        const float am1 = a - 1.0f;
+        const float bm1 = b - 1.0f;
+        const float K = (float)((double)am1 * lx0 + (double)bm1 * l1mx0
+                                - lnB);
+#pragma unroll
+        for (int j = 0; j < PTS_PER_LANE; ++j) {
+            t2[j] = fmaf(am1, dlx[j], fmaf(bm1, dl1mx[j], K));
+            pdf[j] = exp2f(t2[j]);
+        }
+    }
+
     // Trapezoid-cumulative cdf at this lane's 4 points from the wave's pdf.
     __device__ __forceinline__ void cdf4(const float (&pdf)[PTS_PER_LANE],
                                          int lane,
@@ -674,6 +689,46 @@ eig_entropy_kernel(const float* __restrict__ m,            // (C, B, 2H)
     if (lane == 0) h_after[r] = ent;
 }
 
+
+// Per-class table refresh (v2): recompute the H*2 hypothetical Beta
+// curves of ONE class row - what changes between steps (add_label moves
+// only Dirichlet row true_class). One wave per (model, variant) pair;
+// writes EG = 2^(log2 pdf - log2 cdf) and lc = log2 cdf; the tiny
+// delta/s_base combines stay on the host.
+__global__ void __launch_bounds__(BLOCK)
+beta_row_tables_kernel(const float* __restrict__ alpha_col,  // (H,)
+                       const float* __restrict__ beta_col,   // (H,)
+                       float* __restrict__ eg,                // (H, 2, P)
+                       float* __restrict__ lc_out,            // (H, 2, P)
+                       float update_weight, int H) {
+    const int q = blockIdx.x * ROWS_PER_BLOCK + (threadIdx.x >> 6);
+    if (q >= 2 * H) return;
+    const int h = q >> 1, v = q & 1;
+    const int lane = threadIdx.x & 63;
+    const float a = alpha_col[h] + (v ? update_weight : 0.f);
+    const float b = beta_col[h] + (v ? 0.f : update_weight);
+    const double lnB = (lgamma((double)a) + lgamma((double)b)
+                      - lgamma((double)a + (double)b)) * 1.4426950408889634;
+    LaneGrid g;
+    g.init(lane);
+    float pdf[PTS_PER_LANE], t2[PTS_PER_LANE], cdf[PTS_PER_LANE];
+    g.pdf4_log(a, b, lnB, pdf, t2);
+    g.cdf4(pdf, lane, cdf);
+    const size_t base = (size_t)q * P_POINTS + lane * PTS_PER_LANE;
+    float4 lcv, egv;
+    float lcj[PTS_PER_LANE];
+#pragma unroll
+    for (int j = 0; j < PTS_PER_LANE; ++j)
+        lcj[j] = __log2f(fmaxf(cdf[j], kEps));
+    lcv.x = lcj[0]; lcv.y = lcj[1]; lcv.z = lcj[2]; lcv.w = lcj[3];
+    egv.x = exp2f(t2[0] - lcj[0]);
+    egv.y = exp2f(t2[1] - lcj[1]);
+    egv.z = exp2f(t2[2] - lcj[2]);
+    egv.w = exp2f(t2[3] - lcj[3]);
+    *reinterpret_cast<float4*>(lc_out + base) = lcv;
+    *reinterpret_cast<float4*>(eg + base) = egv;
+}
+
 }  // namespace
 
 // ---------------------------------------------------------------------------
@@ -960,6 +1015,26 @@ torch::Tensor eig_entropy(torch::Tensor m, torch::Tensor cls,
     return h_after;
 }
 
+
+std::vector<torch::Tensor> beta_row_tables(torch::Tensor alpha_col,
+                                           torch::Tensor beta_col,
+                                           double update_weight) {
+    check_f32_cuda(alpha_col, "alpha_col");
+    check_f32_cuda(beta_col, "beta_col");
+    const int H = alpha_col.size(0);
+    auto eg = torch::empty({H, 2, P_POINTS}, alpha_col.options());
+    auto lc = torch::empty({H, 2, P_POINTS}, alpha_col.options());
+    const int R = 2 * H;
+    const int blocks = (R + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
+    auto stream = c10::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(beta_row_tables_kernel, dim3(blocks), dim3(BLOCK),
+                       0, stream.stream(), alpha_col.data_ptr<float>(),
+                       beta_col.data_ptr<float>(), eg.data_ptr<float>(),
+                       lc.data_ptr<float>(), (float)update_weight, H);
+    C10_HIP_CHECK(hipGetLastError());
+    return {eg, lc};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "coda_amd fused gfx950 kernels";
     m.def("pbest_from_beta", &pbest_from_beta,
@@ -984,4 +1059,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "sharded v2: local normalizer partials (B,C)");
     m.def("eig_entropy", &eig_entropy,
           "sharded v2: entropy partials from globally-reduced totals");
+    m.def("beta_row_tables", &beta_row_tables,
+          "v2: one class row's H*2 hypothetical curves (EG, log2 cdf)");
 }

@@ -173,6 +173,18 @@ def table_update_rows(tables: EigTables, alpha_cc: torch.Tensor,
     rows = torch.as_tensor(rows, device=alpha_cc.device, dtype=torch.long)
     if rows.numel() == 0:
         return tables
+    import coda_amd.ops as O
+    if (alpha_cc.is_cuda and tables.EG.shape[-1] == PBEST_NUM_POINTS
+            and rows.numel() <= 4 and O._want_hip(alpha_cc)):
+        # per-class refresh kernel: one wave per (model, variant) curve
+        for c in rows.tolist():
+            eg, lc = O._ext.beta_row_tables(
+                alpha_cc[:, c].contiguous(), beta_cc[:, c].contiguous(),
+                float(update_weight))
+            tables.EG[c] = eg
+            tables.delta[c] = lc[:, 1] - lc[:, 0]
+            tables.s_base[c] = lc[:, 0].sum(0)
+        return tables
     sub = table_precompute(alpha_cc[:, rows], beta_cc[:, rows],
                            update_weight, tables.EG.shape[-1])
     tables.EG[rows] = sub.EG

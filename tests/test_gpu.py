@@ -263,3 +263,20 @@ class TestTableFusionKernels:
         got = T.eig_chunk_table_sharded(tables, sba, cls, pb0, pi, pi_xi,
                                         m0, H0, comm)
         torch.testing.assert_close(got, want, rtol=1e-4, atol=1e-6)
+
+    def test_beta_row_tables_kernel_matches_precompute(self, dev):
+        from coda_amd.ops import table as T
+        g = torch.Generator().manual_seed(43)
+        H, C = 12, 6
+        a0 = (torch.rand(H, C, generator=g) * 20 + 1).to(dev)
+        b0 = (torch.rand(H, C, generator=g) * 20 + 1).to(dev)
+        tables = T.table_precompute(a0, b0)
+        a0[:, 2] += 0.01
+        T.table_update_rows(tables, a0, b0, [2])  # kernel path on GPU
+        fresh = T.table_precompute(a0, b0)
+        torch.testing.assert_close(tables.EG, fresh.EG, rtol=2e-3,
+                                   atol=1e30 * 0 + 1e-2)
+        torch.testing.assert_close(tables.delta, fresh.delta, rtol=1e-4,
+                                   atol=1e-4)
+        torch.testing.assert_close(tables.s_base, fresh.s_base,
+                                   rtol=1e-5, atol=1e-3)
