@@ -24,7 +24,7 @@ from .reference import (  # re-export cheap ops + constants
     consensus, confusion_prior, init_dirichlets, dirichlet_to_beta,
     pi_hat_partial, pi_hat_pack, pi_hat_pack_chunked,
     pi_hat_partial_packed, pi_hat_normalize, pi_hat_delta,
-    init_model_stats,
+    pbest_from_beta_hchunked, init_model_stats,
     beta_grid_pdf_cdf, hypothetical_betas,
     mixture_entropy, pred_classes, disagreement_mask,
     accuracy_losses, entropy_acquisition, vma_pairwise, lure_weights,
@@ -73,6 +73,10 @@ def pbest_from_beta(alpha: torch.Tensor, beta: torch.Tensor,
     sizes run the eager formulation (on GPU too - that is a shape choice,
     not a silent fallback).
     """
+    H = alpha.shape[-1]
+    if H > 2048:  # beyond the kernel's LDS budget: chunked eager passes
+        assert not return_unnormalized
+        return reference.pbest_from_beta_hchunked(alpha, beta, num_points)
     if (num_points == PBEST_NUM_POINTS and not return_unnormalized
             and _want_hip(alpha)):
         out = _ext.pbest_from_beta(alpha.contiguous(), beta.contiguous(),

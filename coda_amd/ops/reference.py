@@ -257,6 +257,34 @@ def pbest_from_beta(alpha: torch.Tensor, beta: torch.Tensor,
     return prob / prob.sum(-1, keepdim=True).clamp_min(EPS_PROB)
 
 
+def pbest_from_beta_hchunked(alpha: torch.Tensor, beta: torch.Tensor,
+                             num_points: int = PBEST_NUM_POINTS,
+                             chunk_h: int = 512) -> torch.Tensor:
+    """pbest_from_beta for very wide model axes: two passes chunked over
+    H, so only (R, chunk_h, P) materializes (the plain eager form would
+    need the full (R, H, P)). Used when H exceeds the HIP kernel's LDS
+    budget (~2048 models/row)."""
+    R, H = alpha.shape
+    x = _beta_grid(alpha.device, alpha.dtype, num_points)
+    slog = torch.zeros(R, num_points, dtype=alpha.dtype,
+                       device=alpha.device)
+    for h0 in range(0, H, chunk_h):
+        h1 = min(h0 + chunk_h, H)
+        _, cdf, _ = beta_grid_pdf_cdf(alpha[:, h0:h1], beta[:, h0:h1],
+                                      num_points)
+        slog += torch.log(cdf.clamp_min(EPS_PROB)).sum(dim=1)
+    prob = torch.empty(R, H, dtype=alpha.dtype, device=alpha.device)
+    for h0 in range(0, H, chunk_h):
+        h1 = min(h0 + chunk_h, H)
+        pdf, cdf, _ = beta_grid_pdf_cdf(alpha[:, h0:h1], beta[:, h0:h1],
+                                        num_points)
+        log_cdf = torch.log(cdf.clamp_min(EPS_PROB))
+        pe = torch.exp((slog.unsqueeze(1) - log_cdf)
+                       .clamp(-LOG_CLAMP, LOG_CLAMP))
+        prob[:, h0:h1] = torch.trapz(pdf * pe, x, dim=-1)
+    return prob / prob.sum(-1, keepdim=True).clamp_min(EPS_PROB)
+
+
 def hypothetical_betas(alpha_cc: torch.Tensor, beta_cc: torch.Tensor,
                        pred_classes: torch.Tensor, update_weight: float = 1.0):
     """Hypothetical Beta updates for a candidate chunk.

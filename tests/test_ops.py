@@ -337,3 +337,12 @@ class TestTableEig:
         torch.testing.assert_close(tables.EG, fresh.EG)
         torch.testing.assert_close(tables.delta, fresh.delta)
         torch.testing.assert_close(tables.s_base, fresh.s_base)
+
+
+def test_pbest_hchunked_matches_plain():
+    """Wide-H chunked passes == plain eager (used beyond the kernel's
+    LDS budget, H > 2048)."""
+    a, b = _rand_betas(rows=5, H=37, seed=51)
+    got = R.pbest_from_beta_hchunked(a, b, chunk_h=8)
+    want = R.pbest_from_beta(a, b)
+    torch.testing.assert_close(got, want, rtol=1e-4, atol=1e-6)
