@@ -260,17 +260,20 @@ class CODA(ModelSelector):
             except RuntimeError:
                 pass  # no active tracking run
 
-        # greedy with seeded random tie-breaking (coda/coda.py:306-313)
-        best = q_vals.max()
-        ties = torch.isclose(q_vals, best, rtol=1e-8)
-        n_ties = int(ties.sum())
-        if n_ties > 1:
+        # greedy with seeded random tie-breaking (coda/coda.py:306-313);
+        # max/argmax/tie-count fetched in ONE device sync
+        best_val, best_idx = q_vals.max(0)
+        n_ties = torch.isclose(q_vals, best_val, rtol=1e-8).sum()
+        bv, bi, nt = torch.stack(
+            [best_val, best_idx.to(q_vals.dtype),
+             n_ties.to(q_vals.dtype)]).cpu().tolist()
+        if nt > 1:
+            ties = torch.isclose(q_vals, best_val, rtol=1e-8)
             idx_local = random.choice(
                 torch.nonzero(ties, as_tuple=True)[0].tolist())
             self.stochastic = True
-        else:
-            idx_local = int(torch.argmax(q_vals))
-        return cand[idx_local], float(q_vals[idx_local])
+            return cand[idx_local], float(q_vals[idx_local])
+        return cand[int(bi)], bv
 
     def add_label(self, idx, true_class, selection_prob):
         """Posterior update (K13) + incremental pi_hat refresh.

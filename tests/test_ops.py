@@ -346,3 +346,11 @@ def test_pbest_hchunked_matches_plain():
     got = R.pbest_from_beta_hchunked(a, b, chunk_h=8)
     want = R.pbest_from_beta(a, b)
     torch.testing.assert_close(got, want, rtol=1e-4, atol=1e-6)
+
+
+def test_wide_h_dispatch_path():
+    """ops.pbest_from_beta routes H > 2048 through the chunked passes."""
+    a, b = _rand_betas(rows=2, H=2100, seed=52, lo=1.0, hi=20.0)
+    p = ops.pbest_from_beta(a, b)
+    assert torch.isfinite(p).all()
+    np.testing.assert_allclose(p.sum(-1).numpy(), 1.0, atol=1e-3)
