@@ -107,3 +107,42 @@ def test_delete(db):
     rid = tracking.search_runs(experiment_names=["taskD"]).run_id.values[0]
     tracking.delete_run(rid)
     assert len(tracking.search_runs(experiment_names=["taskD"])) == 0
+
+
+def test_concurrent_writers(db):
+    """8 processes logging into one DB simultaneously (the task-parallel
+    launcher's pattern on an 8-GPU node): WAL + busy timeout must keep
+    every write."""
+    import multiprocessing as mp
+    import subprocess
+    import sys
+
+    code = """
+import sys
+sys.path.insert(0, {repo!r})
+from coda_amd import tracking
+tracking.set_tracking_uri("sqlite:///" + {db!r})
+tracking.set_experiment("stress")
+wid = int(sys.argv[1])
+with tracking.start_run(run_name=f"w{{wid}}"):
+    for s in range(1, 21):
+        tracking.log_metric("m", wid + s * 0.01, step=s)
+    tracking.log_param("worker", wid)
+"""
+    import os as _os
+    repo = _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__)))
+    script = code.format(repo=repo, db=db)
+    procs = [subprocess.Popen([sys.executable, "-c", script, str(i)])
+             for i in range(8)]
+    for p in procs:
+        assert p.wait(timeout=120) == 0
+    import sqlite3
+    conn = sqlite3.connect(db)
+    n_runs = conn.execute("SELECT COUNT(*) FROM runs").fetchone()[0]
+    n_metrics = conn.execute(
+        "SELECT COUNT(*) FROM metrics WHERE key='m'").fetchone()[0]
+    statuses = [r[0] for r in conn.execute("SELECT status FROM runs")]
+    conn.close()
+    assert n_runs == 8
+    assert n_metrics == 8 * 20
+    assert all(s == "FINISHED" for s in statuses)
