@@ -23,7 +23,7 @@ from .reference import (  # re-export cheap ops + constants
     PBEST_NUM_POINTS, GRID_LO, GRID_HI, EPS_PROB, LOG_CLAMP,
     consensus, confusion_prior, init_dirichlets, dirichlet_to_beta,
     pi_hat_partial, pi_hat_pack, pi_hat_pack_chunked,
-    pi_hat_partial_packed, pi_hat_normalize, pi_hat_delta,
+    pi_hat_partial_packed, pi_hat_normalize,
     pbest_from_beta_hchunked, init_model_stats,
     beta_grid_pdf_cdf, hypothetical_betas,
     mixture_entropy, pred_classes, disagreement_mask,
@@ -86,6 +86,20 @@ def pbest_from_beta(alpha: torch.Tensor, beta: torch.Tensor,
         return out
     return reference.pbest_from_beta(alpha, beta, num_points,
                                      return_unnormalized)
+
+
+def pi_hat_delta(preds: torch.Tensor, point_classes: torch.Tensor,
+                 chunk_h: int = 32) -> torch.Tensor:
+    """sum_h preds[h, :, cls_h] -> (N,) fp32 (the rank-1 pi_hat term)."""
+    if (preds.is_cuda and preds.is_contiguous()
+            and preds.dtype in (torch.float32, torch.bfloat16)
+            and _want_hip(preds)):
+        out = _ext.pi_hat_delta(preds,
+                                point_classes.to(torch.int32).contiguous())
+        if DEBUG:
+            _check(out, "pi_hat_delta(kernel)")
+        return out
+    return reference.pi_hat_delta(preds, point_classes, chunk_h)
 
 
 def eig_chunk(alpha_cc: torch.Tensor, beta_cc: torch.Tensor,

@@ -28,6 +28,7 @@
 #include <torch/extension.h>
 #include <hip/hip_runtime.h>
 #include <c10/hip/HIPStream.h>
+#include <hip/hip_bfloat16.h>
 #include <vector>
 
 #define P_POINTS 256
@@ -729,6 +730,26 @@ beta_row_tables_kernel(const float* __restrict__ alpha_col,  // (H,)
     *reinterpret_cast<float4*>(eg + base) = egv;
 }
 
+
+// Rank-1 pi_hat increment: out[n] = sum_h preds[h, n, cls[h]] - the exact
+// per-label posterior-marginal change (only Dirichlet row true_class
+// moves; coda/coda.py:316-317). One thread per point; consecutive
+// threads read consecutive n (coalesced); the class index is
+// wave-uniform per model. fp32 and bf16 prediction storage.
+template <typename T>
+__global__ void __launch_bounds__(BLOCK)
+pi_hat_delta_kernel(const T* __restrict__ preds,  // (H, N, C)
+                    const int* __restrict__ cls,  // (H,)
+                    float* __restrict__ out,      // (N,)
+                    int H, long long N, int C) {
+    const long long n = (long long)blockIdx.x * BLOCK + threadIdx.x;
+    if (n >= N) return;
+    float acc = 0.f;
+    for (int h = 0; h < H; ++h)
+        acc += (float)preds[((long long)h * N + n) * C + cls[h]];
+    out[n] = acc;
+}
+
 }  // namespace
 
 // ---------------------------------------------------------------------------
@@ -1035,6 +1056,35 @@ std::vector<torch::Tensor> beta_row_tables(torch::Tensor alpha_col,
     return {eg, lc};
 }
 
+
+torch::Tensor pi_hat_delta(torch::Tensor preds, torch::Tensor cls) {
+    TORCH_CHECK(preds.is_cuda() && preds.is_contiguous(),
+                "preds must be contiguous on a ROCm device");
+    TORCH_CHECK(cls.scalar_type() == torch::kInt32, "cls must be int32");
+    const int H = preds.size(0), C = preds.size(2);
+    const long long N = preds.size(1);
+    auto out = torch::empty({N}, preds.options().dtype(torch::kFloat32));
+    const int blocks = (int)((N + BLOCK - 1) / BLOCK);
+    auto stream = c10::hip::getCurrentHIPStream();
+    if (preds.scalar_type() == torch::kFloat32) {
+        hipLaunchKernelGGL(pi_hat_delta_kernel<float>, dim3(blocks),
+                           dim3(BLOCK), 0, stream.stream(),
+                           preds.data_ptr<float>(), cls.data_ptr<int>(),
+                           out.data_ptr<float>(), H, N, C);
+    } else if (preds.scalar_type() == torch::kBFloat16) {
+        hipLaunchKernelGGL(pi_hat_delta_kernel<hip_bfloat16>, dim3(blocks),
+                           dim3(BLOCK), 0, stream.stream(),
+                           reinterpret_cast<const hip_bfloat16*>(
+                               preds.data_ptr()),
+                           cls.data_ptr<int>(), out.data_ptr<float>(),
+                           H, N, C);
+    } else {
+        TORCH_CHECK(false, "pi_hat_delta kernel supports fp32/bf16 only");
+    }
+    C10_HIP_CHECK(hipGetLastError());
+    return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "coda_amd fused gfx950 kernels";
     m.def("pbest_from_beta", &pbest_from_beta,
@@ -1061,4 +1111,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "sharded v2: entropy partials from globally-reduced totals");
     m.def("beta_row_tables", &beta_row_tables,
           "v2: one class row's H*2 hypothetical curves (EG, log2 cdf)");
+    m.def("pi_hat_delta", &pi_hat_delta,
+          "rank-1 pi_hat increment: sum_h preds[h, :, cls_h]");
 }

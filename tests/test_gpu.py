@@ -282,3 +282,16 @@ class TestTableFusionKernels:
                                    atol=5e-3)
         torch.testing.assert_close(tables.s_base, fresh.s_base,
                                    rtol=1e-4, atol=1e-2)
+
+    def test_pi_hat_delta_kernel(self, dev):
+        from coda_amd import ops
+        g = torch.Generator().manual_seed(44)
+        H, N, C = 9, 500, 7
+        preds = torch.softmax(torch.randn(H, N, C, generator=g), -1).to(dev)
+        cls = torch.randint(0, C, (H,), generator=g).to(dev)
+        got = ops.pi_hat_delta(preds, cls)
+        want = ops.reference.pi_hat_delta(preds.cpu(), cls.cpu()).to(dev)
+        torch.testing.assert_close(got, want, rtol=1e-5, atol=1e-6)
+        # bf16 storage
+        got16 = ops.pi_hat_delta(preds.to(torch.bfloat16).contiguous(), cls)
+        torch.testing.assert_close(got16, want, rtol=2e-2, atol=1e-2)
