@@ -97,7 +97,11 @@ def run_grid_search(task, data_dir, eps_grid, realisations, pool, budget,
 
 def main():
     ap = argparse.ArgumentParser()
-    ap.add_argument("--task", required=True)
+    ap.add_argument("--task", default=None,
+                    help="one task; or use --missing to sweep all tasks "
+                         "in --data-dir that best_epsilons.json lacks "
+                         "(reference: launch_missing_modelselector.py)")
+    ap.add_argument("--missing", action="store_true")
     ap.add_argument("--data-dir", default="data")
     ap.add_argument("--eps", default="0.35:0.49:0.01",
                     help="start:stop:step grid")
@@ -110,21 +114,38 @@ def main():
 
     lo, hi, st = (float(x) for x in args.eps.split(":"))
     grid = [round(lo + i * st, 4) for i in range(int((hi - lo) / st) + 1)]
-    chosen, results = run_grid_search(
-        args.task, args.data_dir, grid, args.realisations, args.pool,
-        args.budget, args.threshold)
-    print(f"best epsilon for {args.task}: {chosen}")
 
-    best = {}
-    if os.path.exists(args.out):
-        with open(args.out) as f:
-            best = json.load(f)
-    best[args.task] = chosen
-    tmp = args.out + ".tmp"
-    with open(tmp, "w") as f:
-        json.dump(best, f, indent=2, sort_keys=True)
-    os.replace(tmp, args.out)  # atomic: no partial files under concurrency
-    print("wrote", args.out)
+    if args.missing:
+        have = {}
+        if os.path.exists(args.out):
+            with open(args.out) as f:
+                have = json.load(f)
+        tasks = sorted(f[:-3] for f in os.listdir(args.data_dir)
+                       if f.endswith(".pt")
+                       and not f.endswith("_labels.pt")
+                       and f[:-3] not in have)
+        if not tasks:
+            print("no tasks missing from", args.out)
+            return
+    else:
+        assert args.task, "--task or --missing required"
+        tasks = [args.task]
+
+    for task in tasks:
+        chosen, results = run_grid_search(
+            task, args.data_dir, grid, args.realisations, args.pool,
+            args.budget, args.threshold)
+        print(f"best epsilon for {task}: {chosen}")
+        best = {}
+        if os.path.exists(args.out):
+            with open(args.out) as f:
+                best = json.load(f)
+        best[task] = chosen
+        tmp = args.out + ".tmp"
+        with open(tmp, "w") as f:
+            json.dump(best, f, indent=2, sort_keys=True)
+        os.replace(tmp, args.out)  # atomic under concurrency
+        print("wrote", args.out)
 
 
 if __name__ == "__main__":

@@ -148,3 +148,36 @@ def test_sharded_table_impl_equals_single(tmp_path):
     for p in procs:
         p.join(timeout=60)
     assert results[0] == single and results[1] == single
+
+
+@pytest.mark.timeout(300)
+def test_main_sharded_torchrun_cpu(tmp_path):
+    """End-to-end `torchrun --nproc-per-node 2 main.py --sharded` on CPU
+    (gloo): the full harness loop with the model axis sharded; rank 0
+    writes the tracking DB."""
+    import subprocess
+    import sqlite3
+    import sys
+
+    from coda_amd.datasets import write_synthetic_task
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    write_synthetic_task(str(tmp_path / "data"), name="sh", H=6, N=120,
+                         C=4, seed=7)
+    (tmp_path / "main.py").symlink_to(os.path.join(repo, "main.py"))
+    env = dict(os.environ)
+    env["PYTHONPATH"] = repo + os.pathsep + env.get("PYTHONPATH", "")
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29659", "main.py", "--task", "sh",
+         "--data-dir", "data", "--method", "coda", "--iters", "3",
+         "--seeds", "1", "--sharded", "--device", "cpu",
+         "--chunk-size", "32"],
+        cwd=str(tmp_path), env=env, capture_output=True, text=True,
+        timeout=280)
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    conn = sqlite3.connect(str(tmp_path / "coda.sqlite"))
+    n = conn.execute(
+        "SELECT COUNT(*) FROM metrics WHERE key='regret'").fetchone()[0]
+    conn.close()
+    assert n == 3
