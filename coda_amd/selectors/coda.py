@@ -163,7 +163,30 @@ class CODA(ModelSelector):
             adjusted = ops.pi_hat_partial(self.dirichlets, self.dataset.preds)
         self.comm.all_reduce_(adjusted)
         self._adjusted = adjusted
-        self.pi_hat_xi, self.pi_hat = ops.pi_hat_normalize(adjusted)
+        self._row_sums = adjusted.sum(dim=-1)
+        self._refresh_pi_hat()
+
+    def _refresh_pi_hat(self):
+        """pi_hat (C,) from the maintained adjusted/row_sums WITHOUT
+        materializing the (N, C) per-item normalization: the marginal is
+        one GEMV, pi = adjusted^T @ (1/rowsum), then self-normalized
+        (identical math to the reference's pi_hat_xi.sum(0) route,
+        coda/coda.py:229-233). Per-item rows are built on demand
+        (candidate gathers in eig_batched; the pi_hat_xi property)."""
+        inv = 1.0 / self._row_sums.clamp_min(1e-12)
+        pi = torch.mv(self._adjusted.t(), inv)
+        self.pi_hat = pi / pi.sum()
+        self._pi_xi_cache = None
+
+    @property
+    def pi_hat_xi(self):
+        """(N, C) per-item class posterior (materialized on demand)."""
+        if self._pi_xi_cache is None:
+            self._pi_xi_cache = self._adjusted /                 self._row_sums.clamp_min(1e-12).unsqueeze(-1)
+        return self._pi_xi_cache
+
+    def _pi_xi_rows(self, ids):
+        return self._adjusted[ids] /             self._row_sums[ids].clamp_min(1e-12).unsqueeze(-1)
 
     # ------------------------------------------------------------------
     def _pbest_rows_before(self):
@@ -209,7 +232,7 @@ class CODA(ModelSelector):
         for s in range(0, cand.numel(), self.chunk_size):
             ids = cand[s:s + self.chunk_size]
             chunk_classes = self.classes[:, ids].t().contiguous()  # (B, Hl)
-            pi_xi = self.pi_hat_xi[ids]                            # (B, C)
+            pi_xi = self._pi_xi_rows(ids)                          # (B, C)
             if self.comm.is_distributed and tables is not None:
                 from ..ops import table as tops
                 eig = tops.eig_chunk_table_sharded(
@@ -290,7 +313,8 @@ class CODA(ModelSelector):
         delta = ops.pi_hat_delta(self.dataset.preds, self.classes[:, idx])
         self.comm.all_reduce_(delta)
         self._adjusted[:, int(true_class)] += self.update_strength * delta
-        self.pi_hat_xi, self.pi_hat = ops.pi_hat_normalize(self._adjusted)
+        self._row_sums += self.update_strength * delta
+        self._refresh_pi_hat()
         self.labeled_idxs.append(idx)
         self.labels.append(int(true_class))
         self.q_vals.append(selection_prob)
