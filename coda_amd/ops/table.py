@@ -42,6 +42,11 @@ from .reference import (EPS_PROB, GRID_HI, GRID_LO, PBEST_NUM_POINTS,
 _LOG2E = 1.4426950408889634
 
 
+def _use_bf16_gemm() -> bool:
+    import os
+    return os.environ.get("CODA_AMD_V2_GEMM", "bf16") != "fp32"
+
+
 def _class_csr(cls: torch.Tensor, C: int):
     """Sort each candidate's models by predicted class -> CSR buckets.
 
@@ -62,6 +67,11 @@ class EigTables(NamedTuple):
     delta: torch.Tensor    # (C, H, P)
     s_base: torch.Tensor   # (C, P)
     weights: torch.Tensor  # (P,) trapz weights * dx
+    # bf16 mirror of EG reshaped (C, 2H, P) for the GEMM fast path (GPU
+    # only; bf16 inputs + f32 accumulate move the pairing to the bf16
+    # MFMA rate - measured EIG impact ~5e-7 absolute, at the level of
+    # fp32 reduction-order noise). CODA_AMD_V2_GEMM=fp32 disables.
+    eg16: torch.Tensor = None
 
 
 def table_precompute(alpha_cc: torch.Tensor, beta_cc: torch.Tensor,
@@ -99,13 +109,16 @@ def table_precompute(alpha_cc: torch.Tensor, beta_cc: torch.Tensor,
 
     w = torch.full((num_points,), dx, device=dev)
     w[0] = w[-1] = 0.5 * dx
-    return EigTables(EG, delta, s_base, w)
+    eg16 = None
+    if EG.is_cuda and _use_bf16_gemm():
+        eg16 = EG.reshape(C, 2 * H, num_points).to(torch.bfloat16)
+    return EigTables(EG, delta, s_base, w, eg16)
 
 
 def pbest_hyp_table(tables: EigTables,
                     chunk_classes: torch.Tensor) -> torch.Tensor:
     """Normalized hypothetical P(best): (B, H) classes -> (B, C, H)."""
-    EG, delta, s_base, w = tables
+    EG, delta, s_base, w = tables.EG, tables.delta, tables.s_base, tables.weights
     C, H, _, P = EG.shape
     B = chunk_classes.shape[0]
     cls = chunk_classes.long()
@@ -144,14 +157,18 @@ def eig_chunk_table(tables: EigTables, chunk_classes: torch.Tensor,
     GPU path: the es_build / eig_assemble_k fusion kernels around the
     batched GEMM (see pbest.hip); CPU path: the torch composition above.
     """
-    EG, delta, s_base, w = tables
+    EG, delta, s_base, w = tables.EG, tables.delta, tables.s_base, tables.weights
     C, H, _, P = EG.shape
     import coda_amd.ops as O
     if EG.is_cuda and P == PBEST_NUM_POINTS and O._want_hip(EG):
         cls32 = chunk_classes.to(torch.int32).contiguous()
         hvals, offsets = _class_csr(chunk_classes.long(), C)
         ES = O._ext.es_build(s_base, delta, hvals, offsets, w)  # (C, B, P)
-        M = torch.bmm(ES, EG.reshape(C, 2 * H, P).transpose(1, 2))
+        if tables.eg16 is not None:
+            M = torch.bmm(ES.to(torch.bfloat16),
+                          tables.eg16.transpose(1, 2)).float()
+        else:
+            M = torch.bmm(ES, EG.reshape(C, 2 * H, P).transpose(1, 2))
         h_after = O._ext.eig_assemble_k(M, cls32, pi_hat.contiguous(),
                                         pbest_before.contiguous(),
                                         mixture0.contiguous())  # (B, C)
@@ -177,17 +194,23 @@ def table_update_rows(tables: EigTables, alpha_cc: torch.Tensor,
     if (alpha_cc.is_cuda and tables.EG.shape[-1] == PBEST_NUM_POINTS
             and rows.numel() <= 4 and O._want_hip(alpha_cc)):
         # per-class refresh kernel: one wave per (model, variant) curve
+        H = alpha_cc.shape[0]
         for c in rows.tolist():
             eg, lc = O._ext.beta_row_tables(
                 alpha_cc[:, c].contiguous(), beta_cc[:, c].contiguous(),
                 float(update_weight))
             tables.EG[c] = eg
+            if tables.eg16 is not None:
+                tables.eg16[c] = eg.reshape(2 * H, -1).to(torch.bfloat16)
             tables.delta[c] = lc[:, 1] - lc[:, 0]
             tables.s_base[c] = lc[:, 0].sum(0)
         return tables
     sub = table_precompute(alpha_cc[:, rows], beta_cc[:, rows],
                            update_weight, tables.EG.shape[-1])
     tables.EG[rows] = sub.EG
+    if tables.eg16 is not None:
+        H = alpha_cc.shape[0]
+        tables.eg16[rows] = sub.EG.reshape(rows.numel(), 2 * H, -1)             .to(torch.bfloat16)
     tables.delta[rows] = sub.delta
     tables.s_base[rows] = sub.s_base
     return tables
@@ -217,7 +240,7 @@ def eig_chunk_table_sharded(tables: EigTables, s_base_all: torch.Tensor,
                             mixture0_local: torch.Tensor,
                             H_before: torch.Tensor, comm) -> torch.Tensor:
     """(B,) EIG with local tables; identical on every rank."""
-    EG, delta, _, w = tables
+    EG, delta, w = tables.EG, tables.delta, tables.weights
     C, Hl, _, P = EG.shape
     B = cls_local.shape[0]
     cls_l = cls_local.long()
@@ -235,7 +258,11 @@ def eig_chunk_table_sharded(tables: EigTables, s_base_all: torch.Tensor,
         hvals, offsets = _class_csr(cls_all, C)
         ES = O._ext.es_build_gathered(s_base_all.contiguous(), sel_all,
                                       hvals, offsets, w)     # (C, B, P)
-        M = torch.bmm(ES, EG.reshape(C, 2 * Hl, P).transpose(1, 2))
+        if tables.eg16 is not None:
+            M = torch.bmm(ES.to(torch.bfloat16),
+                          tables.eg16.transpose(1, 2)).float()
+        else:
+            M = torch.bmm(ES, EG.reshape(C, 2 * Hl, P).transpose(1, 2))
         tot = O._ext.eig_totals(M, cls_l32)                  # (B, C) partial
         comm.all_reduce_(tot)
         h_after = O._ext.eig_entropy(M, cls_l32, tot.contiguous(),
