@@ -60,7 +60,7 @@ def test_launcher_idempotent(pipeline_dir):
          "--pred-dir", "data",
          "--methods", "iid,coda-lr=0.01-mult=2.0-no-prefilter",
          "--seeds", "2", "--iters", "3", "--gpus", "0"],
-        cwd=str(d), env=env, capture_output=True, text=True, timeout=120)
+        cwd=str(d), env=env, capture_output=True, text=True, timeout=400)
     assert r.returncode == 0
     assert "No jobs to run!" in r.stdout
 
@@ -70,7 +70,7 @@ def test_aggregate_and_analysis(pipeline_dir):
     r = subprocess.run(
         [sys.executable, os.path.join(REPO, "scripts",
                                       "aggregate_results.py")],
-        cwd=str(d), env=env, capture_output=True, text=True, timeout=120)
+        cwd=str(d), env=env, capture_output=True, text=True, timeout=400)
     assert r.returncode == 0, r.stderr
     conn = sqlite3.connect(str(d / "coda.sqlite"))
     n_mean = conn.execute(
@@ -81,7 +81,7 @@ def test_aggregate_and_analysis(pipeline_dir):
     r = subprocess.run(
         [sys.executable, os.path.join(REPO, "analysis", "tab1.py"),
          "--step", "3"],
-        cwd=str(d), env=env, capture_output=True, text=True, timeout=120)
+        cwd=str(d), env=env, capture_output=True, text=True, timeout=400)
     assert r.returncode == 0, r.stderr
     assert "taskx" in r.stdout and "tasky" in r.stdout
     assert "Random Sampling" in r.stdout and "CODA (Ours)" in r.stdout
@@ -89,27 +89,27 @@ def test_aggregate_and_analysis(pipeline_dir):
     r = subprocess.run(
         [sys.executable, os.path.join(REPO, "analysis", "fig1.py"),
          "--out", "fig1.png"],
-        cwd=str(d), env=env, capture_output=True, text=True, timeout=120)
+        cwd=str(d), env=env, capture_output=True, text=True, timeout=400)
     assert r.returncode == 0, r.stderr
     assert (d / "fig1.png").exists()
 
     r = subprocess.run(
         [sys.executable, os.path.join(REPO, "analysis", "fig5.py"),
          "--out", "fig5.png"],
-        cwd=str(d), env=env, capture_output=True, text=True, timeout=120)
+        cwd=str(d), env=env, capture_output=True, text=True, timeout=400)
     assert r.returncode == 0, r.stderr
     assert (d / "fig5.png").exists()
 
     r = subprocess.run(
         [sys.executable, os.path.join(REPO, "analysis", "fig4.py")],
-        cwd=str(d), env=env, capture_output=True, text=True, timeout=120)
+        cwd=str(d), env=env, capture_output=True, text=True, timeout=400)
     assert r.returncode == 0, r.stderr
     assert "failure rate" in r.stdout
 
     r = subprocess.run(
         [sys.executable, os.path.join(REPO, "analysis", "fig3.py"),
          "--data-dir", "data"],
-        cwd=str(d), env=env, capture_output=True, text=True, timeout=120)
+        cwd=str(d), env=env, capture_output=True, text=True, timeout=400)
     assert r.returncode == 0, r.stderr
     assert "by class count" in r.stdout
 
@@ -119,7 +119,7 @@ def test_clear_db_selected(pipeline_dir):
     r = subprocess.run(
         [sys.executable, os.path.join(REPO, "scripts", "clear_db.py"),
          "--experiments", "tasky", "--yes"],
-        cwd=str(d), env=env, capture_output=True, text=True, timeout=60)
+        cwd=str(d), env=env, capture_output=True, text=True, timeout=200)
     assert r.returncode == 0, r.stderr
     conn = sqlite3.connect(str(d / "coda.sqlite"))
     exps = [x[0] for x in conn.execute(

@@ -135,7 +135,7 @@ with tracking.start_run(run_name=f"w{{wid}}"):
     procs = [subprocess.Popen([sys.executable, "-c", script, str(i)])
              for i in range(8)]
     for p in procs:
-        assert p.wait(timeout=120) == 0
+        assert p.wait(timeout=400) == 0
     import sqlite3
     conn = sqlite3.connect(db)
     n_runs = conn.execute("SELECT COUNT(*) FROM runs").fetchone()[0]
