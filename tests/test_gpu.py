@@ -216,9 +216,11 @@ class TestStorageGPU:
 
 
 class TestTableFusionKernels:
-    def test_eig_chunk_table_gpu_matches_cpu(self, dev):
+    def test_eig_chunk_table_gpu_matches_cpu(self, dev, monkeypatch):
         """The es_build/eig_assemble_k fused GPU path == the torch
-        composition on CPU."""
+        composition on CPU (fp32 GEMM for exact comparison; the default
+        bf16 GEMM is checked separately with an absolute tolerance)."""
+        monkeypatch.setenv("CODA_AMD_V2_GEMM", "fp32")
         from coda_amd import ops
         from coda_amd.ops import table as T
         g = torch.Generator().manual_seed(41)
@@ -236,10 +238,22 @@ class TestTableFusionKernels:
         want = T.eig_chunk_table(t_cpu, cls, pb0, pi, pi_xi, m0, H0)
 
         t_gpu = T.table_precompute(a0.to(dev), b0.to(dev))
+        assert t_gpu.eg16 is None  # fp32 mode
         got = T.eig_chunk_table(t_gpu, cls.to(dev), pb0.to(dev),
                                 pi.to(dev), pi_xi.to(dev), m0.to(dev),
                                 H0.to(dev)).cpu()
         torch.testing.assert_close(got, want, rtol=5e-3, atol=1e-5)
+
+        # default bf16 pairing GEMM: absolute accuracy at the fp32
+        # reduction-noise level (EIG entropies are insensitive to the
+        # 0.4% per-element input rounding; measured ~5e-7)
+        monkeypatch.delenv("CODA_AMD_V2_GEMM")
+        t16 = T.table_precompute(a0.to(dev), b0.to(dev))
+        assert t16.eg16 is not None
+        got16 = T.eig_chunk_table(t16, cls.to(dev), pb0.to(dev),
+                                  pi.to(dev), pi_xi.to(dev), m0.to(dev),
+                                  H0.to(dev)).cpu()
+        assert float((got16 - want).abs().max()) < 5e-5
 
     def test_sharded_table_gpu_path_single_rank(self, dev):
         """es_build_gathered + eig_totals/eig_entropy with a no-op Comm
