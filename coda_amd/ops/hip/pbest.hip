@@ -524,13 +524,14 @@ eig_phase2_kernel(const float* __restrict__ alpha_t,
 //       v = [cls(b,h)==c], normalize over h, and emit the log2-entropy
 //       H_after[b,c] of the updated P(best) mixture.
 // ---------------------------------------------------------------------------
+template <typename TOUT>
 __global__ void __launch_bounds__(BLOCK)
 es_build_kernel(const float* __restrict__ s_base,   // (C, P)
                 const float* __restrict__ delta,    // (C, H, P)
                 const int* __restrict__ hvals,      // (B, H) h sorted by class
                 const int* __restrict__ offsets,    // (B, C+1) CSR
                 const float* __restrict__ w,        // (P,)
-                float* __restrict__ es,             // (C, B, P)
+                TOUT* __restrict__ es,              // (C, B, P)
                 int B, int C, int H) {
     const int r = blockIdx.x * ROWS_PER_BLOCK + (threadIdx.x >> 6);
     if (r >= B * C) return;
@@ -554,17 +555,16 @@ es_build_kernel(const float* __restrict__ s_base,   // (C, P)
         acc[0] += d.x; acc[1] += d.y; acc[2] += d.z; acc[3] += d.w;
     }
     const float4 wv = *reinterpret_cast<const float4*>(w + p0);
-    float4 out;
-    out.x = exp2f(acc[0]) * wv.x;
-    out.y = exp2f(acc[1]) * wv.y;
-    out.z = exp2f(acc[2]) * wv.z;
-    out.w = exp2f(acc[3]) * wv.w;
-    *reinterpret_cast<float4*>(
-        es + ((size_t)c * B + b) * P_POINTS + p0) = out;
+    TOUT* dst = es + ((size_t)c * B + b) * P_POINTS + p0;
+    dst[0] = (TOUT)(exp2f(acc[0]) * wv.x);
+    dst[1] = (TOUT)(exp2f(acc[1]) * wv.y);
+    dst[2] = (TOUT)(exp2f(acc[2]) * wv.z);
+    dst[3] = (TOUT)(exp2f(acc[3]) * wv.w);
 }
 
+template <typename TM>
 __global__ void __launch_bounds__(BLOCK)
-eig_assemble_kernel(const float* __restrict__ m,            // (C, B, 2H)
+eig_assemble_kernel(const TM* __restrict__ m,               // (C, B, 2H)
                     const int* __restrict__ cls,            // (B, H)
                     const float* __restrict__ pi_hat,       // (C,)
                     const float* __restrict__ pbest_before, // (C, H)
@@ -575,12 +575,12 @@ eig_assemble_kernel(const float* __restrict__ m,            // (C, B, 2H)
     if (r >= B * C) return;
     const int b = r / C, c = r - b * C;
     const int lane = threadIdx.x & 63;
-    const float* row = m + ((size_t)c * B + b) * (2 * H);
+    const TM* row = m + ((size_t)c * B + b) * (2 * H);
 
     float total = 0.f;
     for (int h = lane; h < H; h += 64) {
         const int v = (cls[(size_t)b * H + h] == c) ? 1 : 0;
-        total += row[2 * h + v];
+        total += (float)row[2 * h + v];
     }
     total = wave_reduce_sum(total);
     const float inv = 1.0f / fmaxf(total, kEps);
@@ -589,7 +589,7 @@ eig_assemble_kernel(const float* __restrict__ m,            // (C, B, 2H)
     float ent = 0.f;
     for (int h = lane; h < H; h += 64) {
         const int v = (cls[(size_t)b * H + h] == c) ? 1 : 0;
-        const float pb = row[2 * h + v] * inv;
+        const float pb = (float)row[2 * h + v] * inv;
         float mm = mixture0[h]
                  + pi_c * (pb - pbest_before[(size_t)c * H + h]);
         mm = fmaxf(mm, 1e-12f);
@@ -605,13 +605,14 @@ eig_assemble_kernel(const float* __restrict__ m,            // (C, B, 2H)
 // normalization needs a GLOBAL total, so the assemble step splits into a
 // totals kernel (partial sums, all-reduced by the host) and an entropy
 // kernel consuming the reduced totals.
+template <typename TOUT>
 __global__ void __launch_bounds__(BLOCK)
 es_build_gathered_kernel(const float* __restrict__ s_base_all,  // (C, P)
                          const float* __restrict__ sel_all,  // (B, Hg, P)
                          const int* __restrict__ hvals,      // (B, Hg) CSR
                          const int* __restrict__ offsets,    // (B, C+1)
                          const float* __restrict__ w,        // (P,)
-                         float* __restrict__ es,             // (C, B, P)
+                         TOUT* __restrict__ es,              // (C, B, P)
                          int B, int C, int Hg) {
     const int r = blockIdx.x * ROWS_PER_BLOCK + (threadIdx.x >> 6);
     if (r >= B * C) return;
@@ -633,17 +634,16 @@ es_build_gathered_kernel(const float* __restrict__ s_base_all,  // (C, P)
         acc[0] += d.x; acc[1] += d.y; acc[2] += d.z; acc[3] += d.w;
     }
     const float4 wv = *reinterpret_cast<const float4*>(w + p0);
-    float4 out;
-    out.x = exp2f(acc[0]) * wv.x;
-    out.y = exp2f(acc[1]) * wv.y;
-    out.z = exp2f(acc[2]) * wv.z;
-    out.w = exp2f(acc[3]) * wv.w;
-    *reinterpret_cast<float4*>(
-        es + ((size_t)c * B + b) * P_POINTS + p0) = out;
+    TOUT* dst = es + ((size_t)c * B + b) * P_POINTS + p0;
+    dst[0] = (TOUT)(exp2f(acc[0]) * wv.x);
+    dst[1] = (TOUT)(exp2f(acc[1]) * wv.y);
+    dst[2] = (TOUT)(exp2f(acc[2]) * wv.z);
+    dst[3] = (TOUT)(exp2f(acc[3]) * wv.w);
 }
 
+template <typename TM>
 __global__ void __launch_bounds__(BLOCK)
-eig_totals_kernel(const float* __restrict__ m,    // (C, B, 2H)
+eig_totals_kernel(const TM* __restrict__ m,       // (C, B, 2H)
                   const int* __restrict__ cls,    // (B, H) local
                   float* __restrict__ totals,     // (B, C) partial
                   int B, int C, int H) {
@@ -651,18 +651,19 @@ eig_totals_kernel(const float* __restrict__ m,    // (C, B, 2H)
     if (r >= B * C) return;
     const int b = r / C, c = r - b * C;
     const int lane = threadIdx.x & 63;
-    const float* row = m + ((size_t)c * B + b) * (2 * H);
+    const TM* row = m + ((size_t)c * B + b) * (2 * H);
     float total = 0.f;
     for (int h = lane; h < H; h += 64) {
         const int v = (cls[(size_t)b * H + h] == c) ? 1 : 0;
-        total += row[2 * h + v];
+        total += (float)row[2 * h + v];
     }
     total = wave_reduce_sum(total);
     if (lane == 0) totals[r] = total;
 }
 
+template <typename TM>
 __global__ void __launch_bounds__(BLOCK)
-eig_entropy_kernel(const float* __restrict__ m,            // (C, B, 2H)
+eig_entropy_kernel(const TM* __restrict__ m,               // (C, B, 2H)
                    const int* __restrict__ cls,            // (B, H) local
                    const float* __restrict__ totals,       // (B, C) GLOBAL
                    const float* __restrict__ pi_hat,       // (C,)
@@ -674,13 +675,13 @@ eig_entropy_kernel(const float* __restrict__ m,            // (C, B, 2H)
     if (r >= B * C) return;
     const int b = r / C, c = r - b * C;
     const int lane = threadIdx.x & 63;
-    const float* row = m + ((size_t)c * B + b) * (2 * H);
+    const TM* row = m + ((size_t)c * B + b) * (2 * H);
     const float inv = 1.0f / fmaxf(totals[r], kEps);
     const float pi_c = pi_hat[c];
     float ent = 0.f;
     for (int h = lane; h < H; h += 64) {
         const int v = (cls[(size_t)b * H + h] == c) ? 1 : 0;
-        const float pb = row[2 * h + v] * inv;
+        const float pb = (float)row[2 * h + v] * inv;
         float mm = mixture0[h]
                  + pi_c * (pb - pbest_before[(size_t)c * H + h]);
         mm = fmaxf(mm, 1e-12f);
@@ -925,7 +926,7 @@ std::vector<torch::Tensor> eig_phase2(torch::Tensor alpha_cc,
 
 torch::Tensor es_build(torch::Tensor s_base, torch::Tensor delta,
                        torch::Tensor hvals, torch::Tensor offsets,
-                       torch::Tensor w) {
+                       torch::Tensor w, bool bf16_out) {
     check_f32_cuda(s_base, "s_base");
     check_f32_cuda(delta, "delta");
     check_f32_cuda(w, "w");
@@ -935,15 +936,29 @@ torch::Tensor es_build(torch::Tensor s_base, torch::Tensor delta,
     TORCH_CHECK(s_base.size(1) == P_POINTS, "P must be 256");
     const int C = s_base.size(0), H = delta.size(1);
     const int B = hvals.size(0);
-    auto es = torch::empty({C, B, P_POINTS}, s_base.options());
+    auto es = torch::empty({C, B, P_POINTS},
+                           s_base.options().dtype(
+                               bf16_out ? torch::kBFloat16
+                                        : torch::kFloat32));
     const int R = B * C;
     const int blocks = (R + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
     auto stream = c10::hip::getCurrentHIPStream();
-    hipLaunchKernelGGL(es_build_kernel, dim3(blocks), dim3(BLOCK), 0,
-                       stream.stream(), s_base.data_ptr<float>(),
-                       delta.data_ptr<float>(), hvals.data_ptr<int>(),
-                       offsets.data_ptr<int>(),
-                       w.data_ptr<float>(), es.data_ptr<float>(), B, C, H);
+    if (bf16_out) {
+        hipLaunchKernelGGL(es_build_kernel<hip_bfloat16>, dim3(blocks),
+                           dim3(BLOCK), 0, stream.stream(),
+                           s_base.data_ptr<float>(),
+                           delta.data_ptr<float>(), hvals.data_ptr<int>(),
+                           offsets.data_ptr<int>(), w.data_ptr<float>(),
+                           reinterpret_cast<hip_bfloat16*>(es.data_ptr()),
+                           B, C, H);
+    } else {
+        hipLaunchKernelGGL(es_build_kernel<float>, dim3(blocks),
+                           dim3(BLOCK), 0, stream.stream(),
+                           s_base.data_ptr<float>(),
+                           delta.data_ptr<float>(), hvals.data_ptr<int>(),
+                           offsets.data_ptr<int>(), w.data_ptr<float>(),
+                           es.data_ptr<float>(), B, C, H);
+    }
     C10_HIP_CHECK(hipGetLastError());
     return es;
 }
@@ -952,23 +967,36 @@ torch::Tensor eig_assemble_k(torch::Tensor m, torch::Tensor cls,
                              torch::Tensor pi_hat,
                              torch::Tensor pbest_before,
                              torch::Tensor mixture0) {
-    check_f32_cuda(m, "m");
+    TORCH_CHECK(m.is_cuda() && m.is_contiguous(), "m");
     check_f32_cuda(pi_hat, "pi_hat");
     check_f32_cuda(pbest_before, "pbest_before");
     check_f32_cuda(mixture0, "mixture0");
     TORCH_CHECK(cls.scalar_type() == torch::kInt32, "cls must be int32");
     const int C = m.size(0), B = m.size(1);
     const int H = m.size(2) / 2;
-    auto h_after = torch::empty({B, C}, m.options());
+    auto h_after = torch::empty({B, C},
+                                m.options().dtype(torch::kFloat32));
     const int R = B * C;
     const int blocks = (R + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
     auto stream = c10::hip::getCurrentHIPStream();
-    hipLaunchKernelGGL(eig_assemble_kernel, dim3(blocks), dim3(BLOCK), 0,
-                       stream.stream(), m.data_ptr<float>(),
-                       cls.data_ptr<int>(), pi_hat.data_ptr<float>(),
-                       pbest_before.data_ptr<float>(),
-                       mixture0.data_ptr<float>(),
-                       h_after.data_ptr<float>(), B, C, H);
+    if (m.scalar_type() == torch::kBFloat16) {
+        hipLaunchKernelGGL(eig_assemble_kernel<hip_bfloat16>, dim3(blocks),
+                           dim3(BLOCK), 0, stream.stream(),
+                           reinterpret_cast<const hip_bfloat16*>(
+                               m.data_ptr()),
+                           cls.data_ptr<int>(), pi_hat.data_ptr<float>(),
+                           pbest_before.data_ptr<float>(),
+                           mixture0.data_ptr<float>(),
+                           h_after.data_ptr<float>(), B, C, H);
+    } else {
+        hipLaunchKernelGGL(eig_assemble_kernel<float>, dim3(blocks),
+                           dim3(BLOCK), 0, stream.stream(),
+                           m.data_ptr<float>(),
+                           cls.data_ptr<int>(), pi_hat.data_ptr<float>(),
+                           pbest_before.data_ptr<float>(),
+                           mixture0.data_ptr<float>(),
+                           h_after.data_ptr<float>(), B, C, H);
+    }
     C10_HIP_CHECK(hipGetLastError());
     return h_after;
 }
@@ -977,7 +1005,7 @@ torch::Tensor eig_assemble_k(torch::Tensor m, torch::Tensor cls,
 torch::Tensor es_build_gathered(torch::Tensor s_base_all,
                                 torch::Tensor sel_all,
                                 torch::Tensor hvals, torch::Tensor offsets,
-                                torch::Tensor w) {
+                                torch::Tensor w, bool bf16_out) {
     check_f32_cuda(s_base_all, "s_base_all");
     check_f32_cuda(sel_all, "sel_all");
     check_f32_cuda(w, "w");
@@ -986,30 +1014,56 @@ torch::Tensor es_build_gathered(torch::Tensor s_base_all,
                 "hvals/offsets must be int32");
     const int C = s_base_all.size(0);
     const int B = hvals.size(0), Hg = hvals.size(1);
-    auto es = torch::empty({C, B, P_POINTS}, s_base_all.options());
+    auto es = torch::empty({C, B, P_POINTS},
+                           s_base_all.options().dtype(
+                               bf16_out ? torch::kBFloat16
+                                        : torch::kFloat32));
     const int R = B * C;
     const int blocks = (R + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
     auto stream = c10::hip::getCurrentHIPStream();
-    hipLaunchKernelGGL(es_build_gathered_kernel, dim3(blocks), dim3(BLOCK),
-                       0, stream.stream(), s_base_all.data_ptr<float>(),
-                       sel_all.data_ptr<float>(), hvals.data_ptr<int>(),
-                       offsets.data_ptr<int>(),
-                       w.data_ptr<float>(), es.data_ptr<float>(), B, C, Hg);
+    if (bf16_out) {
+        hipLaunchKernelGGL(es_build_gathered_kernel<hip_bfloat16>,
+                           dim3(blocks), dim3(BLOCK), 0, stream.stream(),
+                           s_base_all.data_ptr<float>(),
+                           sel_all.data_ptr<float>(),
+                           hvals.data_ptr<int>(), offsets.data_ptr<int>(),
+                           w.data_ptr<float>(),
+                           reinterpret_cast<hip_bfloat16*>(es.data_ptr()),
+                           B, C, Hg);
+    } else {
+        hipLaunchKernelGGL(es_build_gathered_kernel<float>, dim3(blocks),
+                           dim3(BLOCK), 0, stream.stream(),
+                           s_base_all.data_ptr<float>(),
+                           sel_all.data_ptr<float>(),
+                           hvals.data_ptr<int>(), offsets.data_ptr<int>(),
+                           w.data_ptr<float>(), es.data_ptr<float>(),
+                           B, C, Hg);
+    }
     C10_HIP_CHECK(hipGetLastError());
     return es;
 }
 
 torch::Tensor eig_totals(torch::Tensor m, torch::Tensor cls) {
-    check_f32_cuda(m, "m");
+    TORCH_CHECK(m.is_cuda() && m.is_contiguous(), "m");
     const int C = m.size(0), B = m.size(1), H = m.size(2) / 2;
-    auto totals = torch::empty({B, C}, m.options());
+    auto totals = torch::empty({B, C},
+                               m.options().dtype(torch::kFloat32));
     const int R = B * C;
     const int blocks = (R + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
     auto stream = c10::hip::getCurrentHIPStream();
-    hipLaunchKernelGGL(eig_totals_kernel, dim3(blocks), dim3(BLOCK), 0,
-                       stream.stream(), m.data_ptr<float>(),
-                       cls.data_ptr<int>(), totals.data_ptr<float>(),
-                       B, C, H);
+    if (m.scalar_type() == torch::kBFloat16) {
+        hipLaunchKernelGGL(eig_totals_kernel<hip_bfloat16>, dim3(blocks),
+                           dim3(BLOCK), 0, stream.stream(),
+                           reinterpret_cast<const hip_bfloat16*>(
+                               m.data_ptr()),
+                           cls.data_ptr<int>(), totals.data_ptr<float>(),
+                           B, C, H);
+    } else {
+        hipLaunchKernelGGL(eig_totals_kernel<float>, dim3(blocks),
+                           dim3(BLOCK), 0, stream.stream(),
+                           m.data_ptr<float>(), cls.data_ptr<int>(),
+                           totals.data_ptr<float>(), B, C, H);
+    }
     C10_HIP_CHECK(hipGetLastError());
     return totals;
 }
@@ -1018,20 +1072,34 @@ torch::Tensor eig_entropy(torch::Tensor m, torch::Tensor cls,
                           torch::Tensor totals, torch::Tensor pi_hat,
                           torch::Tensor pbest_before,
                           torch::Tensor mixture0) {
-    check_f32_cuda(m, "m");
+    TORCH_CHECK(m.is_cuda() && m.is_contiguous(), "m");
     check_f32_cuda(totals, "totals");
     const int C = m.size(0), B = m.size(1), H = m.size(2) / 2;
-    auto h_after = torch::empty({B, C}, m.options());
+    auto h_after = torch::empty({B, C},
+                                m.options().dtype(torch::kFloat32));
     const int R = B * C;
     const int blocks = (R + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
     auto stream = c10::hip::getCurrentHIPStream();
-    hipLaunchKernelGGL(eig_entropy_kernel, dim3(blocks), dim3(BLOCK), 0,
-                       stream.stream(), m.data_ptr<float>(),
-                       cls.data_ptr<int>(), totals.data_ptr<float>(),
-                       pi_hat.data_ptr<float>(),
-                       pbest_before.data_ptr<float>(),
-                       mixture0.data_ptr<float>(),
-                       h_after.data_ptr<float>(), B, C, H);
+    if (m.scalar_type() == torch::kBFloat16) {
+        hipLaunchKernelGGL(eig_entropy_kernel<hip_bfloat16>, dim3(blocks),
+                           dim3(BLOCK), 0, stream.stream(),
+                           reinterpret_cast<const hip_bfloat16*>(
+                               m.data_ptr()),
+                           cls.data_ptr<int>(), totals.data_ptr<float>(),
+                           pi_hat.data_ptr<float>(),
+                           pbest_before.data_ptr<float>(),
+                           mixture0.data_ptr<float>(),
+                           h_after.data_ptr<float>(), B, C, H);
+    } else {
+        hipLaunchKernelGGL(eig_entropy_kernel<float>, dim3(blocks),
+                           dim3(BLOCK), 0, stream.stream(),
+                           m.data_ptr<float>(), cls.data_ptr<int>(),
+                           totals.data_ptr<float>(),
+                           pi_hat.data_ptr<float>(),
+                           pbest_before.data_ptr<float>(),
+                           mixture0.data_ptr<float>(),
+                           h_after.data_ptr<float>(), B, C, H);
+    }
     C10_HIP_CHECK(hipGetLastError());
     return h_after;
 }

@@ -163,10 +163,10 @@ def eig_chunk_table(tables: EigTables, chunk_classes: torch.Tensor,
     if EG.is_cuda and P == PBEST_NUM_POINTS and O._want_hip(EG):
         cls32 = chunk_classes.to(torch.int32).contiguous()
         hvals, offsets = _class_csr(chunk_classes.long(), C)
-        ES = O._ext.es_build(s_base, delta, hvals, offsets, w)  # (C, B, P)
-        if tables.eg16 is not None:
-            M = torch.bmm(ES.to(torch.bfloat16),
-                          tables.eg16.transpose(1, 2)).float()
+        bf16 = tables.eg16 is not None
+        ES = O._ext.es_build(s_base, delta, hvals, offsets, w, bf16)
+        if bf16:
+            M = torch.bmm(ES, tables.eg16.transpose(1, 2))
         else:
             M = torch.bmm(ES, EG.reshape(C, 2 * H, P).transpose(1, 2))
         h_after = O._ext.eig_assemble_k(M, cls32, pi_hat.contiguous(),
@@ -256,11 +256,11 @@ def eig_chunk_table_sharded(tables: EigTables, s_base_all: torch.Tensor,
     if EG.is_cuda and P == PBEST_NUM_POINTS and O._want_hip(EG):
         cls_l32 = cls_l.to(torch.int32).contiguous()
         hvals, offsets = _class_csr(cls_all, C)
+        bf16 = tables.eg16 is not None
         ES = O._ext.es_build_gathered(s_base_all.contiguous(), sel_all,
-                                      hvals, offsets, w)     # (C, B, P)
-        if tables.eg16 is not None:
-            M = torch.bmm(ES.to(torch.bfloat16),
-                          tables.eg16.transpose(1, 2)).float()
+                                      hvals, offsets, w, bf16)
+        if bf16:
+            M = torch.bmm(ES, tables.eg16.transpose(1, 2))
         else:
             M = torch.bmm(ES, EG.reshape(C, 2 * Hl, P).transpose(1, 2))
         tot = O._ext.eig_totals(M, cls_l32)                  # (B, C) partial
