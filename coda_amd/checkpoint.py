@@ -69,6 +69,8 @@ def load_state_dict(selector, state: Dict[str, Any]):
     if cls == "CODA":
         selector._tables = None       # force a fresh v2 table build
         selector._tables_dirty = set()
+        selector._posterior_version += 1
+        selector._pbest_rows_cache = (-1, None)
         selector._active_candidates = [
             i for i in selector.unlabeled_idxs
             if selector._disagreement_host[i]]

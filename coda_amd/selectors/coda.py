@@ -64,6 +64,8 @@ class CODA(ModelSelector):
         self.eig_impl = eig_impl
         self._tables = None          # persistent v2 curve tables
         self._tables_dirty = set()   # class rows touched since last build
+        self._posterior_version = 0  # bumped by add_label
+        self._pbest_rows_cache = (-1, None)
 
         # hyperparams (reference names: coda/coda.py:189-190)
         self.prior_strength = 1.0 - alpha
@@ -192,13 +194,23 @@ class CODA(ModelSelector):
 
     # ------------------------------------------------------------------
     def _pbest_rows_before(self):
-        """(C, Hl) P(best | class row c) under the current posterior."""
+        """(C, Hl) P(best | class row c) under the current posterior.
+
+        Memoized on the posterior version: the rows computed by get_pbest
+        right after a label are identical to the next acquisition's
+        pbest_before (nothing moves the Dirichlets in between)."""
+        ver, cached = self._pbest_rows_cache
+        if ver == self._posterior_version and cached is not None:
+            return cached
         alpha_cc, beta_cc = ops.dirichlet_to_beta(self.dirichlets)  # (Hl, C)
         a, b = alpha_cc.t().contiguous(), beta_cc.t().contiguous()  # (C, Hl)
         if self.comm.is_distributed:
-            return shops.pbest_from_beta_sharded(a, b, self.comm,
+            rows = shops.pbest_from_beta_sharded(a, b, self.comm,
                                                  self.num_points)
-        return ops.pbest_from_beta(a, b, self.num_points)
+        else:
+            rows = ops.pbest_from_beta(a, b, self.num_points)
+        self._pbest_rows_cache = (self._posterior_version, rows)
+        return rows
 
     def eig_batched(self):
         """EIG for every candidate point (reference: coda/coda.py:235-281)."""
@@ -312,6 +324,7 @@ class CODA(ModelSelector):
             self.classes[:, idx], self.C).to(self.dirichlets.dtype)  # (Hl, C)
         self.dirichlets[:, int(true_class)] += self.update_strength * onehot
         self._tables_dirty.add(int(true_class))
+        self._posterior_version += 1
         delta = ops.pi_hat_delta(self.dataset.preds, self.classes[:, idx])
         self.comm.all_reduce_(delta)
         self._adjusted[:, int(true_class)] += self.update_strength * delta
