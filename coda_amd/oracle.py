@@ -18,6 +18,10 @@ class Oracle:
         self.device = dataset.device
         self.labels = dataset.labels
         assert self.labels is not None, "Oracle needs labels!"
+        # host-side copy: label serving is a per-step scalar fetch and
+        # must not synchronize the device (reference syncs every step,
+        # oracle.py:24)
+        self._labels_host = self.labels.cpu().tolist()
 
     def true_losses(self, preds: torch.Tensor) -> torch.Tensor:
         """Mean loss per model: (H, N, C) post-softmax scores -> (H,)."""
@@ -28,4 +32,4 @@ class Oracle:
                             reduction="none").view(H, N).mean(dim=1)
 
     def __call__(self, idx) -> int:
-        return int(self.labels[idx].item())
+        return int(self._labels_host[int(idx)])
