@@ -59,7 +59,7 @@ def _worker(rank, world, init_file, preds, labels, steps, q):
         dist.destroy_process_group()
 
 
-@pytest.mark.timeout(300)
+@pytest.mark.timeout(600)
 def test_sharded_equals_single(tmp_path):
     preds, labels = make_synthetic_task(H=7, N=200, C=4, seed=5)
     single = _single_trajectory(preds, labels)
@@ -74,7 +74,7 @@ def test_sharded_equals_single(tmp_path):
         p.start()
     results = {}
     for _ in range(2):
-        rank, res = q.get(timeout=240)
+        rank, res = q.get(timeout=400)
         results[rank] = res
     for p in procs:
         p.join(timeout=60)
@@ -115,7 +115,7 @@ def _worker_table(rank, world, init_file, preds, labels, steps, q):
         dist.destroy_process_group()
 
 
-@pytest.mark.timeout(300)
+@pytest.mark.timeout(600)
 def test_sharded_table_impl_equals_single(tmp_path):
     """The table-factored (v2) sharded path: same selections as the
     single-process table path."""
@@ -143,14 +143,14 @@ def test_sharded_table_impl_equals_single(tmp_path):
         p.start()
     results = {}
     for _ in range(2):
-        rank, res = q.get(timeout=240)
+        rank, res = q.get(timeout=400)
         results[rank] = res
     for p in procs:
         p.join(timeout=60)
     assert results[0] == single and results[1] == single
 
 
-@pytest.mark.timeout(300)
+@pytest.mark.timeout(600)
 def test_main_sharded_torchrun_cpu(tmp_path):
     """End-to-end `torchrun --nproc-per-node 2 main.py --sharded` on CPU
     (gloo): the full harness loop with the model axis sharded; rank 0
@@ -174,7 +174,7 @@ def test_main_sharded_torchrun_cpu(tmp_path):
          "--seeds", "1", "--sharded", "--device", "cpu",
          "--chunk-size", "32"],
         cwd=str(tmp_path), env=env, capture_output=True, text=True,
-        timeout=280)
+        timeout=500)
     assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
     conn = sqlite3.connect(str(tmp_path / "coda.sqlite"))
     n = conn.execute(
