@@ -45,7 +45,7 @@ def state_dict(selector) -> Dict[str, Any]:
             v = v.detach().cpu().clone()
         elif isinstance(v, list) and v and torch.is_tensor(v[0]):
             v = [t.detach().cpu().clone() for t in v]
-        elif isinstance(v, list):
+        elif isinstance(v, (list, tuple)) or type(v).__name__ == "SortedList":
             v = list(v)
         out[f] = v
     return out
@@ -64,6 +64,9 @@ def load_state_dict(selector, state: Dict[str, Any]):
             v = v.to(device=device, dtype=cur.dtype)
         elif isinstance(v, list) and v and torch.is_tensor(v[0]):
             v = [t.to(device) for t in v]
+        elif type(cur).__name__ == "SortedList":
+            from sortedcontainers import SortedList
+            v = SortedList(v)
         setattr(selector, f, v)
     # derived state that depends on the posterior
     if cls == "CODA":
@@ -71,9 +74,10 @@ def load_state_dict(selector, state: Dict[str, Any]):
         selector._tables_dirty = set()
         selector._posterior_version += 1
         selector._pbest_rows_cache = (-1, None)
-        selector._active_candidates = [
+        from sortedcontainers import SortedList
+        selector._active_candidates = SortedList(
             i for i in selector.unlabeled_idxs
-            if selector._disagreement_host[i]]
+            if selector._disagreement_host[i])
         selector.update_pi_hat()
     return selector
 

@@ -24,6 +24,7 @@ from __future__ import annotations
 import random
 
 import torch
+from sortedcontainers import SortedList
 
 from ..base import ModelSelector
 from ..parallel import Comm, get_comm
@@ -107,14 +108,15 @@ class CODA(ModelSelector):
         # host-side copy for the per-step candidate filter (indexing a GPU
         # tensor point-by-point would be one device sync per point)
         self._disagreement_host = self._disagreement.cpu().tolist()
-        # persistently maintained unlabeled-and-disagreeing candidate list
-        # (the mask is static; add_label removes one entry) - avoids an
-        # O(N) Python filter pass every step
-        self._active_candidates = [
-            i for i in range(self.N) if self._disagreement_host[i]]
+        # persistently maintained unlabeled-and-disagreeing candidates
+        # (the mask is static; add_label removes one entry). SortedList
+        # keeps the reference's ascending order with O(log N) removal -
+        # a plain list.remove is an O(N) scan per label.
+        self._active_candidates = SortedList(
+            i for i in range(self.N) if self._disagreement_host[i])
 
         self.labeled_idxs, self.labels = [], []
-        self.unlabeled_idxs = list(range(self.N))
+        self.unlabeled_idxs = SortedList(range(self.N))
         self.q_vals = []
         self.stochastic = False
         self.step = 0
@@ -214,8 +216,9 @@ class CODA(ModelSelector):
 
     def eig_batched(self):
         """EIG for every candidate point (reference: coda/coda.py:235-281)."""
-        candidate_ids = self._prefilter(self.unlabeled_idxs) or self.unlabeled_idxs
-        cand = torch.tensor(candidate_ids, device=self.device)
+        candidate_ids = self._prefilter(self.unlabeled_idxs) \
+            or self.unlabeled_idxs
+        cand = torch.tensor(list(candidate_ids), device=self.device)
 
         pbest_before = self._pbest_rows_before()            # (C, Hl)
         alpha_cc, beta_cc = ops.dirichlet_to_beta(self.dirichlets)
