@@ -184,3 +184,22 @@ def test_model_picker_reads_tuned_epsilon(task_dir, capsys):
         assert abs(sel.epsilon - 0.41) < 1e-9
     finally:
         os.chdir(cwd)
+
+
+def test_convert_task_roundtrip(task_dir):
+    import subprocess, sys
+    import torch
+    r = subprocess.run(
+        [sys.executable,
+         os.path.join(os.path.dirname(os.path.dirname(
+             os.path.abspath(__file__))), "scripts", "convert_task.py"),
+         "--task", "synthtask", "--data-dir", str(task_dir / "data"),
+         "--dtype", "bf16", "--out-dir", str(task_dir / "data_bf16")],
+        capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0, r.stderr
+    from coda_amd.datasets import Dataset
+    a = Dataset(str(task_dir / "data" / "synthtask.pt"), "cpu")
+    b = Dataset(str(task_dir / "data_bf16" / "synthtask.pt"), "cpu")
+    assert b.preds.dtype == torch.float32  # up-cast on load
+    torch.testing.assert_close(a.preds, b.preds, rtol=1e-2, atol=5e-3)
+    assert b.labels is not None
