@@ -138,20 +138,19 @@ def set_experiment(name: str) -> dict:
     conn = _connect()
     try:
         _ensure_schema(conn)
-        row = conn.execute(
-            "SELECT experiment_id, name FROM experiments WHERE name = ?",
-            (name,)).fetchone()
-        if row is None:
-            now = _now_ms()
-            cur = conn.execute(
-                "INSERT INTO experiments (name, artifact_location, "
-                "lifecycle_stage, creation_time, last_update_time) "
-                "VALUES (?, ?, 'active', ?, ?)",
-                (name, f"./mlruns/{name}", now, now))
-            conn.commit()
-            exp_id = cur.lastrowid
-        else:
-            exp_id = row[0]
+        now = _now_ms()
+        # race-safe against concurrent harness processes creating the same
+        # experiment (the task-parallel launcher runs task x method jobs
+        # concurrently): INSERT OR IGNORE, then read back
+        conn.execute(
+            "INSERT OR IGNORE INTO experiments (name, artifact_location, "
+            "lifecycle_stage, creation_time, last_update_time) "
+            "VALUES (?, ?, 'active', ?, ?)",
+            (name, f"./mlruns/{name}", now, now))
+        conn.commit()
+        exp_id = conn.execute(
+            "SELECT experiment_id FROM experiments WHERE name = ?",
+            (name,)).fetchone()[0]
         _EXPERIMENT = {"experiment_id": exp_id, "name": name}
         return _EXPERIMENT
     finally:

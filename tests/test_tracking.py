@@ -146,3 +146,34 @@ with tracking.start_run(run_name=f"w{{wid}}"):
     assert n_runs == 8
     assert n_metrics == 8 * 20
     assert all(s == "FINISHED" for s in statuses)
+
+
+def test_concurrent_set_experiment_same_name(db):
+    """Two processes racing to create the SAME experiment (the launcher
+    runs taskX-iid and taskX-coda concurrently) must both succeed - this
+    was a UNIQUE-constraint crash before INSERT OR IGNORE."""
+    import subprocess
+    import sys
+    import os as _os
+
+    repo = _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__)))
+    code = (
+        "import sys; sys.path.insert(0, {repo!r})\n"
+        "from coda_amd import tracking\n"
+        "tracking.set_tracking_uri('sqlite:///' + {db!r})\n"
+        "tracking.set_experiment('racetask')\n"
+        "with tracking.start_run(run_name='r' + sys.argv[1]):\n"
+        "    tracking.log_metric('m', 1.0, step=1)\n"
+    ).format(repo=repo, db=db)
+    procs = [subprocess.Popen([sys.executable, "-c", code, str(i)])
+             for i in range(6)]
+    rcs = [p.wait(timeout=200) for p in procs]
+    assert all(rc == 0 for rc in rcs), rcs
+    import sqlite3
+    conn = sqlite3.connect(db)
+    n_exp = conn.execute(
+        "SELECT COUNT(*) FROM experiments WHERE name='racetask'"
+    ).fetchone()[0]
+    n_runs = conn.execute("SELECT COUNT(*) FROM runs").fetchone()[0]
+    conn.close()
+    assert n_exp == 1 and n_runs == 6
