@@ -11,8 +11,11 @@ import os
 
 import torch
 
-_env = os.environ.get("CODA_AMD_DEBUG")
-DEBUG = _env == "1" if _env is not None else True
+# The reference ships its _DEBUG flag ON (coda/coda.py:10); here the
+# equivalent guards are OPT-IN (CODA_AMD_DEBUG=1): every NaN/Inf check is
+# a host-device synchronization (~5 per acquisition step measured, ~1 ms
+# of the step at the headline config). The test suites enable them.
+DEBUG = os.environ.get("CODA_AMD_DEBUG") == "1"
 
 
 class Ensemble:

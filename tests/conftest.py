@@ -1,5 +1,11 @@
+import os
+
 import pytest
 import torch
+
+# run the suites with the numeric guards ON (they are opt-in in
+# production because each check synchronizes the device)
+os.environ.setdefault("CODA_AMD_DEBUG", "1")
 
 
 def pytest_configure(config):
