@@ -1,9 +1,9 @@
 """Numeric sanitizers, ensemble consensus, debug plotting.
 
 Reference parity: coda/util.py:7-66. The `_check`/`_check_prob` runtime
-asserts are the de-facto correctness oracle for the math pipeline; they are
-gated by CODA_AMD_DEBUG (default on for CPU, off on GPU hot paths unless
-explicitly enabled, since each check synchronizes the device).
+asserts are the de-facto correctness oracle for the math pipeline; they
+are opt-in via CODA_AMD_DEBUG=1 (each check synchronizes the device; the
+test suites turn them on).
 """
 from __future__ import annotations
 
