@@ -751,6 +751,39 @@ pi_hat_delta_kernel(const T* __restrict__ preds,  // (H, N, C)
     out[n] = acc;
 }
 
+
+// pi marginal: out[c] = sum_n adjusted[n, c] / max(row_sums[n], 1e-12)
+// (reference coda/coda.py:229-233 without materializing the normalized
+// (N, C) matrix). One streaming pass: each thread owns fixed columns
+// c = tid + k*BLOCK (register accumulators, no LDS), blocks stride over
+// row slabs, one atomicAdd per (block, column) at the end.
+__global__ void __launch_bounds__(BLOCK)
+pi_marginal_kernel(const float* __restrict__ adjusted,  // (N, C)
+                   const float* __restrict__ row_sums,  // (N,)
+                   float* __restrict__ out,             // (C,) pre-zeroed
+                   long long N, int C) {
+    const int tid = threadIdx.x;
+    const int ncols = (C + BLOCK - 1) / BLOCK;
+    float acc[8];  // supports C <= 8*BLOCK = 2048
+    for (int k = 0; k < ncols && k < 8; ++k) acc[k] = 0.f;
+    const long long rows_per_block =
+        (N + gridDim.x - 1) / gridDim.x;
+    const long long n0 = (long long)blockIdx.x * rows_per_block;
+    const long long n1 = min(n0 + rows_per_block, N);
+    for (long long n = n0; n < n1; ++n) {
+        const float inv = 1.0f / fmaxf(row_sums[n], 1e-12f);
+        const float* row = adjusted + n * C;
+        for (int k = 0; k < ncols && k < 8; ++k) {
+            const int c = tid + k * BLOCK;
+            if (c < C) acc[k] += row[c] * inv;
+        }
+    }
+    for (int k = 0; k < ncols && k < 8; ++k) {
+        const int c = tid + k * BLOCK;
+        if (c < C) atomicAdd(out + c, acc[k]);
+    }
+}
+
 }  // namespace
 
 // ---------------------------------------------------------------------------
@@ -1153,6 +1186,24 @@ torch::Tensor pi_hat_delta(torch::Tensor preds, torch::Tensor cls) {
     return out;
 }
 
+
+torch::Tensor pi_marginal(torch::Tensor adjusted, torch::Tensor row_sums) {
+    check_f32_cuda(adjusted, "adjusted");
+    check_f32_cuda(row_sums, "row_sums");
+    const long long N = adjusted.size(0);
+    const int C = adjusted.size(1);
+    TORCH_CHECK(C <= 8 * BLOCK, "C too large for pi_marginal kernel");
+    auto out = torch::zeros({C}, adjusted.options());
+    const int blocks = 512;
+    auto stream = c10::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(pi_marginal_kernel, dim3(blocks), dim3(BLOCK), 0,
+                       stream.stream(), adjusted.data_ptr<float>(),
+                       row_sums.data_ptr<float>(), out.data_ptr<float>(),
+                       N, C);
+    C10_HIP_CHECK(hipGetLastError());
+    return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "coda_amd fused gfx950 kernels";
     m.def("pbest_from_beta", &pbest_from_beta,
@@ -1181,4 +1232,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "v2: one class row's H*2 hypothetical curves (EG, log2 cdf)");
     m.def("pi_hat_delta", &pi_hat_delta,
           "rank-1 pi_hat increment: sum_h preds[h, :, cls_h]");
+    m.def("pi_marginal", &pi_marginal,
+          "pi[c] = sum_n adjusted[n,c]/clamp(rowsum[n]) in one pass");
 }

@@ -177,10 +177,14 @@ class CODA(ModelSelector):
         (identical math to the reference's pi_hat_xi.sum(0) route,
         coda/coda.py:229-233). Per-item rows are built on demand
         (candidate gathers in eig_batched; the pi_hat_xi property)."""
-        inv = 1.0 / self._row_sums.clamp_min(1e-12)
-        # row-vector @ matrix streams adjusted once; mv on the transposed
-        # view is 5x slower on ROCm (strided column reduction)
-        pi = inv @ self._adjusted
+        if (self._adjusted.is_cuda and self.C <= 2048
+                and ops.hip_available()):
+            pi = ops._ext.pi_marginal(self._adjusted, self._row_sums)
+        else:
+            inv = 1.0 / self._row_sums.clamp_min(1e-12)
+            # row-vector @ matrix streams adjusted once; transposed mv is
+            # 5x slower on ROCm (strided column reduction)
+            pi = inv @ self._adjusted
         self.pi_hat = pi / pi.sum()
         self._pi_xi_cache = None
 

@@ -309,3 +309,13 @@ class TestTableFusionKernels:
         # bf16 storage
         got16 = ops.pi_hat_delta(preds.to(torch.bfloat16).contiguous(), cls)
         torch.testing.assert_close(got16, want, rtol=2e-2, atol=1e-2)
+
+    def test_pi_marginal_kernel(self, dev):
+        from coda_amd import ops
+        g = torch.Generator().manual_seed(45)
+        N, C = 777, 133
+        A = (torch.rand(N, C, generator=g) + 0.01).to(dev)
+        rs = A.sum(-1)
+        got = ops._ext.pi_marginal(A, rs)
+        want = (A / rs.clamp_min(1e-12).unsqueeze(-1)).sum(0)
+        torch.testing.assert_close(got, want, rtol=1e-4, atol=1e-4)
