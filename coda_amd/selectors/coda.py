@@ -192,11 +192,13 @@ class CODA(ModelSelector):
     def pi_hat_xi(self):
         """(N, C) per-item class posterior (materialized on demand)."""
         if self._pi_xi_cache is None:
-            self._pi_xi_cache = self._adjusted /                 self._row_sums.clamp_min(1e-12).unsqueeze(-1)
+            rs = self._row_sums.clamp_min(1e-12).unsqueeze(-1)
+            self._pi_xi_cache = self._adjusted / rs
         return self._pi_xi_cache
 
     def _pi_xi_rows(self, ids):
-        return self._adjusted[ids] /             self._row_sums[ids].clamp_min(1e-12).unsqueeze(-1)
+        rs = self._row_sums[ids].clamp_min(1e-12).unsqueeze(-1)
+        return self._adjusted[ids] / rs
 
     # ------------------------------------------------------------------
     def _pbest_rows_before(self):
@@ -291,7 +293,7 @@ class CODA(ModelSelector):
             ens_sum = self.dataset.preds.sum(dim=0)
             self.comm.all_reduce_(ens_sum)
             ent = ops.entropy_acquisition(ens_sum / self.H)
-            q_vals = ent[torch.tensor(cand, device=self.device)]
+            q_vals = ent[torch.tensor(list(cand), device=self.device)]
         else:
             raise NotImplementedError(self.q)
 
