@@ -33,7 +33,7 @@ def fill_csr(cls):
     hv_s.copy_(hv); off_s.copy_(off)
 
 def body():
-    ES = ops._ext.es_build(s_base, delta, hv_s, off_s, w)
+    ES = ops._ext.es_build(s_base, delta, hv_s, off_s, w, False)
     M = torch.bmm(ES, EG.reshape(C, 2 * H, 256).transpose(1, 2))
     h_after = ops._ext.eig_assemble_k(M, cls_s, pi, pb0, m0)
     return H0_s - (pixi_s * h_after).sum(-1)
