@@ -319,3 +319,33 @@ class TestTableFusionKernels:
         got = ops._ext.pi_marginal(A, rs)
         want = (A / rs.clamp_min(1e-12).unsqueeze(-1)).sum(0)
         torch.testing.assert_close(got, want, rtol=1e-4, atol=1e-4)
+
+
+class TestWidePoolGPU:
+    def test_hchunked_pbest_gpu(self, dev):
+        """H beyond the kernel LDS budget routes through the chunked eager
+        passes on GPU too."""
+        from coda_amd import ops
+        a, b = _rand_betas(rows=3, H=2100, seed=61, lo=1.0, hi=20.0,
+                           device=dev)
+        p = ops.pbest_from_beta(a, b)
+        assert torch.isfinite(p).all()
+        np.testing.assert_allclose(p.sum(-1).cpu().numpy(), 1.0, atol=1e-3)
+
+    def test_coda_wide_pool_gpu(self, dev):
+        """End-to-end CODA on a 2100-model pool (wide-H fallback for the
+        class-row posterior + v2 tables for EIG)."""
+        import bench
+        from coda_amd import CODA, Oracle
+        from coda_amd.datasets import Dataset
+        from coda_amd.options import LOSS_FNS
+        preds, labels = bench.synth_preds(list(range(2100)), 500, 10, dev)
+        ds = Dataset.from_tensors(preds, labels, dev)
+        oracle = Oracle(ds, LOSS_FNS["acc"])
+        random.seed(0); torch.manual_seed(0)
+        sel = CODA(ds, chunk_size=64, prefilter_n=64)
+        for _ in range(2):
+            i, q = sel.get_next_item_to_label()
+            sel.add_label(i, oracle(int(i)), q)
+        p = sel.get_pbest()
+        assert p.shape == (2100,) and torch.isfinite(p).all()
