@@ -24,6 +24,20 @@
 // log-space clamp +-80, entropy clamp 1e-12).
 //
 // Workgroup = 256 threads = 4 waves = 4 independent rows.
+//
+// Kernel inventory (host bindings at the bottom):
+//   pbest_kernel           fused P(best) over (R, H) rows (v1)
+//   eig_hyp_kernel         fused hypothetical P(best) + entropy (v1 EIG)
+//   pbest/eig_phase1/2     two-phase variants with the model-axis
+//                          all-reduce seam between passes (sharded v1)
+//   es_build[_gathered]    v2: CSR-bucketed log-cdf scatter + exp2 +
+//                          trapz weights -> the GEMM B operand (bf16/f32)
+//   eig_assemble_kernel    v2: variant select + normalize + log2 entropy
+//   eig_totals/entropy     v2 sharded split around the normalizer
+//                          all-reduce
+//   beta_row_tables_kernel v2: per-class incremental table refresh
+//   pi_hat_delta_kernel    rank-1 posterior-marginal increment
+//   pi_marginal_kernel     streaming scaled column sum for pi_hat
 
 #include <torch/extension.h>
 #include <hip/hip_runtime.h>

@@ -195,6 +195,19 @@ def create_app(dataset, method: str = "coda", oracle=None, seed: int = 0,
     def pbest():
         return {"pbest": session.pbest()}
 
+    @app.get("/pbest.png")
+    def pbest_png():
+        """Live P(best) bar chart (the reference demo's probability
+        chart, demo/app.py:212-255)."""
+        import io
+        from fastapi.responses import Response
+        from ..util import plot_bar
+        img = plot_bar(session.pbest(), title="P(model is best)",
+                       xlabel="model", ylabel="probability")
+        buf = io.BytesIO()
+        img.save(buf, format="PNG")
+        return Response(content=buf.getvalue(), media_type="image/png")
+
     @app.get("/state")
     def state():
         return session.state()

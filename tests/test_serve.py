@@ -80,3 +80,11 @@ def test_synthetic_builder(tmp_path):
     t = torch.load(str(out), weights_only=True)
     assert t.shape == (4, 60, 5)
     assert (tmp_path / "demo_labels.pt").exists()
+
+
+def test_pbest_png(client):
+    c, _ = client
+    r = c.get("/pbest.png")
+    assert r.status_code == 200
+    assert r.headers["content-type"] == "image/png"
+    assert r.content[:4] == b"\x89PNG"
