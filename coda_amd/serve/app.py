@@ -156,14 +156,26 @@ refresh();
 
 
 def create_app(dataset, method: str = "coda", oracle=None, seed: int = 0,
+               images_dir: str = None, class_names=None,
                **selector_kwargs):
-    from fastapi import Body, FastAPI
+    """images_dir: optional folder whose sorted image files correspond to
+    point indices (the order coda_amd.serve.build_predictions uses) - the
+    reference demo's image-quiz flow (demo/app.py:137-172)."""
+    from fastapi import Body, FastAPI, HTTPException
     from fastapi.responses import HTMLResponse
 
     session = SelectorSession(dataset, method=method, seed=seed,
                               oracle=oracle, **selector_kwargs)
     app = FastAPI(title="coda_amd serving")
     app.state.session = session
+
+    image_paths = None
+    if images_dir:
+        from .build_predictions import list_images
+        image_paths, _, detected = list_images(images_dir)
+        if class_names is None:
+            class_names = detected
+    app.state.class_names = class_names
 
     @app.get("/", response_class=HTMLResponse)
     def index():
@@ -210,7 +222,21 @@ def create_app(dataset, method: str = "coda", oracle=None, seed: int = 0,
 
     @app.get("/state")
     def state():
-        return session.state()
+        st = session.state()
+        if class_names:
+            st["class_names"] = list(class_names)
+        return st
+
+    @app.get("/image/{index}")
+    def image(index: int):
+        """The point's image (sorted-order correspondence with the
+        prediction tensor built by build_predictions)."""
+        from fastapi.responses import FileResponse
+        if image_paths is None:
+            raise HTTPException(404, "no images_dir configured")
+        if not 0 <= index < len(image_paths):
+            raise HTTPException(404, "index out of range")
+        return FileResponse(image_paths[index])
 
     return app
 
@@ -226,6 +252,8 @@ def main():
     ap.add_argument("--task", required=True)
     ap.add_argument("--data-dir", default="data")
     ap.add_argument("--method", default="coda")
+    ap.add_argument("--images-dir", default=None,
+                    help="serve point images from this folder")
     ap.add_argument("--device", default=None)
     ap.add_argument("--host", default="127.0.0.1")
     ap.add_argument("--port", type=int, default=7860)
@@ -235,7 +263,8 @@ def main():
     import os
     ds = Dataset(os.path.join(args.data_dir, args.task + ".pt"), device)
     oracle = Oracle(ds, LOSS_FNS["acc"]) if ds.labels is not None else None
-    app = create_app(ds, method=args.method, oracle=oracle)
+    app = create_app(ds, method=args.method, oracle=oracle,
+                     images_dir=args.images_dir)
     uvicorn.run(app, host=args.host, port=args.port)
 
 

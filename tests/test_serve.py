@@ -88,3 +88,23 @@ def test_pbest_png(client):
     assert r.status_code == 200
     assert r.headers["content-type"] == "image/png"
     assert r.content[:4] == b"\x89PNG"
+
+
+def test_image_endpoint(tmp_path):
+    from PIL import Image
+    import numpy as np
+    d = tmp_path / "imgs" / "cat"
+    d.mkdir(parents=True)
+    for i in range(3):
+        Image.fromarray(
+            (np.random.rand(8, 8, 3) * 255).astype("uint8")).save(
+            d / f"im{i}.png")
+    preds, labels = make_synthetic_task(H=3, N=3, C=2, seed=8)
+    ds = Dataset.from_tensors(preds, labels, "cpu")
+    app = create_app(ds, method="iid", images_dir=str(tmp_path / "imgs"))
+    c = TestClient(app)
+    r = c.get("/image/1")
+    assert r.status_code == 200
+    assert r.content[:4] == b"\x89PNG"
+    assert c.get("/image/99").status_code == 404
+    assert c.get("/state").json()["class_names"] == ["cat"]
