@@ -210,7 +210,8 @@ def table_update_rows(tables: EigTables, alpha_cc: torch.Tensor,
     tables.EG[rows] = sub.EG
     if tables.eg16 is not None:
         H = alpha_cc.shape[0]
-        tables.eg16[rows] = sub.EG.reshape(rows.numel(), 2 * H, -1)             .to(torch.bfloat16)
+        tables.eg16[rows] = sub.EG.reshape(
+            rows.numel(), 2 * H, -1).to(torch.bfloat16)
     tables.delta[rows] = sub.delta
     tables.s_base[rows] = sub.s_base
     return tables
