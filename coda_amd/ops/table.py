@@ -239,8 +239,12 @@ def eig_chunk_table_sharded(tables: EigTables, s_base_all: torch.Tensor,
                             pi_hat: torch.Tensor,
                             pi_hat_xi_chunk: torch.Tensor,
                             mixture0_local: torch.Tensor,
-                            H_before: torch.Tensor, comm) -> torch.Tensor:
-    """(B,) EIG with local tables; identical on every rank."""
+                            H_before: torch.Tensor, comm,
+                            hsizes=None) -> torch.Tensor:
+    """(B,) EIG with local tables; identical on every rank.
+
+    hsizes: per-rank model counts (comm.shard_sizes(H_global)) - lets the
+    per-chunk gathers skip their size-exchange collectives."""
     EG, delta, w = tables.EG, tables.delta, tables.weights
     C, Hl, _, P = EG.shape
     B = cls_local.shape[0]
@@ -249,8 +253,9 @@ def eig_chunk_table_sharded(tables: EigTables, s_base_all: torch.Tensor,
     flat = delta.permute(1, 0, 2).reshape(Hl * C, P)
     sel = flat[(torch.arange(Hl, device=cls_l.device) * C).unsqueeze(0)
                + cls_l].contiguous()                         # (B, Hl, P)
-    sel_all = comm.all_gather_cat(sel, dim=1).contiguous()   # (B, H, P)
-    cls_all = comm.all_gather_cat(cls_l, dim=1)              # (B, H)
+    sel_all = comm.all_gather_cat(sel, dim=1,
+                                  sizes=hsizes).contiguous()  # (B, H, P)
+    cls_all = comm.all_gather_cat(cls_l, dim=1, sizes=hsizes)  # (B, H)
     Hg = cls_all.shape[1]
 
     import coda_amd.ops as O

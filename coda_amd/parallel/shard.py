@@ -50,18 +50,29 @@ class Comm:
             dist.broadcast(t, src=src)
         return t
 
-    def all_gather_cat(self, t: torch.Tensor, dim: int = 0) -> torch.Tensor:
+    def shard_sizes(self, total: int):
+        """Per-rank shard sizes under strided sharding of `total` items."""
+        return [len(range(r, total, self.world))
+                for r in range(self.world)]
+
+    def all_gather_cat(self, t: torch.Tensor, dim: int = 0,
+                       sizes=None) -> torch.Tensor:
         """Gather per-rank shards and concatenate along `dim`.
 
         Shards may have unequal sizes along `dim` (strided H sharding when
-        H % world != 0).
+        H % world != 0). Pass `sizes` (per-rank sizes along `dim`, e.g.
+        from shard_sizes) to skip the size-exchange collective - one
+        round-trip saved per call, which matters for the per-chunk
+        gathers inside the EIG loop.
         """
         if self.world == 1:
             return t
-        sizes = [torch.zeros(1, dtype=torch.long, device=t.device)
-                 for _ in range(self.world)]
-        dist.all_gather(sizes, torch.tensor([t.shape[dim]], device=t.device))
-        sizes = [int(s.item()) for s in sizes]
+        if sizes is None:
+            sz = [torch.zeros(1, dtype=torch.long, device=t.device)
+                  for _ in range(self.world)]
+            dist.all_gather(sz, torch.tensor([t.shape[dim]],
+                                             device=t.device))
+            sizes = [int(x.item()) for x in sz]
         maxd = max(sizes)
         padded_shape = list(t.shape)
         padded_shape[dim] = maxd

@@ -260,7 +260,8 @@ class CODA(ModelSelector):
                 from ..ops import table as tops
                 eig = tops.eig_chunk_table_sharded(
                     tables, s_base_all, chunk_classes, pbest_before,
-                    self.pi_hat, pi_xi, mixture0, H_before, self.comm)
+                    self.pi_hat, pi_xi, mixture0, H_before, self.comm,
+                    hsizes=self.comm.shard_sizes(self.H))
             elif self.comm.is_distributed:
                 eig = shops.eig_chunk_sharded(
                     alpha_cc, beta_cc, chunk_classes, pbest_before,
