@@ -140,3 +140,12 @@ def test_eps_gridsearch(pipeline_dir):
     assert r.returncode == 0, r.stdout + r.stderr
     best = json.loads((d / "best_epsilons.json").read_text())
     assert "taskx" in best
+
+
+def test_latency_report(pipeline_dir):
+    d, env = pipeline_dir
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "analysis", "latency.py")],
+        cwd=str(d), env=env, capture_output=True, text=True, timeout=400)
+    assert r.returncode == 0, r.stderr
+    assert "steps/s" in r.stdout and "taskx" in r.stdout
