@@ -115,3 +115,82 @@ def test_oracle_losses_parity(ref_modules):
                   loss_fn=LOSS_FNS["acc"])
     torch.testing.assert_close(ref.true_losses(preds),
                                mine.true_losses(preds))
+
+
+def test_coda_prefilter_and_ablation_parity(ref_modules):
+    """Prefiltered candidates + no-diag-prior ablation match the
+    reference's trajectories too."""
+    ref_coda, _ = ref_modules
+    from coda_amd.datasets import Dataset
+    from coda_amd import CODA
+
+    preds, labels = _mk(seed=6, H=6, N=250, C=4)
+    rds = _RefDS()
+    rds.preds, rds.labels, rds.device = preds.clone(), labels.clone(), \
+        torch.device("cpu")
+
+    random.seed(3); torch.manual_seed(3)
+    ref = ref_coda.CODA(rds, prefilter_n=40, disable_diag_prior=True)
+    random.seed(3); torch.manual_seed(3)
+    mine = CODA(Dataset.from_tensors(preds, labels, "cpu"),
+                prefilter_n=40, disable_diag_prior=True)
+
+    for m in range(3):
+        random.seed(50 + m)
+        with contextlib.redirect_stderr(io.StringIO()):
+            ir, qr = ref.get_next_item_to_label()
+        random.seed(50 + m)
+        im, qm = mine.get_next_item_to_label()
+        assert int(ir) == int(im), (m, int(ir), int(im))
+        y = int(labels[int(ir)])
+        ref.add_label(int(ir), y, qr)
+        mine.add_label(int(im), y, qm)
+
+
+def test_baseline_acquisition_distributions_parity(ref_modules):
+    """ActiveTesting / VMA acquisition DISTRIBUTIONS equal the
+    reference's (selection draws consume the same seeded RNG, so equal
+    distributions => equal trajectories)."""
+    sys.path.insert(0, REF)
+    try:
+        from coda.baselines.activetesting import ActiveTesting as RefAT
+        from coda.baselines.vma import VMA as RefVMA
+        from coda.options import LOSS_FNS as REF_LOSS
+    finally:
+        sys.path.remove(REF)
+    from coda_amd.datasets import Dataset
+    from coda_amd.baselines import ActiveTesting, VMA
+    from coda_amd.options import LOSS_FNS
+
+    preds, labels = _mk(seed=7, H=5, N=120, C=4)
+    rds = _RefDS()
+    rds.preds, rds.labels, rds.device = preds.clone(), labels.clone(), \
+        torch.device("cpu")
+    ds = Dataset.from_tensors(preds, labels, "cpu")
+
+    # ActiveTesting: same normalized scores over the unlabeled pool
+    rat = RefAT(rds, REF_LOSS["acc"])
+    mat = ActiveTesting(ds, LOSS_FNS["acc"])
+    pi_y = rds.preds.mean(0)
+    cls = rds.preds.argmax(2)
+    y_star = pi_y[torch.arange(120), cls]
+    ref_scores = (1 - y_star).sum(0)
+    ref_scores = ref_scores / ref_scores.sum()
+    mine_scores = mat._acq_mass / mat._acq_mass.sum()
+    torch.testing.assert_close(mine_scores, ref_scores, rtol=1e-5,
+                               atol=1e-7)
+    # identical seeded draw
+    random.seed(11)
+    ri, rq = rat.get_next_item_to_label()
+    random.seed(11)
+    mi, mq = mat.get_next_item_to_label()
+    assert int(ri) == int(mi) and abs(rq - mq) < 1e-6
+
+    # VMA: same pairwise-difference masses (ours via the sorted identity)
+    rv = RefVMA(rds, REF_LOSS["acc"])
+    mv = VMA(ds, LOSS_FNS["acc"])
+    random.seed(12)
+    ri, rq = rv.get_next_item_to_label()
+    random.seed(12)
+    mi, mq = mv.get_next_item_to_label()
+    assert int(ri) == int(mi) and abs(rq - mq) < 1e-6
