@@ -74,6 +74,7 @@ def load_state_dict(selector, state: Dict[str, Any]):
         selector._tables_dirty = set()
         selector._posterior_version += 1
         selector._pbest_rows_cache = (-1, None)
+        selector._label_graph = None  # graph buffers alias replaced state
         from sortedcontainers import SortedList
         selector._active_candidates = SortedList(
             i for i in selector.unlabeled_idxs

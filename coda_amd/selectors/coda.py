@@ -67,6 +67,14 @@ class CODA(ModelSelector):
         self._tables_dirty = set()   # class rows touched since last build
         self._posterior_version = 0  # bumped by add_label
         self._pbest_rows_cache = (-1, None)
+        # hipGraph capture of the post-label update pipeline (Dirichlet row
+        # update -> rank-1 pi_hat -> table row refresh -> posterior rows):
+        # ~20 static-shape launches replayed as one graph. Auto on
+        # single-device GPU; CODA_AMD_NO_GRAPH=1 disables.
+        self._label_graph = None
+        self._use_label_graph = (
+            self.device.type == "cuda" and not self.comm.is_distributed
+            and os.environ.get("CODA_AMD_NO_GRAPH") != "1")
 
         # hyperparams (reference names: coda/coda.py:189-190)
         self.prior_strength = 1.0 - alpha
@@ -322,14 +330,104 @@ class CODA(ModelSelector):
             return cand[idx_local], float(q_vals[idx_local])
         return cand[int(bi)], bv
 
+    # -- hipGraph label-update pipeline --------------------------------
+    def _label_update_body(self):
+        """The tensor part of add_label as static-shape ops on the static
+        input buffers self._g_idx / self._g_y (both (1,) int64). Writes
+        tables in place and leaves the updated posterior rows in
+        self._g_rows. Captured once into a hipGraph; identical math to
+        the eager path (it IS the eager path, replayed)."""
+        lr = self.update_strength
+        idx_t, y_t = self._g_idx, self._g_y
+        col = self.classes.index_select(1, idx_t).squeeze(1)     # (Hl,)
+        onehot = torch.nn.functional.one_hot(col, self.C).to(
+            self.dirichlets.dtype)
+        self.dirichlets.index_add_(1, y_t, (lr * onehot).unsqueeze(1))
+        delta = ops.pi_hat_delta(self.dataset.preds,
+                                 col) * lr                       # (N,)
+        self._adjusted.index_add_(1, y_t, delta.unsqueeze(1))
+        self._row_sums += delta
+        pi = ops._ext.pi_marginal(self._adjusted, self._row_sums)             if (self.C <= 2048 and ops.hip_available()) else             (1.0 / self._row_sums.clamp_min(1e-12)) @ self._adjusted
+        self._g_pi.copy_(pi / pi.sum())
+        # refresh the labeled class's table row (v2 tables)
+        if self._tables is not None:
+            row = self.dirichlets.index_select(1, y_t).squeeze(1)  # (Hl,C)
+            a_col = row.gather(1, y_t.view(1, 1).expand(self.Hl, 1))                 .squeeze(1).contiguous()
+            b_col = (row.sum(1) - a_col).contiguous()
+            eg, lc = ops._ext.beta_row_tables(a_col, b_col, 1.0)
+            t = self._tables
+            t.EG.index_copy_(0, y_t, eg.unsqueeze(0))
+            if t.eg16 is not None:
+                t.eg16.index_copy_(
+                    0, y_t, eg.reshape(1, 2 * self.Hl, -1)
+                    .to(torch.bfloat16))
+            t.delta.index_copy_(0, y_t, (lc[:, 1] - lc[:, 0]).unsqueeze(0))
+            t.s_base.index_copy_(0, y_t, lc[:, 0].sum(0).unsqueeze(0))
+        # posterior rows for the next acquisition / get_pbest
+        alpha_cc, beta_cc = ops.dirichlet_to_beta(self.dirichlets)
+        rows = ops.pbest_from_beta(alpha_cc.t().contiguous(),
+                                   beta_cc.t().contiguous(),
+                                   self.num_points)
+        self._g_rows.copy_(rows)
+
+    def _graphed_add_label(self, idx: int, true_class: int):
+        if self._label_graph is None:
+            self._g_idx = torch.zeros(1, dtype=torch.long,
+                                      device=self.device)
+            self._g_y = torch.zeros(1, dtype=torch.long, device=self.device)
+            self._g_pi = torch.empty_like(self.pi_hat)
+            self._g_rows = torch.empty(self.C, self.Hl, device=self.device)
+            self._g_idx.fill_(idx)
+            self._g_y.fill_(true_class)
+            # warmup on a side stream (required before capture), then
+            # capture a second execution
+            # warmup executes the body for real (THIS label's update) on
+            # a side stream, as stream capture requires; the capture pass
+            # then only RECORDS the ops (no execution, no state change),
+            # so the first label must not additionally replay.
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                self._label_update_body()
+            torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self._label_update_body()
+            self._label_graph = g
+            self.pi_hat = self._g_pi
+            self._pbest_rows_cache = (self._posterior_version + 1,
+                                      self._g_rows)
+            return
+        self._g_idx.fill_(idx)
+        self._g_y.fill_(true_class)
+        self._label_graph.replay()
+        self.pi_hat = self._g_pi
+        self._pbest_rows_cache = (self._posterior_version + 1,
+                                  self._g_rows)
+
     def add_label(self, idx, true_class, selection_prob):
         """Posterior update (K13) + incremental pi_hat refresh.
 
         Reference semantics (coda/coda.py:315-323); the pi_hat refresh is
         the exact rank-1 increment (only Dirichlet row `true_class`
-        moved), all-reduced over shards.
-        """
+        moved), all-reduced over shards. On single-device GPU the whole
+        tensor pipeline replays as one hipGraph."""
         idx = int(idx)
+        if (self._use_label_graph and self._tables is not None
+                and not self._tables_dirty):
+            self._graphed_add_label(idx, int(true_class))
+            self._posterior_version += 1
+            self._pi_xi_cache = None
+            self.labeled_idxs.append(idx)
+            self.labels.append(int(true_class))
+            self.q_vals.append(selection_prob)
+            self.unlabeled_idxs.remove(idx)
+            try:
+                self._active_candidates.remove(idx)
+            except ValueError:
+                pass
+            return
         onehot = torch.nn.functional.one_hot(
             self.classes[:, idx], self.C).to(self.dirichlets.dtype)  # (Hl, C)
         self.dirichlets[:, int(true_class)] += self.update_strength * onehot

@@ -349,3 +349,38 @@ class TestWidePoolGPU:
             sel.add_label(i, oracle(int(i)), q)
         p = sel.get_pbest()
         assert p.shape == (2100,) and torch.isfinite(p).all()
+
+
+class TestLabelGraph:
+    def test_graphed_trajectory_equals_eager(self, dev, monkeypatch):
+        """hipGraph label-update replay == the eager pipeline: identical
+        selections and P(best) over 6 steps."""
+        import bench
+        from coda_amd import CODA, Oracle
+        from coda_amd.datasets import Dataset
+        from coda_amd.options import LOSS_FNS
+
+        preds, labels = bench.synth_preds(list(range(12)), 600, 8, dev)
+        ds = Dataset.from_tensors(preds, labels, dev)
+        oracle = Oracle(ds, LOSS_FNS["acc"])
+
+        def run(no_graph):
+            if no_graph:
+                monkeypatch.setenv("CODA_AMD_NO_GRAPH", "1")
+            else:
+                monkeypatch.delenv("CODA_AMD_NO_GRAPH", raising=False)
+            random.seed(0); torch.manual_seed(0)
+            sel = CODA(ds, chunk_size=64, prefilter_n=64)
+            traj = []
+            for _ in range(6):
+                i, q = sel.get_next_item_to_label()
+                sel.add_label(i, oracle(int(i)), q)
+                traj.append((int(i), int(sel.get_best_model_prediction())))
+            assert (sel._label_graph is None) == no_graph
+            return traj, sel.get_pbest().cpu(), sel.pi_hat.cpu()
+
+        t_e, p_e, pi_e = run(True)
+        t_g, p_g, pi_g = run(False)
+        assert t_e == t_g
+        torch.testing.assert_close(p_e, p_g, rtol=1e-5, atol=1e-7)
+        torch.testing.assert_close(pi_e, pi_g, rtol=1e-5, atol=1e-7)
