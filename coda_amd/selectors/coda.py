@@ -72,8 +72,11 @@ class CODA(ModelSelector):
         # ~20 static-shape launches replayed as one graph. Auto on
         # single-device GPU; CODA_AMD_NO_GRAPH=1 disables.
         self._label_graph = None
+        # (debug guards synchronize per op and cannot run under stream
+        # capture - graph replay is a production-mode path)
         self._use_label_graph = (
             self.device.type == "cuda" and not self.comm.is_distributed
+            and not DEBUG
             and os.environ.get("CODA_AMD_NO_GRAPH") != "1")
 
         # hyperparams (reference names: coda/coda.py:189-190)

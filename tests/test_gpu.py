@@ -364,6 +364,9 @@ class TestLabelGraph:
         ds = Dataset.from_tensors(preds, labels, dev)
         oracle = Oracle(ds, LOSS_FNS["acc"])
 
+        import coda_amd.selectors.coda as coda_mod
+        monkeypatch.setattr(coda_mod, "DEBUG", False)
+
         def run(no_graph):
             if no_graph:
                 monkeypatch.setenv("CODA_AMD_NO_GRAPH", "1")
