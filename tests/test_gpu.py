@@ -364,8 +364,13 @@ class TestLabelGraph:
         ds = Dataset.from_tensors(preds, labels, dev)
         oracle = Oracle(ds, LOSS_FNS["acc"])
 
+        # production mode for both the selector gate and the op guards
         import coda_amd.selectors.coda as coda_mod
+        import coda_amd.ops as ops_mod
+        import coda_amd.util as util_mod
         monkeypatch.setattr(coda_mod, "DEBUG", False)
+        monkeypatch.setattr(ops_mod, "DEBUG", False)
+        monkeypatch.setattr(util_mod, "DEBUG", False)
 
         def run(no_graph):
             if no_graph:
