@@ -28,10 +28,14 @@ def make(impl, gemm):
         sel.get_best_model_prediction()
     return sel
 
+os.environ["CODA_AMD_NO_GRAPH"] = "1"
+nograph = make("table", "bf16")
+os.environ.pop("CODA_AMD_NO_GRAPH")
 variants = {
     "fused": make("fused", "fp32"),
     "table-fp32": make("table", "fp32"),
     "table-bf16": make("table", "bf16"),
+    "bf16-nograph": nograph,
 }
 res = {k: [] for k in variants}
 for rnd in range(6):
