@@ -538,7 +538,6 @@ eig_phase2_kernel(const float* __restrict__ alpha_t,
 //       v = [cls(b,h)==c], normalize over h, and emit the log2-entropy
 //       H_after[b,c] of the updated P(best) mixture.
 // ---------------------------------------------------------------------------
-#define ES_BPW 4  // candidates per wave: share the class's base row
 template <typename TOUT>
 __global__ void __launch_bounds__(BLOCK)
 es_build_kernel(const float* __restrict__ s_base,   // (C, P)
@@ -548,35 +547,33 @@ es_build_kernel(const float* __restrict__ s_base,   // (C, P)
                 const float* __restrict__ w,        // (P,)
                 TOUT* __restrict__ es,              // (C, B, P)
                 int B, int C, int H) {
-    const int nbq = (B + ES_BPW - 1) / ES_BPW;
     const int r = blockIdx.x * ROWS_PER_BLOCK + (threadIdx.x >> 6);
-    if (r >= C * nbq) return;
-    const int c = r / nbq, b0 = (r - c * nbq) * ES_BPW;
+    if (r >= B * C) return;
+    const int b = r / C, c = r - b * C;
     const int lane = threadIdx.x & 63;
     const int p0 = lane * PTS_PER_LANE;
 
+    float acc[PTS_PER_LANE];
     const float4 sb = *reinterpret_cast<const float4*>(
         s_base + (size_t)c * P_POINTS + p0);
-    const float4 wv = *reinterpret_cast<const float4*>(w + p0);
+    acc[0] = sb.x; acc[1] = sb.y; acc[2] = sb.z; acc[3] = sb.w;
     const size_t dbase = (size_t)c * H * P_POINTS + p0;
-    for (int b = b0; b < min(b0 + ES_BPW, B); ++b) {
-        float acc[PTS_PER_LANE] = {sb.x, sb.y, sb.z, sb.w};
-        // CSR over "models whose argmax class on candidate b is c":
-        // avg H/C iterations instead of an H-long load chain
-        const int k0 = offsets[(size_t)b * (C + 1) + c];
-        const int k1 = offsets[(size_t)b * (C + 1) + c + 1];
-        for (int k = k0; k < k1; ++k) {
-            const int h = hvals[(size_t)b * H + k];
-            const float4 d = *reinterpret_cast<const float4*>(
-                delta + dbase + (size_t)h * P_POINTS);
-            acc[0] += d.x; acc[1] += d.y; acc[2] += d.z; acc[3] += d.w;
-        }
-        TOUT* dst = es + ((size_t)c * B + b) * P_POINTS + p0;
-        dst[0] = (TOUT)(exp2f(acc[0]) * wv.x);
-        dst[1] = (TOUT)(exp2f(acc[1]) * wv.y);
-        dst[2] = (TOUT)(exp2f(acc[2]) * wv.z);
-        dst[3] = (TOUT)(exp2f(acc[3]) * wv.w);
+    // CSR over "models whose argmax class on candidate b is c": avg H/C
+    // iterations instead of an H-long wave-uniform load chain
+    const int k0 = offsets[(size_t)b * (C + 1) + c];
+    const int k1 = offsets[(size_t)b * (C + 1) + c + 1];
+    for (int k = k0; k < k1; ++k) {
+        const int h = hvals[(size_t)b * H + k];
+        const float4 d = *reinterpret_cast<const float4*>(
+            delta + dbase + (size_t)h * P_POINTS);
+        acc[0] += d.x; acc[1] += d.y; acc[2] += d.z; acc[3] += d.w;
     }
+    const float4 wv = *reinterpret_cast<const float4*>(w + p0);
+    TOUT* dst = es + ((size_t)c * B + b) * P_POINTS + p0;
+    dst[0] = (TOUT)(exp2f(acc[0]) * wv.x);
+    dst[1] = (TOUT)(exp2f(acc[1]) * wv.y);
+    dst[2] = (TOUT)(exp2f(acc[2]) * wv.z);
+    dst[3] = (TOUT)(exp2f(acc[3]) * wv.w);
 }
 
 template <typename TM>
@@ -631,33 +628,31 @@ es_build_gathered_kernel(const float* __restrict__ s_base_all,  // (C, P)
                          const float* __restrict__ w,        // (P,)
                          TOUT* __restrict__ es,              // (C, B, P)
                          int B, int C, int Hg) {
-    const int nbq = (B + ES_BPW - 1) / ES_BPW;
     const int r = blockIdx.x * ROWS_PER_BLOCK + (threadIdx.x >> 6);
-    if (r >= C * nbq) return;
-    const int c = r / nbq, b0 = (r - c * nbq) * ES_BPW;
+    if (r >= B * C) return;
+    const int b = r / C, c = r - b * C;
     const int lane = threadIdx.x & 63;
     const int p0 = lane * PTS_PER_LANE;
 
+    float acc[PTS_PER_LANE];
     const float4 sb = *reinterpret_cast<const float4*>(
         s_base_all + (size_t)c * P_POINTS + p0);
-    const float4 wv = *reinterpret_cast<const float4*>(w + p0);
-    for (int b = b0; b < min(b0 + ES_BPW, B); ++b) {
-        float acc[PTS_PER_LANE] = {sb.x, sb.y, sb.z, sb.w};
-        const size_t sbase = (size_t)b * Hg * P_POINTS + p0;
-        const int k0 = offsets[(size_t)b * (C + 1) + c];
-        const int k1 = offsets[(size_t)b * (C + 1) + c + 1];
-        for (int k = k0; k < k1; ++k) {
-            const int h = hvals[(size_t)b * Hg + k];
-            const float4 d = *reinterpret_cast<const float4*>(
-                sel_all + sbase + (size_t)h * P_POINTS);
-            acc[0] += d.x; acc[1] += d.y; acc[2] += d.z; acc[3] += d.w;
-        }
-        TOUT* dst = es + ((size_t)c * B + b) * P_POINTS + p0;
-        dst[0] = (TOUT)(exp2f(acc[0]) * wv.x);
-        dst[1] = (TOUT)(exp2f(acc[1]) * wv.y);
-        dst[2] = (TOUT)(exp2f(acc[2]) * wv.z);
-        dst[3] = (TOUT)(exp2f(acc[3]) * wv.w);
+    acc[0] = sb.x; acc[1] = sb.y; acc[2] = sb.z; acc[3] = sb.w;
+    const size_t sbase = (size_t)b * Hg * P_POINTS + p0;
+    const int k0 = offsets[(size_t)b * (C + 1) + c];
+    const int k1 = offsets[(size_t)b * (C + 1) + c + 1];
+    for (int k = k0; k < k1; ++k) {
+        const int h = hvals[(size_t)b * Hg + k];
+        const float4 d = *reinterpret_cast<const float4*>(
+            sel_all + sbase + (size_t)h * P_POINTS);
+        acc[0] += d.x; acc[1] += d.y; acc[2] += d.z; acc[3] += d.w;
     }
+    const float4 wv = *reinterpret_cast<const float4*>(w + p0);
+    TOUT* dst = es + ((size_t)c * B + b) * P_POINTS + p0;
+    dst[0] = (TOUT)(exp2f(acc[0]) * wv.x);
+    dst[1] = (TOUT)(exp2f(acc[1]) * wv.y);
+    dst[2] = (TOUT)(exp2f(acc[2]) * wv.z);
+    dst[3] = (TOUT)(exp2f(acc[3]) * wv.w);
 }
 
 template <typename TM>
@@ -992,7 +987,7 @@ torch::Tensor es_build(torch::Tensor s_base, torch::Tensor delta,
                            s_base.options().dtype(
                                bf16_out ? torch::kBFloat16
                                         : torch::kFloat32));
-    const int R = C * ((B + 3) / 4);  // ES_BPW = 4 candidates per wave
+    const int R = B * C;
     const int blocks = (R + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
     auto stream = c10::hip::getCurrentHIPStream();
     if (bf16_out) {
@@ -1070,7 +1065,7 @@ torch::Tensor es_build_gathered(torch::Tensor s_base_all,
                            s_base_all.options().dtype(
                                bf16_out ? torch::kBFloat16
                                         : torch::kFloat32));
-    const int R = C * ((B + 3) / 4);  // ES_BPW = 4 candidates per wave
+    const int R = B * C;
     const int blocks = (R + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
     auto stream = c10::hip::getCurrentHIPStream();
     if (bf16_out) {
