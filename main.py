@@ -153,8 +153,9 @@ def do_model_selection_experiment(dataset, oracle, args, loss_fn, seed=0,
     start_m, cumulative_regret_loss = 0, 0.0
     if args.checkpoint_every:
         os.makedirs(args.checkpoint_dir, exist_ok=True)
+        safe_task = args.task.replace("/", "_")  # GLUE tasks: 'glue/cola'
         ckpt_path = os.path.join(
-            args.checkpoint_dir, f"{args.task}-{args.method}-{seed}.pt")
+            args.checkpoint_dir, f"{safe_task}-{args.method}-{seed}.pt")
         if ckpt.exists(ckpt_path, comm.rank, comm.world):
             blob = torch.load(
                 ckpt_path + (f".rank{comm.rank}" if comm.world > 1 else ""),

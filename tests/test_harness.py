@@ -203,3 +203,32 @@ def test_convert_task_roundtrip(task_dir):
     assert b.preds.dtype == torch.float32  # up-cast on load
     torch.testing.assert_close(a.preds, b.preds, rtol=1e-2, atol=5e-3)
     assert b.labels is not None
+
+
+def test_slashed_task_names(tmp_path):
+    """GLUE-style task names contain '/' (TASK_EPS: 'glue/cola'); the
+    harness, tracking DB and checkpoints must all handle them."""
+    from coda_amd.datasets import write_synthetic_task
+    import sqlite3 as sq
+    write_synthetic_task(str(tmp_path / "data" / "glue"), name="cola",
+                         H=4, N=80, C=3, seed=5)
+    cwd = os.getcwd()
+    os.chdir(tmp_path)
+    try:
+        from coda_amd import tracking
+        tracking.set_tracking_uri("sqlite:///coda.sqlite")
+        tracking._EXPERIMENT = None
+        tracking._RUN_STACK.clear()
+        import main as harness
+        harness.main(["--task", "glue/cola", "--data-dir", "data",
+                      "--device", "cpu", "--method", "iid", "--iters", "2",
+                      "--seeds", "1", "--checkpoint-every", "1",
+                      "--checkpoint-dir", "ck"])
+    finally:
+        os.chdir(cwd)
+    conn = sq.connect(str(tmp_path / "coda.sqlite"))
+    exps = [r[0] for r in conn.execute("SELECT name FROM experiments")]
+    n = conn.execute("SELECT COUNT(*) FROM metrics WHERE key='regret'"
+                     ).fetchone()[0]
+    conn.close()
+    assert "glue/cola" in exps and n == 2
