@@ -350,12 +350,16 @@ class CODA(ModelSelector):
                                  col) * lr                       # (N,)
         self._adjusted.index_add_(1, y_t, delta.unsqueeze(1))
         self._row_sums += delta
-        pi = ops._ext.pi_marginal(self._adjusted, self._row_sums)             if (self.C <= 2048 and ops.hip_available()) else             (1.0 / self._row_sums.clamp_min(1e-12)) @ self._adjusted
+        if self.C <= 2048 and ops.hip_available():
+            pi = ops._ext.pi_marginal(self._adjusted, self._row_sums)
+        else:
+            pi = (1.0 / self._row_sums.clamp_min(1e-12)) @ self._adjusted
         self._g_pi.copy_(pi / pi.sum())
         # refresh the labeled class's table row (v2 tables)
         if self._tables is not None:
             row = self.dirichlets.index_select(1, y_t).squeeze(1)  # (Hl,C)
-            a_col = row.gather(1, y_t.view(1, 1).expand(self.Hl, 1))                 .squeeze(1).contiguous()
+            a_col = row.gather(
+                1, y_t.view(1, 1).expand(self.Hl, 1)).squeeze(1).contiguous()
             b_col = (row.sum(1) - a_col).contiguous()
             eg, lc = ops._ext.beta_row_tables(a_col, b_col, 1.0)
             t = self._tables
