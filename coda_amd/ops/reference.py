@@ -28,7 +28,6 @@ kernels validated against these (see coda_amd/ops/__init__.py dispatch).
 from __future__ import annotations
 
 import torch
-import torch.nn.functional as F
 
 # Grid constants for the Beta P(best) integral (coda/coda.py:80,86)
 PBEST_NUM_POINTS = 256

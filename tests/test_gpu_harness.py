@@ -2,7 +2,6 @@
 
 Marked gpu; gives the round-end GPU suite coverage of the complete
 user-facing loop (not just the kernels)."""
-import os
 import sqlite3
 
 import pytest

@@ -10,7 +10,6 @@ at all.
 import os
 import re
 import subprocess
-import sys
 
 import pytest
 

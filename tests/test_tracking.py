@@ -1,5 +1,4 @@
 """Tracking store: MLflow-SQLite schema fidelity + harness query patterns."""
-import os
 import sqlite3
 
 import pytest
@@ -113,7 +112,6 @@ def test_concurrent_writers(db):
     """8 processes logging into one DB simultaneously (the task-parallel
     launcher's pattern on an 8-GPU node): WAL + busy timeout must keep
     every write."""
-    import multiprocessing as mp
     import subprocess
     import sys
 
