@@ -74,8 +74,13 @@ def pbest_from_beta(alpha: torch.Tensor, beta: torch.Tensor,
     not a silent fallback).
     """
     H = alpha.shape[-1]
-    if H > 2048:  # beyond the kernel's LDS budget: chunked eager passes
+    if H > 2048:  # beyond the fused kernel's LDS budget
         assert not return_unnormalized
+        if num_points == PBEST_NUM_POINTS and _want_hip(alpha):
+            out = _pbest_wide_hip(alpha, beta)
+            if DEBUG:
+                _check(out, "pbest(wide)")
+            return out
         return reference.pbest_from_beta_hchunked(alpha, beta, num_points)
     if (num_points == PBEST_NUM_POINTS and not return_unnormalized
             and _want_hip(alpha)):
@@ -88,14 +93,37 @@ def pbest_from_beta(alpha: torch.Tensor, beta: torch.Tensor,
                                      return_unnormalized)
 
 
+def _pbest_wide_hip(alpha: torch.Tensor, beta: torch.Tensor,
+                    hc: int = 512) -> torch.Tensor:
+    """Wide-H P(best) through the two-pass window kernels.
+
+    The LDS-window chunks play the role of ranks in the sharded math
+    (ops/sharded.py): pass-A slog2 partials sum to the global H-coupling,
+    pass B integrates each window against it. Identical math to the
+    fused kernel, two launches + two tiny reductions total.
+    """
+    a, b = alpha.contiguous(), beta.contiguous()
+    slog2 = _ext.pbest_phase1_wide(a, b, hc).sum(0)
+    pb, tot_part = _ext.pbest_phase2_wide(a, b, slog2.contiguous(), hc)
+    tot = tot_part.sum(0)
+    return pb / tot.clamp_min(reference.EPS_PROB).unsqueeze(-1)
+
+
 def pi_hat_delta(preds: torch.Tensor, point_classes: torch.Tensor,
                  chunk_h: int = 32) -> torch.Tensor:
     """sum_h preds[h, :, cls_h] -> (N,) fp32 (the rank-1 pi_hat term)."""
     if (preds.is_cuda and preds.is_contiguous()
             and preds.dtype in (torch.float32, torch.bfloat16)
             and _want_hip(preds)):
-        out = _ext.pi_hat_delta(preds,
-                                point_classes.to(torch.int32).contiguous())
+        cls32 = point_classes.to(torch.int32).contiguous()
+        H, N = preds.shape[0], preds.shape[1]
+        # wide pools cannot fill the chip with N threads alone: chunk H
+        # across blockIdx.y so (N/256)*(H/256) blocks cover 256 CUs, then
+        # reduce partials deterministically on-device
+        if H > 512 and N < (1 << 20):
+            out = _ext.pi_hat_delta_part(preds, cls32, 256).sum(0)
+        else:
+            out = _ext.pi_hat_delta(preds, cls32)
         if DEBUG:
             _check(out, "pi_hat_delta(kernel)")
         return out

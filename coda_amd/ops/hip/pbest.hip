@@ -43,6 +43,7 @@
 #include <hip/hip_runtime.h>
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_bfloat16.h>
+#include <algorithm>
 #include <vector>
 
 #define P_POINTS 256
@@ -466,6 +467,93 @@ pbest_phase2_kernel(const float* __restrict__ alpha,
     if (lane == 0) tot_out[r] = total;
 }
 
+// ---------------------------------------------------------------------------
+// Wide-H single-device variants: the same two-pass split, but the "shards"
+// are LDS-sized column windows of one (R, Htot) matrix processed by
+// blockIdx.y, coupled through global memory instead of RCCL. Lifts the
+// H <= 2048 LDS ceiling of the fused kernel (model pools of 10k+) while
+// keeping every pass in LDS and the whole op at two launches + two tiny
+// sums (vs the chunked eager fallback's ~1 GB log-cdf intermediates).
+// ---------------------------------------------------------------------------
+__device__ void stage_window(const float* alpha, const float* beta, int R,
+                             int Htot, int h0, int Hc, int row0,
+                             double* lnB_all, float* f_all) {
+    for (int idx = threadIdx.x; idx < ROWS_PER_BLOCK * Hc; idx += BLOCK) {
+        const int rl = idx / Hc, h = idx - rl * Hc;
+        const int r = row0 + rl;
+        if (r >= R || h0 + h >= Htot) continue;
+        float a = alpha[(size_t)r * Htot + h0 + h];
+        float b = beta[(size_t)r * Htot + h0 + h];
+        f_all[rl * 3 * Hc + h] = a;
+        f_all[rl * 3 * Hc + Hc + h] = b;
+        lnB_all[rl * Hc + h] = (lgamma((double)a) + lgamma((double)b)
+                             - lgamma((double)a + (double)b))
+                             * 1.4426950408889634;
+    }
+    __syncthreads();
+}
+
+__global__ void CODA_LB
+pbest_phase1_wide_kernel(const float* __restrict__ alpha,
+                         const float* __restrict__ beta,
+                         float* __restrict__ slog2_part,  // (KH, R, P)
+                         int R, int Htot, int Hc) {
+    extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+    double* lnB_all = reinterpret_cast<double*>(smem_raw);
+    float* f_all = reinterpret_cast<float*>(lnB_all + ROWS_PER_BLOCK * Hc);
+    const int k = blockIdx.y, h0 = k * Hc;
+    const int Hcur = min(Hc, Htot - h0);
+    const int row0 = blockIdx.x * ROWS_PER_BLOCK;
+    stage_window(alpha, beta, R, Htot, h0, Hc, row0, lnB_all, f_all);
+    const int rl = threadIdx.x >> 6;
+    const int r = row0 + rl;
+    if (r >= R) return;
+    const int lane = threadIdx.x & 63;
+    LaneGrid g;
+    g.init(lane);
+    float slog2[PTS_PER_LANE];
+    // stage_window packs the window at stride Hc, so pass A sees a dense
+    // Hcur-model row exactly like a shard's local slice
+    pbest_pass_a(f_all + rl * 3 * Hc, f_all + rl * 3 * Hc + Hc,
+                 lnB_all + rl * Hc, Hcur, g, lane, slog2);
+#pragma unroll
+    for (int j = 0; j < PTS_PER_LANE; ++j)
+        slog2_part[((size_t)k * R + r) * P_POINTS
+                   + lane * PTS_PER_LANE + j] = slog2[j];
+}
+
+__global__ void CODA_LB
+pbest_phase2_wide_kernel(const float* __restrict__ alpha,
+                         const float* __restrict__ beta,
+                         const float* __restrict__ slog2_in,  // (R, P) global
+                         float* __restrict__ pb_out,          // (R, Htot)
+                         float* __restrict__ tot_part,        // (KH, R)
+                         int R, int Htot, int Hc) {
+    extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+    double* lnB_all = reinterpret_cast<double*>(smem_raw);
+    float* f_all = reinterpret_cast<float*>(lnB_all + ROWS_PER_BLOCK * Hc);
+    const int k = blockIdx.y, h0 = k * Hc;
+    const int Hcur = min(Hc, Htot - h0);
+    const int row0 = blockIdx.x * ROWS_PER_BLOCK;
+    stage_window(alpha, beta, R, Htot, h0, Hc, row0, lnB_all, f_all);
+    const int rl = threadIdx.x >> 6;
+    const int r = row0 + rl;
+    if (r >= R) return;
+    const int lane = threadIdx.x & 63;
+    LaneGrid g;
+    g.init(lane);
+    float slog2[PTS_PER_LANE];
+#pragma unroll
+    for (int j = 0; j < PTS_PER_LANE; ++j)
+        slog2[j] = slog2_in[(size_t)r * P_POINTS + lane * PTS_PER_LANE + j];
+    float* s_pb = f_all + rl * 3 * Hc + 2 * Hc;
+    float total = pbest_pass_b(f_all + rl * 3 * Hc, f_all + rl * 3 * Hc + Hc,
+                               lnB_all + rl * Hc, s_pb, Hcur, g, lane, slog2);
+    for (int h = lane; h < Hcur; h += 64)
+        pb_out[(size_t)r * Htot + h0 + h] = s_pb[h];
+    if (lane == 0) tot_part[(size_t)k * R + r] = total;
+}
+
 __global__ void CODA_LB
 eig_phase1_kernel(const float* __restrict__ alpha_t,
                   const float* __restrict__ beta_t,
@@ -765,6 +853,27 @@ pi_hat_delta_kernel(const T* __restrict__ preds,  // (H, N, C)
     out[n] = acc;
 }
 
+// H-chunked variant for wide model pools: with only N threads the plain
+// kernel cannot fill 256 CUs (N=5k, H=10k left it 13x slower than its
+// memory floor). blockIdx.y sums an H window into partial[(k, n)]; the
+// host reduces partials with a deterministic torch sum (no atomics, so
+// trajectories stay bit-reproducible).
+template <typename T>
+__global__ void CODA_LB
+pi_hat_delta_part_kernel(const T* __restrict__ preds,   // (H, N, C)
+                         const int* __restrict__ cls,   // (H,)
+                         float* __restrict__ partial,   // (KH, N)
+                         int H, long long N, int C, int Hc) {
+    const int hbeg = blockIdx.y * Hc;
+    const int hend = min(H, hbeg + Hc);
+    const long long n = (long long)blockIdx.x * BLOCK + threadIdx.x;
+    if (n >= N) return;
+    float acc = 0.f;
+    for (int h = hbeg; h < hend; ++h)
+        acc += (float)preds[((long long)h * N + n) * C + cls[h]];
+    partial[(size_t)blockIdx.y * N + n] = acc;
+}
+
 
 // pi marginal: out[c] = sum_n adjusted[n, c] / max(row_sums[n], 1e-12)
 // (reference coda/coda.py:229-233 without materializing the normalized
@@ -913,6 +1022,53 @@ std::vector<torch::Tensor> pbest_phase2(torch::Tensor alpha,
                        pb.data_ptr<float>(), tot.data_ptr<float>(), R, H);
     C10_HIP_CHECK(hipGetLastError());
     return {pb, tot};
+}
+
+torch::Tensor pbest_phase1_wide(torch::Tensor alpha, torch::Tensor beta,
+                                int64_t hc) {
+    check_f32_cuda(alpha, "alpha");
+    check_f32_cuda(beta, "beta");
+    const int R = alpha.size(0), Htot = alpha.size(1);
+    const int Hc = std::min<int64_t>(hc, Htot);
+    const int KH = (Htot + Hc - 1) / Hc;
+    auto part = torch::empty({KH, R, P_POINTS}, alpha.options());
+    if (R == 0) return part;
+    const size_t smem = lds_bytes(Hc);
+    TORCH_CHECK(smem <= 160 * 1024, "Hc too large for LDS: ", Hc);
+    const int bx = (R + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
+    auto stream = c10::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(pbest_phase1_wide_kernel, dim3(bx, KH), dim3(BLOCK),
+                       smem, stream.stream(), alpha.data_ptr<float>(),
+                       beta.data_ptr<float>(), part.data_ptr<float>(),
+                       R, Htot, Hc);
+    C10_HIP_CHECK(hipGetLastError());
+    return part;
+}
+
+std::vector<torch::Tensor> pbest_phase2_wide(torch::Tensor alpha,
+                                             torch::Tensor beta,
+                                             torch::Tensor slog2,
+                                             int64_t hc) {
+    check_f32_cuda(alpha, "alpha");
+    check_f32_cuda(beta, "beta");
+    check_f32_cuda(slog2, "slog2");
+    const int R = alpha.size(0), Htot = alpha.size(1);
+    const int Hc = std::min<int64_t>(hc, Htot);
+    const int KH = (Htot + Hc - 1) / Hc;
+    auto pb = torch::empty({R, Htot}, alpha.options());
+    auto tot_part = torch::empty({KH, R}, alpha.options());
+    if (R == 0) return {pb, tot_part};
+    const size_t smem = lds_bytes(Hc);
+    TORCH_CHECK(smem <= 160 * 1024, "Hc too large for LDS: ", Hc);
+    const int bx = (R + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
+    auto stream = c10::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(pbest_phase2_wide_kernel, dim3(bx, KH), dim3(BLOCK),
+                       smem, stream.stream(), alpha.data_ptr<float>(),
+                       beta.data_ptr<float>(), slog2.data_ptr<float>(),
+                       pb.data_ptr<float>(), tot_part.data_ptr<float>(),
+                       R, Htot, Hc);
+    C10_HIP_CHECK(hipGetLastError());
+    return {pb, tot_part};
 }
 
 torch::Tensor eig_phase1(torch::Tensor alpha_cc, torch::Tensor beta_cc,
@@ -1201,6 +1357,39 @@ torch::Tensor pi_hat_delta(torch::Tensor preds, torch::Tensor cls) {
 }
 
 
+torch::Tensor pi_hat_delta_part(torch::Tensor preds, torch::Tensor cls,
+                                int64_t hc) {
+    TORCH_CHECK(preds.is_cuda() && preds.is_contiguous(),
+                "preds must be contiguous on a ROCm device");
+    TORCH_CHECK(cls.scalar_type() == torch::kInt32, "cls must be int32");
+    const int H = preds.size(0), C = preds.size(2);
+    const long long N = preds.size(1);
+    const int Hc = (int)std::min<int64_t>(hc, H);
+    const int KH = (H + Hc - 1) / Hc;
+    auto partial = torch::empty({KH, N},
+                                preds.options().dtype(torch::kFloat32));
+    const int bx = (int)((N + BLOCK - 1) / BLOCK);
+    auto stream = c10::hip::getCurrentHIPStream();
+    if (preds.scalar_type() == torch::kFloat32) {
+        hipLaunchKernelGGL(pi_hat_delta_part_kernel<float>, dim3(bx, KH),
+                           dim3(BLOCK), 0, stream.stream(),
+                           preds.data_ptr<float>(), cls.data_ptr<int>(),
+                           partial.data_ptr<float>(), H, N, C, Hc);
+    } else if (preds.scalar_type() == torch::kBFloat16) {
+        hipLaunchKernelGGL(pi_hat_delta_part_kernel<hip_bfloat16>,
+                           dim3(bx, KH), dim3(BLOCK), 0, stream.stream(),
+                           reinterpret_cast<const hip_bfloat16*>(
+                               preds.data_ptr()),
+                           cls.data_ptr<int>(), partial.data_ptr<float>(),
+                           H, N, C, Hc);
+    } else {
+        TORCH_CHECK(false, "pi_hat_delta kernel supports fp32/bf16 only");
+    }
+    C10_HIP_CHECK(hipGetLastError());
+    return partial;
+}
+
+
 torch::Tensor pi_marginal(torch::Tensor adjusted, torch::Tensor row_sums) {
     check_f32_cuda(adjusted, "adjusted");
     check_f32_cuda(row_sums, "row_sums");
@@ -1228,6 +1417,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "Sharded pass A: local sum_h log2 cdf partials (R, P)");
     m.def("pbest_phase2", &pbest_phase2,
           "Sharded pass B: unnormalized masses + normalizer partial");
+    m.def("pbest_phase1_wide", &pbest_phase1_wide,
+          "Wide-H pass A: per-window slog2 partials (KH, R, P)");
+    m.def("pbest_phase2_wide", &pbest_phase2_wide,
+          "Wide-H pass B: unnormalized (R, Htot) masses + (KH, R) totals");
+    m.def("pi_hat_delta_part", &pi_hat_delta_part,
+          "H-chunked rank-1 pi_hat increment partials (KH, N)");
     m.def("eig_phase1", &eig_phase1,
           "Sharded hypothetical pass A: slog2 partials (B*C, P)");
     m.def("eig_phase2", &eig_phase2,

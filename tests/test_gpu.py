@@ -323,14 +323,43 @@ class TestTableFusionKernels:
 
 class TestWidePoolGPU:
     def test_hchunked_pbest_gpu(self, dev):
-        """H beyond the kernel LDS budget routes through the chunked eager
-        passes on GPU too."""
+        """H beyond the fused kernel's LDS budget routes through the
+        two-pass window kernels; matches the chunked eager math."""
         from coda_amd import ops
         a, b = _rand_betas(rows=3, H=2100, seed=61, lo=1.0, hi=20.0,
                            device=dev)
         p = ops.pbest_from_beta(a, b)
         assert torch.isfinite(p).all()
         np.testing.assert_allclose(p.sum(-1).cpu().numpy(), 1.0, atol=1e-3)
+        want = ops.reference.pbest_from_beta_hchunked(a.cpu(), b.cpu())
+        torch.testing.assert_close(p.cpu(), want, rtol=2e-3, atol=1e-5)
+
+    def test_wide_pbest_equals_fused_at_boundary(self, dev):
+        """Window kernels (forced) == fused kernel on an H that both
+        support: same math, different coupling route."""
+        from coda_amd import ops
+        a, b = _rand_betas(rows=5, H=1500, seed=62, lo=0.5, hi=40.0,
+                           device=dev)
+        fused = ops.pbest_from_beta(a, b)
+        wide = ops._pbest_wide_hip(a, b, hc=512)
+        torch.testing.assert_close(wide, fused, rtol=1e-4, atol=1e-6)
+        # uneven tail window (1500 = 2*640 + 220)
+        wide2 = ops._pbest_wide_hip(a, b, hc=640)
+        torch.testing.assert_close(wide2, fused, rtol=1e-4, atol=1e-6)
+
+    def test_pi_hat_delta_wide_pool(self, dev):
+        """H-chunked pi_hat_delta partials (H > 512 dispatch) match the
+        eager reference."""
+        from coda_amd import ops
+        g = torch.Generator().manual_seed(46)
+        H, N, C = 600, 300, 7
+        preds = torch.softmax(torch.randn(H, N, C, generator=g), -1).to(dev)
+        cls = torch.randint(0, C, (H,), generator=g).to(dev)
+        got = ops.pi_hat_delta(preds, cls)
+        want = ops.reference.pi_hat_delta(preds.cpu(), cls.cpu()).to(dev)
+        torch.testing.assert_close(got, want, rtol=1e-5, atol=1e-6)
+        got16 = ops.pi_hat_delta(preds.to(torch.bfloat16).contiguous(), cls)
+        torch.testing.assert_close(got16, want, rtol=2e-2, atol=1e-2)
 
     def test_coda_wide_pool_gpu(self, dev):
         """End-to-end CODA on a 2100-model pool (wide-H fallback for the
