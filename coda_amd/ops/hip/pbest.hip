@@ -647,15 +647,41 @@ es_build_kernel(const float* __restrict__ s_base,   // (C, P)
     acc[0] = sb.x; acc[1] = sb.y; acc[2] = sb.z; acc[3] = sb.w;
     const size_t dbase = (size_t)c * H * P_POINTS + p0;
     // CSR over "models whose argmax class on candidate b is c": avg H/C
-    // iterations instead of an H-long wave-uniform load chain
+    // iterations instead of an H-long wave-uniform load chain. Four
+    // accumulator sets break the fp32 add dependency (the compiler may
+    // not reassociate), so four loads are in flight per wave.
     const int k0 = offsets[(size_t)b * (C + 1) + c];
     const int k1 = offsets[(size_t)b * (C + 1) + c + 1];
-    for (int k = k0; k < k1; ++k) {
+    float a1[4] = {0.f, 0.f, 0.f, 0.f}, a2[4] = {0.f, 0.f, 0.f, 0.f},
+          a3[4] = {0.f, 0.f, 0.f, 0.f};
+    int k = k0;
+    for (; k + 3 < k1; k += 4) {
+        const int ha = hvals[(size_t)b * H + k];
+        const int hb = hvals[(size_t)b * H + k + 1];
+        const int hc2 = hvals[(size_t)b * H + k + 2];
+        const int hd = hvals[(size_t)b * H + k + 3];
+        const float4 da = *reinterpret_cast<const float4*>(
+            delta + dbase + (size_t)ha * P_POINTS);
+        const float4 db = *reinterpret_cast<const float4*>(
+            delta + dbase + (size_t)hb * P_POINTS);
+        const float4 dc = *reinterpret_cast<const float4*>(
+            delta + dbase + (size_t)hc2 * P_POINTS);
+        const float4 dd = *reinterpret_cast<const float4*>(
+            delta + dbase + (size_t)hd * P_POINTS);
+        acc[0] += da.x; acc[1] += da.y; acc[2] += da.z; acc[3] += da.w;
+        a1[0] += db.x; a1[1] += db.y; a1[2] += db.z; a1[3] += db.w;
+        a2[0] += dc.x; a2[1] += dc.y; a2[2] += dc.z; a2[3] += dc.w;
+        a3[0] += dd.x; a3[1] += dd.y; a3[2] += dd.z; a3[3] += dd.w;
+    }
+    for (; k < k1; ++k) {
         const int h = hvals[(size_t)b * H + k];
         const float4 d = *reinterpret_cast<const float4*>(
             delta + dbase + (size_t)h * P_POINTS);
         acc[0] += d.x; acc[1] += d.y; acc[2] += d.z; acc[3] += d.w;
     }
+#pragma unroll
+    for (int j = 0; j < PTS_PER_LANE; ++j)
+        acc[j] += (a1[j] + a2[j]) + a3[j];
     const float4 wv = *reinterpret_cast<const float4*>(w + p0);
     TOUT* dst = es + ((size_t)c * B + b) * P_POINTS + p0;
     dst[0] = (TOUT)(exp2f(acc[0]) * wv.x);
@@ -729,12 +755,36 @@ es_build_gathered_kernel(const float* __restrict__ s_base_all,  // (C, P)
     const size_t sbase = (size_t)b * Hg * P_POINTS + p0;
     const int k0 = offsets[(size_t)b * (C + 1) + c];
     const int k1 = offsets[(size_t)b * (C + 1) + c + 1];
-    for (int k = k0; k < k1; ++k) {
+    float a1[4] = {0.f, 0.f, 0.f, 0.f}, a2[4] = {0.f, 0.f, 0.f, 0.f},
+          a3[4] = {0.f, 0.f, 0.f, 0.f};
+    int k = k0;
+    for (; k + 3 < k1; k += 4) {
+        const int ha = hvals[(size_t)b * Hg + k];
+        const int hb = hvals[(size_t)b * Hg + k + 1];
+        const int hc2 = hvals[(size_t)b * Hg + k + 2];
+        const int hd = hvals[(size_t)b * Hg + k + 3];
+        const float4 da = *reinterpret_cast<const float4*>(
+            sel_all + sbase + (size_t)ha * P_POINTS);
+        const float4 db = *reinterpret_cast<const float4*>(
+            sel_all + sbase + (size_t)hb * P_POINTS);
+        const float4 dc = *reinterpret_cast<const float4*>(
+            sel_all + sbase + (size_t)hc2 * P_POINTS);
+        const float4 dd = *reinterpret_cast<const float4*>(
+            sel_all + sbase + (size_t)hd * P_POINTS);
+        acc[0] += da.x; acc[1] += da.y; acc[2] += da.z; acc[3] += da.w;
+        a1[0] += db.x; a1[1] += db.y; a1[2] += db.z; a1[3] += db.w;
+        a2[0] += dc.x; a2[1] += dc.y; a2[2] += dc.z; a2[3] += dc.w;
+        a3[0] += dd.x; a3[1] += dd.y; a3[2] += dd.z; a3[3] += dd.w;
+    }
+    for (; k < k1; ++k) {
         const int h = hvals[(size_t)b * Hg + k];
         const float4 d = *reinterpret_cast<const float4*>(
             sel_all + sbase + (size_t)h * P_POINTS);
         acc[0] += d.x; acc[1] += d.y; acc[2] += d.z; acc[3] += d.w;
     }
+#pragma unroll
+    for (int j = 0; j < PTS_PER_LANE; ++j)
+        acc[j] += (a1[j] + a2[j]) + a3[j];
     const float4 wv = *reinterpret_cast<const float4*>(w + p0);
     TOUT* dst = es + ((size_t)c * B + b) * P_POINTS + p0;
     dst[0] = (TOUT)(exp2f(acc[0]) * wv.x);

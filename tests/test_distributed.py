@@ -181,3 +181,31 @@ def test_main_sharded_torchrun_cpu(tmp_path):
         "SELECT COUNT(*) FROM metrics WHERE key='regret'").fetchone()[0]
     conn.close()
     assert n == 3
+
+
+@pytest.mark.timeout(600)
+def test_bench_torchrun_cpu():
+    """The driver's SCALE invocation shape, on CPU/gloo at a tiny config:
+    `torch.distributed.run --nproc-per-node 2 bench.py --gpus 2` must
+    print the one-line JSON contract from rank 0."""
+    import json
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    env.update(CODA_BENCH_H="6", CODA_BENCH_N="150", CODA_BENCH_C="4",
+               CODA_BENCH_PREFILTER="32", CODA_BENCH_CHUNK="32")
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29663", "bench.py", "--gpus", "2",
+         "--steps", "2", "--warmup", "1"],
+        cwd=repo, env=env, capture_output=True, text=True, timeout=500)
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    d = json.loads(line)
+    assert d["n_gpus"] == 2 and d["steps"] == 2
+    assert d["config"]["H_models"] == 12  # weak scaling: 6 per rank
+    assert d["config"]["parallelism"] == "model-shard2"
+    assert d["value"] > 0 and d["higher_is_better"] is True
