@@ -74,12 +74,22 @@ def main():
     chunk_classes = sel.classes[:, cand].t().contiguous()
     pi_xi = sel.pi_hat_xi[cand]
 
-    t_chunk = timed(lambda: ops.eig_chunk(
-        alpha_cc, beta_cc, chunk_classes, pbest_before, sel.pi_hat,
-        pi_xi, mixture0, H_before), sync)
-    rows = args.chunk * args.c
-    print(f"eig_chunk B={args.chunk:5d}    {t_chunk:8.3f} s "
-          f"({1e9*t_chunk/rows:.0f} ns/row, {rows/t_chunk/1e6:.2f} Mrow/s)")
+    if args.h <= 2048:  # v1 fused kernel path (eager fallback above that)
+        t_chunk = timed(lambda: ops.eig_chunk(
+            alpha_cc, beta_cc, chunk_classes, pbest_before, sel.pi_hat,
+            pi_xi, mixture0, H_before), sync)
+        rows = args.chunk * args.c
+        print(f"eig_chunk B={args.chunk:5d}    {t_chunk:8.3f} s "
+              f"({1e9*t_chunk/rows:.0f} ns/row, {rows/t_chunk/1e6:.2f} Mrow/s)")
+    if device.type == "cuda":
+        from coda_amd.ops import table as tops
+        tables = tops.table_precompute(alpha_cc, beta_cc)
+        t_tab = timed(lambda: tops.eig_chunk_table(
+            tables, chunk_classes, pbest_before, sel.pi_hat, pi_xi,
+            mixture0, H_before), sync)
+        rows = args.chunk * args.c
+        print(f"eig_chunk_table      {t_tab:8.3f} s "
+              f"({1e9*t_tab/rows:.0f} ns/row)")
 
     print(f"eig_batched (full)   {timed(lambda: sel.eig_batched(), sync):8.3f} s")
     print(f"get_pbest            {timed(sel.get_pbest, sync):8.3f} s")
