@@ -94,14 +94,24 @@ def pbest_from_beta(alpha: torch.Tensor, beta: torch.Tensor,
 
 
 def _pbest_wide_hip(alpha: torch.Tensor, beta: torch.Tensor,
-                    hc: int = 512) -> torch.Tensor:
+                    hc: int = 0) -> torch.Tensor:
     """Wide-H P(best) through the two-pass window kernels.
 
     The LDS-window chunks play the role of ranks in the sharded math
     (ops/sharded.py): pass-A slog2 partials sum to the global H-coupling,
     pass B integrates each window against it. Identical math to the
     fused kernel, two launches + two tiny reductions total.
+
+    hc (window size, <= 2048 for LDS) defaults to the largest window
+    that still yields >= ~512 workgroups - few rows (the per-class
+    posterior refresh is R = C rows) need narrow windows to fill
+    256 CUs.
     """
+    R, H = alpha.shape
+    if hc <= 0:
+        bx = max(1, (R + 3) // 4)
+        kh_needed = max(1, (512 + bx - 1) // bx)
+        hc = min(512, max(64, (H + kh_needed - 1) // kh_needed))
     a, b = alpha.contiguous(), beta.contiguous()
     slog2 = _ext.pbest_phase1_wide(a, b, hc).sum(0)
     pb, tot_part = _ext.pbest_phase2_wide(a, b, slog2.contiguous(), hc)
