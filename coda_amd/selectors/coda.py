@@ -348,8 +348,13 @@ class CODA(ModelSelector):
         self.dirichlets.index_add_(1, y_t, (lr * onehot).unsqueeze(1))
         delta = ops.pi_hat_delta(self.dataset.preds,
                                  col) * lr                       # (N,)
-        self._adjusted.index_add_(1, y_t, delta.unsqueeze(1))
-        self._row_sums += delta
+        if self._adjusted.is_cuda and ops.hip_available():
+            # fused column update (torch index_add_ over dim 1 with one
+            # index runs ~90x slower than this elementwise pass)
+            ops._ext.col_add(self._adjusted, self._row_sums, y_t, delta)
+        else:
+            self._adjusted.index_add_(1, y_t, delta.unsqueeze(1))
+            self._row_sums += delta
         if self.C <= 2048 and ops.hip_available():
             pi = ops._ext.pi_marginal(self._adjusted, self._row_sums)
         else:

@@ -310,6 +310,21 @@ class TestTableFusionKernels:
         got16 = ops.pi_hat_delta(preds.to(torch.bfloat16).contiguous(), cls)
         torch.testing.assert_close(got16, want, rtol=2e-2, atol=1e-2)
 
+    def test_col_add_kernel(self, dev):
+        from coda_amd import ops
+        g = torch.Generator().manual_seed(47)
+        N, C = 1234, 56
+        A = torch.rand(N, C, generator=g).to(dev)
+        rs = torch.rand(N, generator=g).to(dev)
+        delta = torch.randn(N, generator=g).to(dev)
+        y = torch.tensor([13], dtype=torch.long, device=dev)
+        A2, rs2 = A.clone(), rs.clone()
+        A2.index_add_(1, y, delta.unsqueeze(1))
+        rs2 += delta
+        ops._ext.col_add(A, rs, y, delta)
+        torch.testing.assert_close(A, A2)
+        torch.testing.assert_close(rs, rs2)
+
     def test_pi_marginal_kernel(self, dev):
         from coda_amd import ops
         g = torch.Generator().manual_seed(45)
