@@ -343,9 +343,14 @@ class CODA(ModelSelector):
         lr = self.update_strength
         idx_t, y_t = self._g_idx, self._g_y
         col = self.classes.index_select(1, idx_t).squeeze(1)     # (Hl,)
-        onehot = torch.nn.functional.one_hot(col, self.C).to(
-            self.dirichlets.dtype)
-        self.dirichlets.index_add_(1, y_t, (lr * onehot).unsqueeze(1))
+        if self.dirichlets.is_cuda and ops.hip_available():
+            # H-thread scatter: torch's one_hot + dim-1 index_add_ on the
+            # (H,C,C) posterior costs ~914 us for H real nonzeros
+            ops._ext.dirichlet_add(self.dirichlets, y_t, col, float(lr))
+        else:
+            onehot = torch.nn.functional.one_hot(col, self.C).to(
+                self.dirichlets.dtype)
+            self.dirichlets.index_add_(1, y_t, (lr * onehot).unsqueeze(1))
         delta = ops.pi_hat_delta(self.dataset.preds,
                                  col) * lr                       # (N,)
         if self._adjusted.is_cuda and ops.hip_available():

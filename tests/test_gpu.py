@@ -310,6 +310,20 @@ class TestTableFusionKernels:
         got16 = ops.pi_hat_delta(preds.to(torch.bfloat16).contiguous(), cls)
         torch.testing.assert_close(got16, want, rtol=2e-2, atol=1e-2)
 
+    def test_dirichlet_add_kernel(self, dev):
+        from coda_amd import ops
+        g = torch.Generator().manual_seed(48)
+        H, C = 37, 11
+        dir0 = torch.rand(H, C, C, generator=g).to(dev)
+        cls = torch.randint(0, C, (H,), generator=g).to(dev)
+        y = torch.tensor([4], dtype=torch.long, device=dev)
+        lr = 0.01
+        want = dir0.clone()
+        onehot = torch.nn.functional.one_hot(cls, C).float()
+        want.index_add_(1, y, (lr * onehot).unsqueeze(1))
+        ops._ext.dirichlet_add(dir0, y, cls, lr)
+        torch.testing.assert_close(dir0, want)
+
     def test_col_add_kernel(self, dev):
         from coda_amd import ops
         g = torch.Generator().manual_seed(47)
