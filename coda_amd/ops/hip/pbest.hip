@@ -36,8 +36,14 @@
 //   eig_totals/entropy     v2 sharded split around the normalizer
 //                          all-reduce
 //   beta_row_tables_kernel v2: per-class incremental table refresh
+//   pbest_phase1/2_wide    wide-H (> 2048 models): the two-pass split with
+//                          LDS column windows as the "ranks", coupled
+//                          through global memory on one device
 //   pi_hat_delta_kernel    rank-1 posterior-marginal increment
+//   pi_hat_delta_part      H-chunked variant for wide pools (blockIdx.y)
 //   pi_marginal_kernel     streaming scaled column sum for pi_hat
+//   col_add_kernel         fused adjusted[:, y] += delta; row_sums += delta
+//   dirichlet_add_kernel   dir[h, y, cls_h] += lr posterior scatter
 
 #include <torch/extension.h>
 #include <hip/hip_runtime.h>
