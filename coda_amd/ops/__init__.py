@@ -23,7 +23,7 @@ from .reference import (  # re-export cheap ops + constants
     PBEST_NUM_POINTS, GRID_LO, GRID_HI, EPS_PROB, LOG_CLAMP,
     consensus, confusion_prior, init_dirichlets, dirichlet_to_beta,
     pi_hat_partial, pi_hat_pack, pi_hat_pack_chunked,
-    pi_hat_partial_packed, pi_hat_normalize,
+    pi_hat_partial_packed, pi_hat_partial_streamed, pi_hat_normalize,
     pbest_from_beta_hchunked, init_model_stats,
     beta_grid_pdf_cdf, hypothetical_betas,
     mixture_entropy, pred_classes, disagreement_mask,

@@ -175,6 +175,33 @@ def pi_hat_partial_packed(dirichlets: torch.Tensor,
     return (packed @ B).float()
 
 
+def pi_hat_partial_streamed(dirichlets: torch.Tensor, preds: torch.Tensor,
+                            chunk_n: int = 32768) -> torch.Tensor:
+    """pi_hat_partial_packed without ever holding the full (N, H*C)
+    operand: pack chunk_n points, GEMM, reuse the buffer.
+
+    Peak extra memory is one (chunk_n, H*C) bf16 buffer (8 GB at
+    H=128 x C=1000) instead of N x H x C x 2 bytes - 256 GB at the
+    1M-point config, which cannot coexist with the prediction pool in
+    288 GB HBM. Same math and the same bf16 MFMA GEMM rate (each chunk
+    is still a huge (chunk_n) x (C) x (K=H*C) GEMM).
+    """
+    H, N, C = preds.shape
+    B = dirichlets.transpose(1, 2).reshape(H * C, C).to(torch.bfloat16)
+    out = torch.empty(N, C, dtype=torch.float32, device=preds.device)
+    m = min(chunk_n, N)
+    buf = torch.empty(m, H * C, dtype=torch.bfloat16, device=preds.device)
+    for n0 in range(0, N, chunk_n):
+        n1 = min(n0 + chunk_n, N)
+        v = buf[: n1 - n0].view(n1 - n0, H, C)
+        for h0 in range(0, H, 16):
+            h1 = min(h0 + 16, H)
+            v[:, h0:h1] = preds[h0:h1, n0:n1].to(
+                torch.bfloat16).permute(1, 0, 2)
+        out[n0:n1] = (buf[: n1 - n0] @ B).float()
+    return out
+
+
 def pi_hat_delta(preds: torch.Tensor, point_classes: torch.Tensor,
                  chunk_h: int = 32) -> torch.Tensor:
     """Incremental pi_hat update term for one labeled point.

@@ -95,8 +95,6 @@ class CODA(ModelSelector):
         if preds.dtype == torch.float8_e4m3fn and pi_hat_precision == "fp32":
             pi_hat_precision = "bf16"  # fp8 storage has no fp32 bmm path
         self.pi_hat_precision = pi_hat_precision
-        self._pi_packed = ops.pi_hat_pack_chunked(preds) \
-            if pi_hat_precision == "bf16" else None
         self._adjusted = None  # maintained incrementally after init
 
         # consensus prior: global mean over H (all-reduce site K1)
@@ -168,12 +166,12 @@ class CODA(ModelSelector):
         go through the exact rank-1 incremental path in add_label
         (ops.pi_hat_delta), which is O(H*N) instead of O(H*N*C^2).
         """
-        if self._pi_packed is not None:
-            adjusted = ops.pi_hat_partial_packed(self.dirichlets,
-                                                 self._pi_packed)
-            # the packed operand is only needed for full refreshes; drop it
-            # to reclaim the (N, H*C) bf16 buffer once initialized
-            self._pi_packed = None
+        if self.pi_hat_precision == "bf16":
+            # N-chunked pack + bf16 MFMA GEMM: never holds the full
+            # (N, H*C) operand, so million-point pools fit alongside the
+            # prediction tensor in HBM
+            adjusted = ops.pi_hat_partial_streamed(self.dirichlets,
+                                                   self.dataset.preds)
         else:
             adjusted = ops.pi_hat_partial(self.dirichlets, self.dataset.preds)
         self.comm.all_reduce_(adjusted)

@@ -260,6 +260,21 @@ class TestPiHatPacked:
         # bf16 inputs: ~0.4% relative tolerance
         torch.testing.assert_close(got, want, rtol=2e-2, atol=1e-3)
 
+    def test_streamed_matches_packed(self):
+        """N-chunked streamed pack+GEMM == the one-shot packed GEMM
+        exactly (same bf16 rounding, including a ragged last chunk)."""
+        g = torch.Generator().manual_seed(22)
+        H, N, C = 5, 37, 4
+        preds = torch.softmax(torch.randn(H, N, C, generator=g), -1)
+        D = torch.rand(H, C, C, generator=g) + 0.1
+        want = ops.pi_hat_partial_packed(D, ops.pi_hat_pack(preds))
+        got = ops.pi_hat_partial_streamed(D, preds, chunk_n=16)
+        torch.testing.assert_close(got, want)
+        # bf16 storage input too
+        got16 = ops.pi_hat_partial_streamed(D, preds.to(torch.bfloat16),
+                                            chunk_n=16)
+        torch.testing.assert_close(got16, want, rtol=2e-2, atol=1e-3)
+
 
 class TestTableEig:
     """v2 (table-factored) EIG vs v1 (fused composition) equivalence."""
