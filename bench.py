@@ -44,14 +44,16 @@ CHUNK = int(os.environ.get("CODA_BENCH_CHUNK", 256))
 STORAGE = os.environ.get("CODA_BENCH_STORAGE", "fp32")
 
 
-def synth_preds(model_idxs, N, C, device, seed_base=1234):
+def synth_preds(model_idxs, N, C, device, seed_base=1234,
+                dtype=torch.float32):
     """Per-model deterministic synthetic predictions: rank-independent.
 
     Model h's tensor depends only on (seed_base + h), so a sharded run sees
     exactly the data the 1-GPU run sees for the same global model.
+    `dtype` is the STORAGE dtype: each model's softmax converts on the fly,
+    so the pool never exists at fp32 (512 GB at the 1M-point config).
     """
-    out = torch.empty(len(model_idxs), N, C, device=device,
-                      dtype=torch.float32)
+    out = torch.empty(len(model_idxs), N, C, device=device, dtype=dtype)
     g = torch.Generator(device=device)
     labels_g = torch.Generator(device=device)
     labels_g.manual_seed(seed_base - 1)
@@ -67,7 +69,7 @@ def synth_preds(model_idxs, N, C, device, seed_base=1234):
         target = torch.where(correct, labels, (labels + wrong) % C)
         logits.scatter_add_(1, target.unsqueeze(1),
                             torch.full((N, 1), 4.0, device=device))
-        out[i] = torch.softmax(logits, dim=-1)
+        out[i] = torch.softmax(logits, dim=-1).to(dtype)
     return out, labels
 
 
@@ -92,10 +94,9 @@ def main():
     model_idxs = list(range(comm.rank, H_TOTAL, comm.world)) \
         if shard else list(range(H_TOTAL))
 
-    preds, labels = synth_preds(model_idxs, N_POINTS, C_CLASSES, device)
-    if STORAGE != "fp32":
-        from coda_amd.datasets import STORAGE_DTYPES
-        preds = preds.to(STORAGE_DTYPES[STORAGE])
+    from coda_amd.datasets import STORAGE_DTYPES
+    preds, labels = synth_preds(model_idxs, N_POINTS, C_CLASSES, device,
+                                dtype=STORAGE_DTYPES[STORAGE])
     ds = Dataset.from_tensors(preds, labels, device, shard=None)
     ds.total_models = H_TOTAL
     ds.shard = shard

@@ -46,12 +46,15 @@ def consensus(preds: torch.Tensor) -> torch.Tensor:
     return init_model_stats(preds)[1] / preds.shape[0]
 
 
-def init_model_stats(preds: torch.Tensor, chunk_h: int = 16):
+def init_model_stats(preds: torch.Tensor, chunk_h: int = 0):
     """One chunked fp32-upcast pass over the model axis at selector init:
     cached argmax classes (H, N) + the consensus sum (N, C) fp32.
     Storage may be fp32 / bf16 / fp8 (coda_amd.datasets.STORAGE_DTYPES);
-    compute is always fp32."""
+    compute is always fp32. chunk_h defaults to <= 1 GiB of fp32 chunk
+    (a fixed 16 was a 64 GB transient at the 1M-point config)."""
     H, N, C = preds.shape
+    if chunk_h <= 0:
+        chunk_h = max(1, min(16, (1 << 28) // max(1, N * C)))
     classes = torch.empty(H, N, dtype=torch.long, device=preds.device)
     ens = torch.zeros(N, C, dtype=torch.float32, device=preds.device)
     for h0 in range(0, H, chunk_h):
