@@ -81,10 +81,15 @@ class Dataset:
     @classmethod
     def from_tensors(cls, preds: torch.Tensor, labels: Optional[torch.Tensor],
                      device, shard: Optional[Tuple[int, int]] = None):
-        """Build a Dataset from in-memory tensors (tests, synthetic tasks)."""
+        """Build a Dataset from in-memory tensors (tests, synthetic tasks).
+
+        A tensor already in a storage dtype (bf16 / fp8-e4m3) is kept as
+        is - up-casting a 1M-point fp8 pool to fp32 would be 512 GB.
+        """
         self = cls.__new__(cls)
         self.device = torch.device(device)
-        preds = preds.float()
+        if preds.dtype not in STORAGE_DTYPES.values():
+            preds = preds.float()
         self.total_models = preds.shape[0]
         self.shard = shard
         if shard is not None:

@@ -123,7 +123,8 @@ def pi_hat_delta(preds: torch.Tensor, point_classes: torch.Tensor,
                  chunk_h: int = 32) -> torch.Tensor:
     """sum_h preds[h, :, cls_h] -> (N,) fp32 (the rank-1 pi_hat term)."""
     if (preds.is_cuda and preds.is_contiguous()
-            and preds.dtype in (torch.float32, torch.bfloat16)
+            and preds.dtype in (torch.float32, torch.bfloat16,
+                                torch.float8_e4m3fn)
             and _want_hip(preds)):
         cls32 = point_classes.to(torch.int32).contiguous()
         H, N = preds.shape[0], preds.shape[1]

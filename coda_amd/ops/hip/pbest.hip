@@ -49,6 +49,7 @@
 #include <hip/hip_runtime.h>
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_bfloat16.h>
+#include <hip/hip_fp8.h>  // __hip_fp8_e4m3 (OCP = torch.float8_e4m3fn)
 #include <algorithm>
 #include <vector>
 
@@ -1449,8 +1450,15 @@ torch::Tensor pi_hat_delta(torch::Tensor preds, torch::Tensor cls) {
                                preds.data_ptr()),
                            cls.data_ptr<int>(), out.data_ptr<float>(),
                            H, N, C);
+    } else if (preds.scalar_type() == torch::kFloat8_e4m3fn) {
+        hipLaunchKernelGGL(pi_hat_delta_kernel<__hip_fp8_e4m3>,
+                           dim3(blocks), dim3(BLOCK), 0, stream.stream(),
+                           reinterpret_cast<const __hip_fp8_e4m3*>(
+                               preds.data_ptr()),
+                           cls.data_ptr<int>(), out.data_ptr<float>(),
+                           H, N, C);
     } else {
-        TORCH_CHECK(false, "pi_hat_delta kernel supports fp32/bf16 only");
+        TORCH_CHECK(false, "pi_hat_delta kernel supports fp32/bf16/fp8");
     }
     C10_HIP_CHECK(hipGetLastError());
     return out;
@@ -1482,8 +1490,15 @@ torch::Tensor pi_hat_delta_part(torch::Tensor preds, torch::Tensor cls,
                                preds.data_ptr()),
                            cls.data_ptr<int>(), partial.data_ptr<float>(),
                            H, N, C, Hc);
+    } else if (preds.scalar_type() == torch::kFloat8_e4m3fn) {
+        hipLaunchKernelGGL(pi_hat_delta_part_kernel<__hip_fp8_e4m3>,
+                           dim3(bx, KH), dim3(BLOCK), 0, stream.stream(),
+                           reinterpret_cast<const __hip_fp8_e4m3*>(
+                               preds.data_ptr()),
+                           cls.data_ptr<int>(), partial.data_ptr<float>(),
+                           H, N, C, Hc);
     } else {
-        TORCH_CHECK(false, "pi_hat_delta kernel supports fp32/bf16 only");
+        TORCH_CHECK(false, "pi_hat_delta kernel supports fp32/bf16/fp8");
     }
     C10_HIP_CHECK(hipGetLastError());
     return partial;
