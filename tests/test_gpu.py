@@ -309,6 +309,12 @@ class TestTableFusionKernels:
         # bf16 storage
         got16 = ops.pi_hat_delta(preds.to(torch.bfloat16).contiguous(), cls)
         torch.testing.assert_close(got16, want, rtol=2e-2, atol=1e-2)
+        # fp8 storage (e4m3): compare against the eager op on the SAME
+        # fp8 values (the storage rounding is semantics, not kernel error)
+        p8 = preds.to(torch.float8_e4m3fn).contiguous()
+        got8 = ops.pi_hat_delta(p8, cls)
+        want8 = ops.reference.pi_hat_delta(p8.cpu(), cls.cpu()).to(dev)
+        torch.testing.assert_close(got8, want8, rtol=1e-3, atol=1e-3)
 
     def test_dirichlet_add_kernel(self, dev):
         from coda_amd import ops
