@@ -976,13 +976,47 @@ pi_marginal_kernel(const float* __restrict__ adjusted,  // (N, C)
                    float* __restrict__ out,             // (C,) pre-zeroed
                    long long N, int C) {
     const int tid = threadIdx.x;
-    const int ncols = (C + BLOCK - 1) / BLOCK;
-    float acc[8];  // supports C <= 8*BLOCK = 2048
-    for (int k = 0; k < ncols && k < 8; ++k) acc[k] = 0.f;
     const long long rows_per_block =
         (N + gridDim.x - 1) / gridDim.x;
     const long long n0 = (long long)blockIdx.x * rows_per_block;
     const long long n1 = min(n0 + rows_per_block, N);
+    if ((C & 3) == 0 && C <= 4 * BLOCK) {
+        // vector path: thread tid owns columns [4*tid, 4*tid+4) as ONE
+        // float4 per row - 4x wider transactions than the scalar path
+        // (the (N, C) read is this kernel's whole cost at large N)
+        const int c4 = tid * 4;
+        if (c4 < C) {
+            float4 acc = {0.f, 0.f, 0.f, 0.f};
+            long long n = n0;
+            for (; n + 1 < n1; n += 2) {  // 2-row unroll
+                const float inv0 = 1.0f / fmaxf(row_sums[n], 1e-12f);
+                const float inv1 = 1.0f / fmaxf(row_sums[n + 1], 1e-12f);
+                const float4 v0 = *reinterpret_cast<const float4*>(
+                    adjusted + n * C + c4);
+                const float4 v1 = *reinterpret_cast<const float4*>(
+                    adjusted + (n + 1) * C + c4);
+                acc.x += v0.x * inv0 + v1.x * inv1;
+                acc.y += v0.y * inv0 + v1.y * inv1;
+                acc.z += v0.z * inv0 + v1.z * inv1;
+                acc.w += v0.w * inv0 + v1.w * inv1;
+            }
+            for (; n < n1; ++n) {
+                const float inv = 1.0f / fmaxf(row_sums[n], 1e-12f);
+                const float4 v = *reinterpret_cast<const float4*>(
+                    adjusted + n * C + c4);
+                acc.x += v.x * inv; acc.y += v.y * inv;
+                acc.z += v.z * inv; acc.w += v.w * inv;
+            }
+            atomicAdd(out + c4 + 0, acc.x);
+            atomicAdd(out + c4 + 1, acc.y);
+            atomicAdd(out + c4 + 2, acc.z);
+            atomicAdd(out + c4 + 3, acc.w);
+        }
+        return;
+    }
+    const int ncols = (C + BLOCK - 1) / BLOCK;
+    float acc[8];  // supports C <= 8*BLOCK = 2048
+    for (int k = 0; k < ncols && k < 8; ++k) acc[k] = 0.f;
     long long n = n0;
     for (; n + 1 < n1; n += 2) {  // 2-row unroll: 2x loads in flight
         const float inv0 = 1.0f / fmaxf(row_sums[n], 1e-12f);
