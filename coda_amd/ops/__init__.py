@@ -128,11 +128,14 @@ def pi_hat_delta(preds: torch.Tensor, point_classes: torch.Tensor,
             and _want_hip(preds)):
         cls32 = point_classes.to(torch.int32).contiguous()
         H, N = preds.shape[0], preds.shape[1]
-        # wide pools cannot fill the chip with N threads alone: chunk H
-        # across blockIdx.y so (N/256)*(H/256) blocks cover 256 CUs, then
+        # N threads alone may not fill 256 CUs (H=128 x N=50k is 196
+        # blocks): chunk H across blockIdx.y until >= 512 blocks, then
         # reduce partials deterministically on-device
-        if H > 512 and N < (1 << 20):
-            out = _ext.pi_hat_delta_part(preds, cls32, 256).sum(0)
+        bx = (N + 255) // 256
+        kh = -(-512 // bx)
+        if kh > 1 and H >= 2 * kh:
+            hc = -(-H // kh)
+            out = _ext.pi_hat_delta_part(preds, cls32, hc).sum(0)
         else:
             out = _ext.pi_hat_delta(preds, cls32)
         if DEBUG:
