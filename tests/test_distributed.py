@@ -209,3 +209,18 @@ def test_bench_torchrun_cpu():
     assert d["config"]["H_models"] == 12  # weak scaling: 6 per rank
     assert d["config"]["parallelism"] == "model-shard2"
     assert d["value"] > 0 and d["higher_is_better"] is True
+
+
+def test_bench_synth_preds_storage_dtype():
+    """bench generates the pool directly in the storage dtype (the fp32
+    intermediate would be 512 GB at the 1M-point config) and rows still
+    sum to ~1 after conversion."""
+    import bench
+    preds, labels = bench.synth_preds([0, 3, 7], 50, 6, "cpu",
+                                      dtype=torch.bfloat16)
+    assert preds.dtype == torch.bfloat16 and preds.shape == (3, 50, 6)
+    sums = preds.float().sum(-1)
+    assert (sums - 1.0).abs().max() < 0.05
+    # per-model determinism: same global model id -> same tensor
+    again, _ = bench.synth_preds([3], 50, 6, "cpu", dtype=torch.bfloat16)
+    torch.testing.assert_close(again[0], preds[1])
