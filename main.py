@@ -86,6 +86,13 @@ def parse_args(argv=None):
                              "path on GPU, fp32 accumulate)")
     parser.add_argument("--sharded", action="store_true",
                         help="Shard the model axis across torchrun ranks.")
+    parser.add_argument("--storage", default="fp32",
+                        choices=["fp32", "bf16", "fp8"],
+                        help="On-device prediction-pool dtype (compute "
+                             "always upcasts to fp32).")
+    parser.add_argument("--stream-chunk-mb", type=int, default=0,
+                        help="Stream the pool host->HBM through pinned "
+                             "buffers of this size (0 = one blocking copy).")
     parser.add_argument("--checkpoint-every", type=int, default=0,
                         help="Save selector state every K steps (0 = off); "
                              "an interrupted seed resumes mid-run.")
@@ -220,7 +227,9 @@ def main(argv=None):
 
     shard = (comm.rank, comm.world) if comm.is_distributed else None
     dataset = Dataset(os.path.join(args.data_dir, args.task + ".pt"),
-                      device=device, shard=shard)
+                      device=device, shard=shard,
+                      storage_dtype=args.storage,
+                      stream_chunk_mb=args.stream_chunk_mb)
     loss_fn = LOSS_FNS[args.loss]
     oracle = Oracle(dataset, loss_fn=loss_fn)
 
