@@ -232,3 +232,11 @@ def test_slashed_task_names(tmp_path):
                      ).fetchone()[0]
     conn.close()
     assert "glue/cola" in exps and n == 2
+
+
+def test_storage_flag_cpu(task_dir):
+    """--storage bf16 runs the harness end-to-end with a bf16 pool on CPU
+    (compute upcasts per chunk)."""
+    _run_cli(task_dir, ["--method", "coda", "--iters", "2", "--seeds", "1",
+                        "--no-mlflow", "--chunk-size", "64",
+                        "--storage", "bf16"])
