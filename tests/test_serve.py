@@ -108,3 +108,23 @@ def test_image_endpoint(tmp_path):
     assert r.content[:4] == b"\x89PNG"
     assert c.get("/image/99").status_code == 404
     assert c.get("/state").json()["class_names"] == ["cat"]
+
+
+@pytest.mark.gpu
+def test_interactive_loop_gpu():
+    """The serving protocol with the selector on cuda:0 (HIP kernels in
+    the loop)."""
+    import torch
+    assert torch.cuda.is_available()
+    preds, labels = make_synthetic_task(H=6, N=200, C=5, seed=9)
+    ds = Dataset.from_tensors(preds, labels, "cuda:0")
+    oracle = Oracle(ds, LOSS_FNS["acc"])
+    app = create_app(ds, method="coda", oracle=oracle, chunk_size=64)
+    c = TestClient(app)
+    for _ in range(2):
+        nxt = c.get("/next").json()
+        st = c.post("/answer", json={"index": nxt["index"],
+                                     "label": oracle(nxt["index"])}).json()
+    assert st["step"] == 2
+    pb = c.get("/pbest").json()["pbest"]
+    assert len(pb) == 6 and abs(sum(pb) - 1.0) < 1e-3
