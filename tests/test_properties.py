@@ -78,3 +78,17 @@ def test_table_matches_fused_random(H, C, B, seed):
     want = ops.pbest_from_beta(ah.reshape(B * C, H),
                                bh.reshape(B * C, H)).reshape(B, C, H)
     torch.testing.assert_close(got, want, rtol=5e-3, atol=1e-4)
+
+
+@settings(max_examples=15, deadline=None)
+@given(st.integers(2, 8), st.integers(5, 60), st.integers(2, 9),
+       st.integers(1, 64), st.integers(0, 9999))
+def test_streamed_pi_hat_random(H, N, C, chunk_n, seed):
+    """N-chunked streamed pack+GEMM == one-shot packed GEMM for any
+    chunking (including chunk_n > N and ragged tails)."""
+    g = torch.Generator().manual_seed(seed)
+    preds = torch.softmax(torch.randn(H, N, C, generator=g), -1)
+    D = torch.rand(H, C, C, generator=g) + 0.05
+    want = ops.pi_hat_partial_packed(D, ops.pi_hat_pack(preds))
+    got = ops.pi_hat_partial_streamed(D, preds, chunk_n=chunk_n)
+    torch.testing.assert_close(got, want)
