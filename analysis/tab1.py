@@ -27,10 +27,13 @@ METHOD_LABELS = {
 
 
 def extract_method_from_run_name(run_name: str) -> str:
-    parts = run_name.split("-")
-    if len(parts) >= 2 and parts[-1].isdigit():
-        parts = parts[:-1]
-    return "-".join(parts[1:]) if len(parts) > 1 else run_name
+    """Child runs are named '<task>-<method>[-<seed>]' (main.py run
+    naming); recover the method string, which may itself contain dashes
+    (e.g. 'coda-lr=0.01-mult=2.0-no-prefilter')."""
+    segments = run_name.split("-")
+    if segments[-1].isdigit() and len(segments) > 1:
+        segments.pop()  # trailing seed number
+    return "-".join(segments[1:]) or run_name
 
 
 def load(db_path: str, metric: str, step: int) -> pd.DataFrame:

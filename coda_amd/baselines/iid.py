@@ -38,8 +38,19 @@ class IID(ModelSelector):
         return idx, 1.0 / len(self.d_u_idxs)
 
     def _point_loss(self, idx: int, label: int) -> torch.Tensor:
-        """(Hl,) loss of each model on one labeled point (acc loss)."""
-        return 1.0 - (self.classes[:, idx] == label).float()
+        """(Hl,) loss of each model on one labeled point.
+
+        Routed through self.loss_fn so a future non-accuracy loss stays
+        consistent with the oracle; the accuracy loss keeps its
+        cached-argmax fast path (no (Hl, C) gather from the pool).
+        """
+        from ..options import accuracy_loss
+        if self.loss_fn is accuracy_loss or self.loss_fn is None:
+            return 1.0 - (self.classes[:, idx] == label).float()
+        point_preds = self.dataset.preds[:, idx].float()      # (Hl, C)
+        label_t = torch.full((self.Hl,), int(label),
+                             device=self.device, dtype=torch.long)
+        return self.loss_fn(point_preds, label_t)
 
     def add_label(self, chosen_idx, true_class, selection_prob=None):
         chosen_idx = int(chosen_idx)

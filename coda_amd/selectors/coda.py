@@ -317,12 +317,15 @@ class CODA(ModelSelector):
                 pass  # no active tracking run
 
         # greedy with seeded random tie-breaking (coda/coda.py:306-313);
-        # max/argmax/tie-count fetched in ONE device sync
+        # max/argmax/tie-count fetched in ONE device sync. The index and
+        # count travel as a separate int64 pair (a float32 round-trip
+        # corrupts indices beyond 2^24, i.e. pools over ~16.7M points).
         best_val, best_idx = q_vals.max(0)
         n_ties = torch.isclose(q_vals, best_val, rtol=1e-8).sum()
         bv, bi, nt = torch.stack(
-            [best_val, best_idx.to(q_vals.dtype),
-             n_ties.to(q_vals.dtype)]).cpu().tolist()
+            [best_val.double(), best_idx.double(),
+             n_ties.double()]).cpu().tolist()
+        bi, nt = int(bi), int(nt)
         if nt > 1:
             ties = torch.isclose(q_vals, best_val, rtol=1e-8)
             idx_local = random.choice(

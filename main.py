@@ -82,8 +82,15 @@ def parse_args(argv=None):
     parser.add_argument("--chunk-size", type=int, default=100)
     parser.add_argument("--pi-hat-precision", default="auto",
                         choices=["auto", "fp32", "bf16"],
-                        help="pi_hat contraction dtype (bf16 = MFMA fast "
-                             "path on GPU, fp32 accumulate)")
+                        help="pi_hat contraction dtype. 'auto' is the "
+                             "documented bf16 MFMA fast path on GPU "
+                             "(fp32 accumulate); pass fp32 for strict "
+                             "bit-parity with the reference's fp32 GEMM.")
+    parser.add_argument("--debug-checks", action="store_true",
+                        help="Enable the per-stage NaN/Inf and "
+                             "probability guards (the reference's "
+                             "default-on _DEBUG behavior; each check "
+                             "synchronizes the device).")
     parser.add_argument("--sharded", action="store_true",
                         help="Shard the model axis across torchrun ranks.")
     parser.add_argument("--storage", default="fp32",
@@ -101,6 +108,15 @@ def parse_args(argv=None):
 
 
 def build_selector(dataset, args, loss_fn, comm=None):
+    if (comm is not None and comm.is_distributed
+            and not args.method.startswith("coda")):
+        # Baselines are shard-unaware: their best-model index would be
+        # shard-LOCAL while regret is evaluated in GLOBAL model order.
+        raise ValueError(
+            f"--sharded supports only CODA methods; '{args.method}' "
+            "would compute regret against the wrong model. Run "
+            "baselines single-process (the task-parallel launcher "
+            "covers multi-GPU baseline sweeps).")
     if args.method == "iid":
         return IID(dataset, loss_fn)
     if args.method == "uncertainty":
@@ -213,6 +229,10 @@ def do_model_selection_experiment(dataset, oracle, args, loss_fn, seed=0,
 
 def main(argv=None):
     args = parse_args(argv)
+
+    if args.debug_checks:
+        from coda_amd.util import set_debug
+        set_debug(True)
 
     comm = init_from_env() if args.sharded else get_comm()
     if args.device is not None:
