@@ -33,11 +33,14 @@ from coda_amd.datasets import Dataset
 from coda_amd.options import LOSS_FNS
 from coda_amd.parallel import init_from_env, get_comm
 
-# Headline config (overridable via env for CPU smoke testing only)
+# Headline config (overridable via env for CPU smoke testing only).
+# PREFILTER=0 is the reference's canonical acquisition (main.py:49, run
+# name `coda-lr=0.01-mult=2.0-no-prefilter`, paper/tab1.py:60): EIG over
+# EVERY disagreeing unlabeled point each step - ~50k candidates here.
 H_PER_GPU = int(os.environ.get("CODA_BENCH_H", 128))
 N_POINTS = int(os.environ.get("CODA_BENCH_N", 50_000))
 C_CLASSES = int(os.environ.get("CODA_BENCH_C", 1000))
-PREFILTER_N = int(os.environ.get("CODA_BENCH_PREFILTER", 256))
+PREFILTER_N = int(os.environ.get("CODA_BENCH_PREFILTER", 0))
 CHUNK = int(os.environ.get("CODA_BENCH_CHUNK", 256))
 # storage dtype for the prediction pool (fp32 | bf16 | fp8); compute is
 # always fp32 (coda_amd/datasets.py STORAGE_DTYPES)
@@ -76,13 +79,22 @@ def synth_preds(model_idxs, N, C, device, seed_base=1234,
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=50)
-    ap.add_argument("--warmup", type=int, default=20)
+    ap.add_argument("--steps", type=int, default=200)
+    ap.add_argument("--warmup", type=int, default=25)
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     comm = init_from_env() if world > 1 else get_comm()
-    n_gpus = max(args.gpus, world)
+    if world > 1 and args.gpus != world:
+        raise SystemExit(
+            f"--gpus {args.gpus} does not match WORLD_SIZE={world}; "
+            "launch one rank per GPU (torch.distributed.run "
+            "--nproc-per-node N ... bench.py --gpus N)")
+    n_gpus = world if world > 1 else args.gpus
+    if n_gpus != 1 and world == 1:
+        raise SystemExit(
+            f"--gpus {n_gpus} requested but WORLD_SIZE is unset; "
+            "multi-GPU runs go through torch.distributed.run")
     H_TOTAL = H_PER_GPU * comm.world  # weak scaling: 128 models per GPU
 
     if torch.cuda.is_available():
@@ -158,7 +170,7 @@ def main():
                 "C_classes": C_CLASSES,
                 "prefilter_n": PREFILTER_N,
                 "chunk_size": CHUNK,
-                "global_batch": PREFILTER_N,
+                "global_batch": PREFILTER_N or N_POINTS,
                 "seq_len": C_CLASSES,
                 "parallelism": f"model-shard{comm.world}",
             },

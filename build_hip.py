@@ -15,7 +15,8 @@ import sys
 import sysconfig
 
 REPO = os.path.dirname(os.path.abspath(__file__))
-SRC = [os.path.join(REPO, "coda_amd", "ops", "hip", "pbest.hip")]
+SRC = [os.path.join(REPO, "coda_amd", "ops", "hip", "pbest.hip"),
+       os.path.join(REPO, "coda_amd", "ops", "hip", "pair.hip")]
 OUT = os.path.join(REPO, "coda_amd", "ops",
                    "_coda_hip.cpython-310-x86_64-linux-gnu.so")
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
@@ -29,8 +30,10 @@ def build(force: bool = False, verbose: bool = True) -> str:
         newest_src = max(os.path.getmtime(s) for s in SRC + [__file__])
         if os.path.getmtime(OUT) >= newest_src:
             if verbose:
-                print(f"[build_hip] up to date: {OUT}")
+                print(f"[build_hip] build_mode=reuse (up to date): {OUT}")
             return OUT
+    if verbose:
+        print("[build_hip] build_mode=compile")
 
     hipcc = os.environ.get("HIPCC", "/opt/rocm/bin/hipcc")
     includes = ce.include_paths() + [sysconfig.get_paths()["include"]]

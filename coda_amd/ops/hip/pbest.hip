@@ -1589,8 +1589,11 @@ torch::Tensor pi_marginal(torch::Tensor adjusted, torch::Tensor row_sums) {
     return out;
 }
 
+void register_pair_ops(pybind11::module_& m);  // pair.hip (v3 engine)
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "coda_amd fused gfx950 kernels";
+    register_pair_ops(m);
     m.def("pbest_from_beta", &pbest_from_beta,
           "Beta-grid P(best) per row: (R,H),(R,H) -> (R,H)");
     m.def("eig_chunk", &eig_chunk,

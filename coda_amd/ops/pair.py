@@ -1,0 +1,238 @@
+"""Pair-factored EIG (v3): hit-sparse acquisition over the full pool.
+
+The canonical acquisition (reference main.py:49 `--prefilter-n 0`,
+hot loop coda/coda.py:235-281) scores EVERY disagreeing unlabeled point
+per step. The v2 table engine (ops/table.py) still does dense work per
+(candidate, class) cell: a (B, C, P) ES tensor + a (C, 2H, P) x (C, P, B)
+GEMM - ~51 GB of ES traffic per step at the headline config (B=50k,
+C=1000, P=256), 85.9 ms/step measured in round 1.
+
+v3 exploits the hit sparsity: for candidate b and hypothesized class c,
+the hypothetical update only deviates from the label-free baseline when
+at least one model PREDICTS class c on point b (coda/coda.py:150-168:
+alpha bumps only for argmax==c models). A candidate's H models hit at
+most min(H, C) distinct classes - typically ~5-40 - so of the B*C cells
+only K = sum_b |distinct classes(b)| ~ B*30 are non-baseline:
+
+    EIG[b] = H_before - sum_c pi_xi[b,c] * h_base[c]
+                      - sum_{c in hit(b)} pi_xi[b,c] * (h_after[b,c] - h_base[c])
+
+with h_base[c] the entropy of the baseline (no-hit) hypothetical - one
+value per class per step. The per-pair work is one P-point curve
+(exp2 of the summed log-cdf deltas) and one (P,) x (P, 2H) pairing GEMV
+against the class's curve table, grouped by class for MFMA tiling
+(ops/hip/pair.hip). ~200x less arithmetic than the dense form at the
+headline shape, and the per-step launch count drops from ~6 per
+256-candidate chunk (~1200) to ~10 total.
+
+The hit structure depends only on the cached argmax classes, so it is
+STATIC across steps: built once over all disagreeing points, reused
+every acquisition (labeled points are skipped at gather time).
+
+Sharding (model axis on disk, candidate axis in compute): every rank
+keeps a replicated (H, C) diagonal-Beta view (updated locally from the
+init-time gathered global classes), computes v3 EIG for a strided slice
+of the candidates, and all-gathers the (B,) EIG values - per-step wire
+is O(B) bytes instead of the 268 MB/chunk curve gather of the v2
+sharded path (ops/table.py:256 in round 1).
+"""
+from __future__ import annotations
+
+from typing import NamedTuple, Optional
+
+import torch
+
+from .reference import EPS_PROB, PBEST_NUM_POINTS
+
+PAIR_TILE = 16  # pairs per MFMA tile row-block; class runs pad to this
+
+
+class PairStructure(NamedTuple):
+    """Static hit structure for a fixed candidate set.
+
+    Pairs are grouped by class; each class run is [base pair] + real
+    pairs + padding to a PAIR_TILE multiple, so one kernel tile never
+    straddles classes. Base/pad pairs carry pair_b = -1 (empty model
+    segment -> they evaluate the class baseline h_base[c]).
+    """
+    cand_ids: torch.Tensor   # (B,) long — point ids covered
+    pair_b: torch.Tensor     # (K,) int32 — row into cand_ids, -1 = base/pad
+    pair_c: torch.Tensor     # (K,) int32 — hypothesized class
+    seg_off: torch.Tensor    # (K+1,) int32 — segment offsets into seg_h
+    seg_h: torch.Tensor      # (S,) int32 — models with argmax c on point b
+    base_pos: torch.Tensor   # (C,) long — position of class c's base pair
+    n_real: int              # real (non-base, non-pad) pair count
+
+    @property
+    def K(self) -> int:
+        return self.pair_b.shape[0]
+
+
+def build_pairs(cls_rows: torch.Tensor, cand_ids: torch.Tensor,
+                C: int) -> PairStructure:
+    """Build the hit structure from candidate argmax classes.
+
+    cls_rows: (B, H) — argmax class of every (global) model on each
+    candidate; cand_ids: (B,) point ids. All work is batched torch ops
+    (one sort over B*H keys), so the one-off cost at N=50k, H=128 is a
+    few ms on device.
+    """
+    device = cls_rows.device
+    B, H = cls_rows.shape
+    cls_l = cls_rows.long()
+    # sort all (b, h) entries by (class, candidate) so pairs come out
+    # grouped by class, and each pair's model segment is contiguous
+    keys = (cls_l * B + torch.arange(B, device=device).unsqueeze(1))
+    flat = keys.reshape(-1)
+    order = torch.argsort(flat, stable=True)
+    sk = flat[order]
+    is_new = torch.ones_like(sk, dtype=torch.bool)
+    is_new[1:] = sk[1:] != sk[:-1]
+    pair_key = sk[is_new]                          # (K_real,)
+    pr_c = pair_key // B
+    pr_b = pair_key - pr_c * B
+    K_real = int(pair_key.shape[0])
+    pid = is_new.long().cumsum(0) - 1
+    seg_len_real = torch.bincount(pid, minlength=K_real)   # (K_real,)
+    seg_h = (order % H).to(torch.int32)            # grouped by pair
+
+    # class runs: base pair + real pairs, padded to PAIR_TILE
+    class_counts = torch.bincount(pr_c, minlength=C)        # (C,)
+    run_len = ((class_counts + 1 + PAIR_TILE - 1)
+               // PAIR_TILE) * PAIR_TILE                    # (C,)
+    run_off = torch.zeros(C + 1, dtype=torch.long, device=device)
+    run_off[1:] = run_len.cumsum(0)
+    K = int(run_off[-1])
+
+    class_off_real = torch.zeros(C, dtype=torch.long, device=device)
+    class_off_real[1:] = class_counts.cumsum(0)[:-1]
+    # position of real pair i: its class run start + 1 (base) + rank
+    rank = torch.arange(K_real, device=device) - class_off_real[pr_c]
+    pos = run_off[:-1][pr_c] + 1 + rank                     # (K_real,)
+
+    pair_b = torch.full((K,), -1, dtype=torch.int32, device=device)
+    pair_b[pos] = pr_b.to(torch.int32)
+    pair_c = torch.repeat_interleave(
+        torch.arange(C, device=device), run_len).to(torch.int32)
+    seg_len = torch.zeros(K, dtype=torch.long, device=device)
+    seg_len[pos] = seg_len_real
+    seg_off = torch.zeros(K + 1, dtype=torch.int32, device=device)
+    seg_off[1:] = seg_len.cumsum(0).to(torch.int32)
+
+    return PairStructure(cand_ids=cand_ids.long(), pair_b=pair_b,
+                         pair_c=pair_c, seg_off=seg_off, seg_h=seg_h,
+                         base_pos=run_off[:-1].clone(), n_real=K_real)
+
+
+def pair_h_after(tables, ps: PairStructure, cls_rows: torch.Tensor,
+                 pbest_before: torch.Tensor, pi_hat: torch.Tensor,
+                 mixture0: torch.Tensor) -> torch.Tensor:
+    """(K,) hypothetical-update entropies, eager formulation.
+
+    Math identical to ops/table.py's dense form restricted to hit cells
+    (the kernel in ops/hip/pair.hip reproduces it in bf16-GEMM form).
+    Used on CPU and as the numerics reference for the kernel; the (K,
+    2H, P) gather makes it unusable at headline scale on purpose.
+    """
+    EG, delta, s_base, w = (tables.EG, tables.delta, tables.s_base,
+                            tables.weights)
+    C, H, _, P = EG.shape
+    K = ps.K
+    device = EG.device
+
+    seg_len = (ps.seg_off[1:] - ps.seg_off[:-1]).long()
+    seg_pair = torch.repeat_interleave(
+        torch.arange(K, device=device), seg_len)
+    pc_long = ps.pair_c.long()
+    dsum = torch.zeros(K, P, device=device)
+    if ps.seg_h.numel():
+        dsum.index_add_(0, seg_pair,
+                        delta[pc_long[seg_pair], ps.seg_h.long()])
+    A = torch.exp2(dsum + s_base[pc_long]) * w              # (K, P)
+    M = torch.einsum('kp,kjp->kj', A,
+                     EG.reshape(C, 2 * H, P)[pc_long])      # (K, 2H)
+
+    b_safe = ps.pair_b.long().clamp_min(0)
+    v = (cls_rows[b_safe] == pc_long.unsqueeze(1)) \
+        & (ps.pair_b >= 0).unsqueeze(1)                     # (K, H)
+    j = 2 * torch.arange(H, device=device).unsqueeze(0) + v.long()
+    pb = M.gather(1, j)                                     # (K, H)
+    tot = pb.sum(-1, keepdim=True).clamp_min(EPS_PROB)
+    pbn = pb / tot
+    mm = (mixture0.unsqueeze(0)
+          + pi_hat[pc_long].unsqueeze(1)
+          * (pbn - pbest_before[pc_long])).clamp_min(1e-12)
+    return -(mm * torch.log2(mm)).sum(-1)                   # (K,)
+
+
+def eig_from_pairs(h_after: torch.Tensor, ps: PairStructure,
+                   adjusted: torch.Tensor, row_sums: torch.Tensor,
+                   H_before) -> torch.Tensor:
+    """(N,) EIG over all points from per-pair entropies.
+
+    Points outside the candidate set get the baseline-only value (never
+    read - the caller gathers cand_ids).
+    """
+    N = adjusted.shape[0]
+    h_base = h_after[ps.base_pos]                           # (C,)
+    inv_rs = 1.0 / row_sums.clamp_min(1e-12)
+    base_n = (adjusted @ h_base) * inv_rs                   # (N,)
+    valid = ps.pair_b >= 0
+    ids = ps.cand_ids[ps.pair_b[valid].long()]              # (Kr,)
+    c_v = ps.pair_c[valid].long()
+    pix = adjusted[ids, c_v] * inv_rs[ids]
+    corr = torch.zeros(N, device=adjusted.device)
+    corr.index_add_(0, ids, pix * (h_after[valid] - h_base[c_v]))
+    if torch.is_tensor(H_before):
+        H_before = H_before.to(base_n.dtype)
+    return H_before - base_n - corr
+
+
+def eig_pairs(tables, ps: PairStructure, cls_rows: torch.Tensor,
+              pbest_before: torch.Tensor, pi_hat: torch.Tensor,
+              mixture0: torch.Tensor, H_before,
+              adjusted: torch.Tensor,
+              row_sums: torch.Tensor) -> torch.Tensor:
+    """Full v3 EIG: (N,) values for every point in one pass.
+
+    Dispatches the per-pair work to the HIP kernels when available
+    (GPU), else the eager formulation.
+    """
+    import coda_amd.ops as O
+    EG = tables.EG
+    C, H, _, P = EG.shape
+    if (EG.is_cuda and P == PBEST_NUM_POINTS
+            and getattr(tables, "egw", None) is not None
+            and O._want_hip(EG)):
+        A16 = O._ext.pair_dsum_es(tables.delta, ps.pair_c, ps.seg_off,
+                                  ps.seg_h)                 # (K, P) bf16
+        h_after = O._ext.pair_gemm_entropy(
+            A16, tables.egw, ps.pair_b, ps.pair_c,
+            cls_rows.to(torch.int32).contiguous(),
+            pi_hat.contiguous(), pbest_before.contiguous(),
+            mixture0.contiguous())                          # (K,)
+    else:
+        h_after = pair_h_after(tables, ps, cls_rows, pbest_before,
+                               pi_hat, mixture0)
+    return eig_from_pairs(h_after, ps, adjusted, row_sums, H_before)
+
+
+def build_egw(tables) -> torch.Tensor:
+    """(C, 2H, P) bf16 MFMA B-operand: EG with the baseline curve and
+    trapezoid weights folded in - egw[c,j,p] = EG[c,j,p] * 2^s_base[c,p]
+    * w[p]. The pair GEMM's A operand is then just 2^dsum."""
+    EG, s_base, w = tables.EG, tables.s_base, tables.weights
+    C, H, _, P = EG.shape
+    esb = torch.exp2(s_base) * w                            # (C, P)
+    return (EG.reshape(C, 2 * H, P)
+            * esb.unsqueeze(1)).to(torch.bfloat16).contiguous()
+
+
+def update_egw_rows(tables, rows) -> None:
+    """Refresh egw for the given class rows (after table_update_rows)."""
+    EG, s_base, w = tables.EG, tables.s_base, tables.weights
+    C, H, _, P = EG.shape
+    for c in rows:
+        esb = torch.exp2(s_base[c]) * w
+        tables.egw[c] = (EG[c].reshape(2 * H, P)
+                         * esb.unsqueeze(0)).to(torch.bfloat16)

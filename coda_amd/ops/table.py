@@ -72,6 +72,10 @@ class EigTables(NamedTuple):
     # MFMA rate - measured EIG impact ~5e-7 absolute, at the level of
     # fp32 reduction-order noise). CODA_AMD_V2_GEMM=fp32 disables.
     eg16: torch.Tensor = None
+    # (C, 2H, P) bf16 pair-GEMM B operand with the baseline curve and
+    # trapz weights folded in (ops/pair.py build_egw); built only by the
+    # v3 pair engine.
+    egw: torch.Tensor = None
 
 
 def table_precompute(alpha_cc: torch.Tensor, beta_cc: torch.Tensor,
@@ -204,6 +208,9 @@ def table_update_rows(tables: EigTables, alpha_cc: torch.Tensor,
                 tables.eg16[c] = eg.reshape(2 * H, -1).to(torch.bfloat16)
             tables.delta[c] = lc[:, 1] - lc[:, 0]
             tables.s_base[c] = lc[:, 0].sum(0)
+        if tables.egw is not None:
+            from .pair import update_egw_rows
+            update_egw_rows(tables, rows.tolist())
         return tables
     sub = table_precompute(alpha_cc[:, rows], beta_cc[:, rows],
                            update_weight, tables.EG.shape[-1])
@@ -214,6 +221,9 @@ def table_update_rows(tables: EigTables, alpha_cc: torch.Tensor,
             rows.numel(), 2 * H, -1).to(torch.bfloat16)
     tables.delta[rows] = sub.delta
     tables.s_base[rows] = sub.s_base
+    if tables.egw is not None:
+        from .pair import update_egw_rows
+        update_egw_rows(tables, rows.tolist())
     return tables
 
 

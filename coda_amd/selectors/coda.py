@@ -61,9 +61,14 @@ class CODA(ModelSelector):
         self.num_points = num_points
         # EIG implementation: 'fused' = the wave-per-row HIP kernel (GPU) /
         # eager (CPU); 'table' = the factored per-step curve tables + MFMA
-        # GEMM (ops/table.py). 'auto' resolves per device at eig time.
+        # GEMM (ops/table.py); 'pair' = the hit-sparse pair engine
+        # (ops/pair.py - the full-pool no-prefilter fast path). 'auto'
+        # resolves per device at eig time.
         self.eig_impl = eig_impl
         self._tables = None          # persistent v2 curve tables
+        self._pairs_static = None    # v3 static hit structure (full pool)
+        self._pair_row_of = None     # point id -> row in the static set
+        self._active_mask = None     # (B_full,) bool device mask
         self._tables_dirty = set()   # class rows touched since last build
         self._posterior_version = 0  # bumped by add_label
         self._pbest_rows_cache = (-1, None)
@@ -229,10 +234,70 @@ class CODA(ModelSelector):
         self._pbest_rows_cache = (self._posterior_version, rows)
         return rows
 
+    def _refresh_tables(self, alpha_cc, beta_cc, want_egw: bool):
+        """Build or incrementally refresh the per-step curve tables."""
+        from ..ops import table as tops
+        if self._tables is None:
+            self._tables = tops.table_precompute(
+                alpha_cc, beta_cc, num_points=self.num_points)
+        elif self._tables_dirty:
+            tops.table_update_rows(self._tables, alpha_cc, beta_cc,
+                                   sorted(self._tables_dirty))
+        self._tables_dirty.clear()
+        if want_egw and self._tables.egw is None:
+            from ..ops import pair as pops
+            self._tables = self._tables._replace(
+                egw=pops.build_egw(self._tables))
+        return self._tables
+
+    def _eig_pair(self, candidate_ids):
+        """v3 hit-sparse EIG over the candidate set (ops/pair.py)."""
+        from ..ops import pair as pops
+        alpha_cc, beta_cc = ops.dirichlet_to_beta(self.dirichlets)
+        tables = self._refresh_tables(alpha_cc, beta_cc, want_egw=True)
+        pbest_before = self._pbest_rows_before()            # (C, H)
+        mixture0, H_before = ops.mixture_entropy(pbest_before, self.pi_hat)
+
+        if candidate_ids is self._active_candidates:
+            # full-pool acquisition: the hit structure is static (argmax
+            # classes never change); labeled points are masked at gather
+            if self._pairs_static is None:
+                ids = torch.tensor(list(candidate_ids), device=self.device)
+                cls_rows = self._global_classes(ids)
+                self._pairs_static = (
+                    pops.build_pairs(cls_rows, ids, self.C), cls_rows)
+                self._pair_row_of = {int(p): r
+                                     for r, p in enumerate(ids.tolist())}
+                self._active_mask = torch.ones(
+                    ids.numel(), dtype=torch.bool, device=self.device)
+            ps, cls_rows = self._pairs_static
+            cand = ps.cand_ids[self._active_mask]
+        else:
+            cand = torch.tensor(list(candidate_ids), device=self.device)
+            cls_rows = self._global_classes(cand)
+            ps = pops.build_pairs(cls_rows, cand, self.C)
+        eig_n = pops.eig_pairs(tables, ps, cls_rows, pbest_before,
+                               self.pi_hat, mixture0, H_before,
+                               self._adjusted, self._row_sums)
+        return eig_n[cand], candidate_ids
+
+    def _global_classes(self, ids: torch.Tensor) -> torch.Tensor:
+        """(B, H) int32 argmax classes of every GLOBAL model on the
+        given points (single-device: classes is already global)."""
+        return self.classes[:, ids].t().to(torch.int32).contiguous()
+
     def eig_batched(self):
         """EIG for every candidate point (reference: coda/coda.py:235-281)."""
         candidate_ids = self._prefilter(self.unlabeled_idxs) \
             or self.unlabeled_idxs
+
+        impl0 = self.eig_impl
+        if impl0 == "auto" and self.device.type == "cuda" \
+                and not self.comm.is_distributed and self.H <= 1024:
+            impl0 = "pair"
+        if impl0 == "pair" and not self.comm.is_distributed:
+            return self._eig_pair(candidate_ids)
+
         cand = torch.tensor(list(candidate_ids), device=self.device)
 
         pbest_before = self._pbest_rows_before()            # (C, Hl)
@@ -243,20 +308,14 @@ class CODA(ModelSelector):
         else:
             mixture0, H_before = ops.mixture_entropy(pbest_before, self.pi_hat)
 
-        impl = self.eig_impl
-        if impl == "auto":
+        impl = impl0
+        if impl == "auto" or impl == "pair":
             impl = "table" if self.device.type == "cuda" else "fused"
         tables = s_base_all = None
         if impl == "table":
             from ..ops import table as tops
-            if self._tables is None:
-                self._tables = tops.table_precompute(
-                    alpha_cc, beta_cc, num_points=self.num_points)
-            elif self._tables_dirty:
-                tops.table_update_rows(self._tables, alpha_cc, beta_cc,
-                                       sorted(self._tables_dirty))
-            self._tables_dirty.clear()
-            tables = self._tables
+            tables = self._refresh_tables(alpha_cc, beta_cc,
+                                          want_egw=False)
             if self.comm.is_distributed:
                 s_base_all = tops.s_base_global(tables, self.comm)
 
@@ -381,6 +440,12 @@ class CODA(ModelSelector):
                     .to(torch.bfloat16))
             t.delta.index_copy_(0, y_t, (lc[:, 1] - lc[:, 0]).unsqueeze(0))
             t.s_base.index_copy_(0, y_t, lc[:, 0].sum(0).unsqueeze(0))
+            if t.egw is not None:
+                esb = torch.exp2(t.s_base.index_select(0, y_t)) \
+                    * t.weights                             # (1, P)
+                t.egw.index_copy_(
+                    0, y_t, (eg.reshape(1, 2 * self.Hl, -1)
+                             * esb.unsqueeze(1)).to(torch.bfloat16))
         # posterior rows for the next acquisition / get_pbest
         alpha_cc, beta_cc = ops.dirichlet_to_beta(self.dirichlets)
         rows = ops.pbest_from_beta(alpha_cc.t().contiguous(),
@@ -441,10 +506,7 @@ class CODA(ModelSelector):
             self.labels.append(int(true_class))
             self.q_vals.append(selection_prob)
             self.unlabeled_idxs.remove(idx)
-            try:
-                self._active_candidates.remove(idx)
-            except ValueError:
-                pass
+            self._deactivate(idx)
             return
         onehot = torch.nn.functional.one_hot(
             self.classes[:, idx], self.C).to(self.dirichlets.dtype)  # (Hl, C)
@@ -460,10 +522,18 @@ class CODA(ModelSelector):
         self.labels.append(int(true_class))
         self.q_vals.append(selection_prob)
         self.unlabeled_idxs.remove(idx)
+        self._deactivate(idx)
+
+    def _deactivate(self, idx: int):
+        """Remove a point from the candidate pool (labeled or skipped)."""
         try:
             self._active_candidates.remove(idx)
         except ValueError:
-            pass  # labeled point was not a disagreeing candidate
+            pass  # point was not a disagreeing candidate
+        if self._pair_row_of is not None:
+            row = self._pair_row_of.get(idx)
+            if row is not None:
+                self._active_mask[row] = False
 
     def skip(self, idx):
         """Drop a point without labeling it (the demo's "I don't know"
@@ -471,10 +541,7 @@ class CODA(ModelSelector):
         idx = int(idx)
         if idx in self.unlabeled_idxs:
             self.unlabeled_idxs.remove(idx)
-        try:
-            self._active_candidates.remove(idx)
-        except ValueError:
-            pass
+        self._deactivate(idx)
 
     # ------------------------------------------------------------------
     def get_pbest(self):

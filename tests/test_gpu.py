@@ -456,3 +456,84 @@ class TestLabelGraph:
         assert t_e == t_g
         torch.testing.assert_close(p_e, p_g, rtol=1e-5, atol=1e-7)
         torch.testing.assert_close(pi_e, pi_g, rtol=1e-5, atol=1e-7)
+
+
+class TestPairKernels:
+    """v3 pair engine kernels (ops/hip/pair.hip) vs the eager pair math."""
+
+    def test_mfma_fragment_layout(self, dev):
+        """16x16x32 bf16 MFMA probe: A (16,32) x B^T rows (16,32) vs a
+        bf16-rounded torch matmul. Asymmetric operands catch row/col
+        swaps (guide section 3)."""
+        from coda_amd import ops
+        g = torch.Generator().manual_seed(7)
+        a = torch.randn(16, 32, generator=g).to(dev)
+        bt = (torch.arange(16 * 32, dtype=torch.float32).reshape(16, 32)
+              / 100.0 - 2.0).to(dev)
+        got = ops._ext.mfma_probe(a, bt).cpu()
+        want = (a.cpu().to(torch.bfloat16).float()
+                @ bt.cpu().to(torch.bfloat16).float().t())
+        torch.testing.assert_close(got, want, rtol=1e-2, atol=1e-2)
+
+    def test_pair_kernels_vs_eager(self, dev):
+        """pair_dsum_es + pair_gemm_entropy vs the fp32 eager pair
+        formulation, through the full eig_pairs dispatch."""
+        from coda_amd.ops import pair as pops
+        from coda_amd.ops import table as tops
+        from coda_amd.ops import reference as R
+        from tests.test_pair import _random_problem
+        for H, N, C in [(10, 120, 7), (3, 60, 2), (16, 200, 126),
+                        (128, 500, 50)]:
+            (preds, cls, dirichlets, pi_hat, adjusted,
+             row_sums) = _random_problem(H, N, C, seed=H + C)
+            alpha_cc, beta_cc = R.dirichlet_to_beta(dirichlets)
+            alpha_cc, beta_cc = alpha_cc.to(dev), beta_cc.to(dev)
+            tables = tops.table_precompute(alpha_cc, beta_cc)
+            tables = tables._replace(egw=pops.build_egw(tables))
+            pbest_before = R.pbest_from_beta(
+                alpha_cc.t().contiguous(), beta_cc.t().contiguous())
+            mixture0, H_before = R.mixture_entropy(
+                pbest_before, pi_hat.to(dev))
+            ids = torch.arange(N, device=dev)
+            cls_rows = cls.to(dev)[:, ids].t().contiguous()
+            ps = pops.build_pairs(cls_rows, ids, C)
+            eig_k = pops.eig_pairs(
+                tables, ps, cls_rows, pbest_before, pi_hat.to(dev),
+                mixture0, H_before, adjusted.to(dev), row_sums.to(dev))
+            h_eager = pops.pair_h_after(
+                tables, ps, cls_rows, pbest_before, pi_hat.to(dev),
+                mixture0)
+            eig_e = pops.eig_from_pairs(
+                h_eager, ps, adjusted.to(dev), row_sums.to(dev),
+                H_before)
+            # bf16 GEMM operands: absolute tolerance at the EIG scale
+            torch.testing.assert_close(eig_k, eig_e, rtol=5e-3,
+                                       atol=5e-5)
+
+    def test_pair_trajectory_matches_table(self, dev):
+        """Full selector on cuda: eig_impl='pair' (the GPU default) vs
+        'table' - same selections over 8 steps of the no-prefilter
+        acquisition."""
+        import bench
+        from coda_amd import CODA, Oracle
+        from coda_amd.datasets import Dataset
+        from coda_amd.options import LOSS_FNS
+
+        preds, labels = bench.synth_preds(list(range(16)), 800, 12, dev)
+        ds = Dataset.from_tensors(preds, labels, dev)
+        oracle = Oracle(ds, LOSS_FNS["acc"])
+
+        def run(impl):
+            random.seed(0); torch.manual_seed(0)
+            sel = CODA(ds, eig_impl=impl)
+            traj = []
+            for _ in range(8):
+                i, q = sel.get_next_item_to_label()
+                sel.add_label(i, oracle(int(i)), q)
+                traj.append((int(i), int(sel.get_best_model_prediction())))
+            return traj, sel.get_pbest().cpu()
+
+        t_t, p_t = run("table")
+        t_p, p_p = run("pair")
+        assert t_t == t_p
+        torch.testing.assert_close(p_t, p_p, rtol=1e-3, atol=1e-5)
