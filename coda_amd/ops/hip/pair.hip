@@ -2,25 +2,32 @@
 // the math and the sparsity argument; reference semantics:
 // coda/coda.py:150-168 + :235-281 restricted to hit cells).
 //
-// Two kernels per acquisition step, covering the WHOLE candidate pool:
+// Three kernels per acquisition step, covering the WHOLE candidate pool:
 //
 //   pair_dsum_es_kernel    one wave per (candidate, class) hit pair:
 //                          sums the pair's selected log2-cdf delta
-//                          curves and exponentiates -> the bf16 MFMA A
-//                          operand (K, P). The baseline curve
-//                          exp2(s_base)*w is folded into the B operand
-//                          (egw, built host-side once per step delta).
+//                          curves (fp16 table - traffic-bound) and
+//                          exponentiates -> the bf16 MFMA A operand
+//                          (K, P). The baseline curve exp2(s_base)*w is
+//                          folded into the B operand (egw).
 //
-//   pair_gemm_entropy_kernel
-//                          one workgroup per 16-pair class-uniform
-//                          tile: bf16 MFMA (16 x 2H x P) pairing GEMM
-//                          with the variant-select + normalize + log2
+//   pair_gemm_entropy{16,64}_kernel
+//                          one workgroup per class-uniform pair tile:
+//                          bf16 MFMA (tile x 2H x P) pairing GEMM with
+//                          the variant-select + normalize + log2
 //                          entropy epilogue fused (the (K, 2H) M tensor
-//                          never reaches global memory). A operand
-//                          staged in LDS; B operand (egw rows of the
-//                          tile's class) streamed from L2 - tiles are
-//                          class-sorted so each XCD reads a class's
-//                          131 KB table once.
+//                          never reaches global memory). The 64-pair
+//                          variant (2H <= 512) stages both operands in
+//                          LDS, quartering the dominant B traffic vs
+//                          the 16-pair tile; the 16-pair variant covers
+//                          H up to 1024.
+//
+//   pair_eig_finalize_kernel
+//                          one wave per candidate: EIG[b] = H_before -
+//                          (pi_xi row) . h_base - sum over the
+//                          candidate's hit pairs of pi_xi*(h_after -
+//                          h_base), in a fixed lane-strided order
+//                          (deterministic - no atomics).
 //
 // MFMA: v_mfma_f32_16x16x32_bf16. Lane mapping (cdna_hip_programming.md
 // section 3): A[i][k] i=lane&15, k=(lane>>4)*8+e; B[k][j] j=lane&15,
@@ -36,12 +43,12 @@
 
 #define P_POINTS 256
 #define BLOCK 256
-#define PAIR_TILE 16
 
 namespace pairops {
 
 typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
+typedef _Float16 half4v __attribute__((ext_vector_type(4)));
 
 __device__ __forceinline__ float wave_reduce(float v) {
 #pragma unroll
@@ -51,14 +58,14 @@ __device__ __forceinline__ float wave_reduce(float v) {
 }
 
 // ---------------------------------------------------------------------
-// A-operand build: a16[k, p] = 2^(sum_{h in seg(k)} delta[c_k, h, p])
+// A-operand build: a16[k, p] = 2^(sum_{h in seg(k)} delta16[c_k, h, p])
 // ---------------------------------------------------------------------
 __global__ void __launch_bounds__(BLOCK)
-pair_dsum_es_kernel(const float* __restrict__ delta,   // (C, H, P)
-                    const int* __restrict__ pair_c,    // (K,)
-                    const int* __restrict__ seg_off,   // (K+1,)
-                    const int* __restrict__ seg_h,     // (S,)
-                    hip_bfloat16* __restrict__ a16,  // (K, P)
+pair_dsum_es_kernel(const _Float16* __restrict__ delta16,  // (C, H, P)
+                    const int* __restrict__ pair_c,        // (K,)
+                    const int* __restrict__ seg_off,       // (K+1,)
+                    const int* __restrict__ seg_h,         // (S,)
+                    hip_bfloat16* __restrict__ a16,        // (K, P)
                     int K, int H) {
     const int k = blockIdx.x * 4 + (threadIdx.x >> 6);
     if (k >= K) return;
@@ -74,26 +81,25 @@ pair_dsum_es_kernel(const float* __restrict__ delta,   // (C, H, P)
           a2[4] = {0.f, 0.f, 0.f, 0.f}, a3[4] = {0.f, 0.f, 0.f, 0.f};
     int s = s0;
     for (; s + 3 < s1; s += 4) {
-        const int ha = seg_h[s], hb = seg_h[s + 1];
-        const int hc = seg_h[s + 2], hd = seg_h[s + 3];
-        const float4 da = *reinterpret_cast<const float4*>(
-            delta + dbase + (size_t)ha * P_POINTS);
-        const float4 db = *reinterpret_cast<const float4*>(
-            delta + dbase + (size_t)hb * P_POINTS);
-        const float4 dc = *reinterpret_cast<const float4*>(
-            delta + dbase + (size_t)hc * P_POINTS);
-        const float4 dd = *reinterpret_cast<const float4*>(
-            delta + dbase + (size_t)hd * P_POINTS);
-        a0[0] += da.x; a0[1] += da.y; a0[2] += da.z; a0[3] += da.w;
-        a1[0] += db.x; a1[1] += db.y; a1[2] += db.z; a1[3] += db.w;
-        a2[0] += dc.x; a2[1] += dc.y; a2[2] += dc.z; a2[3] += dc.w;
-        a3[0] += dd.x; a3[1] += dd.y; a3[2] += dd.z; a3[3] += dd.w;
+        const half4v da = *reinterpret_cast<const half4v*>(
+            delta16 + dbase + (size_t)seg_h[s] * P_POINTS);
+        const half4v db = *reinterpret_cast<const half4v*>(
+            delta16 + dbase + (size_t)seg_h[s + 1] * P_POINTS);
+        const half4v dc = *reinterpret_cast<const half4v*>(
+            delta16 + dbase + (size_t)seg_h[s + 2] * P_POINTS);
+        const half4v dd = *reinterpret_cast<const half4v*>(
+            delta16 + dbase + (size_t)seg_h[s + 3] * P_POINTS);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            a0[j] += (float)da[j]; a1[j] += (float)db[j];
+            a2[j] += (float)dc[j]; a3[j] += (float)dd[j];
+        }
     }
     for (; s < s1; ++s) {
-        const int h = seg_h[s];
-        const float4 d = *reinterpret_cast<const float4*>(
-            delta + dbase + (size_t)h * P_POINTS);
-        a0[0] += d.x; a0[1] += d.y; a0[2] += d.z; a0[3] += d.w;
+        const half4v d = *reinterpret_cast<const half4v*>(
+            delta16 + dbase + (size_t)seg_h[s] * P_POINTS);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) a0[j] += (float)d[j];
     }
     ushort4 out;
     unsigned short* o = reinterpret_cast<unsigned short*>(&out);
@@ -106,34 +112,32 @@ pair_dsum_es_kernel(const float* __restrict__ delta,   // (C, H, P)
 }
 
 // ---------------------------------------------------------------------
-// Pairing GEMM + fused entropy epilogue.
-// Tile: PAIR_TILE pairs (class-uniform by construction) x 2H columns,
-// K-loop over P in steps of 32. M stays in LDS; h_after (K,) out.
+// 16-pair tile GEMM+entropy (any H <= 1024): B streamed from L2.
 // ---------------------------------------------------------------------
 __global__ void __launch_bounds__(BLOCK)
-pair_gemm_entropy_kernel(const hip_bfloat16* __restrict__ a16,  // (K, P)
-                         const hip_bfloat16* __restrict__ egw,  // (C, 2H, P)
-                         const int* __restrict__ pair_b,          // (K,)
-                         const int* __restrict__ pair_c,          // (K,)
-                         const int* __restrict__ cls,             // (B, H)
-                         const float* __restrict__ pi_hat,        // (C,)
-                         const float* __restrict__ pbest_before,  // (C, H)
-                         const float* __restrict__ mixture0,      // (H,)
-                         float* __restrict__ h_after,             // (K,)
-                         int H, int mstride) {
+pair_gemm_entropy16_kernel(const hip_bfloat16* __restrict__ a16,
+                           const hip_bfloat16* __restrict__ egw,
+                           const int* __restrict__ pair_b,
+                           const int* __restrict__ pair_c,
+                           const int* __restrict__ cls,
+                           const float* __restrict__ pi_hat,
+                           const float* __restrict__ pbest_before,
+                           const float* __restrict__ mixture0,
+                           float* __restrict__ h_after,
+                           int H, int mstride) {
     extern __shared__ char smem[];
     hip_bfloat16* a_lds = reinterpret_cast<hip_bfloat16*>(smem);
     float* m_tile = reinterpret_cast<float*>(
-        smem + PAIR_TILE * P_POINTS * sizeof(hip_bfloat16));
+        smem + 16 * P_POINTS * sizeof(hip_bfloat16));
 
-    const int k0 = blockIdx.x * PAIR_TILE;
+    const int k0 = blockIdx.x * 16;
     const int c = pair_c[k0];
     const int tid = threadIdx.x;
     const int wave = tid >> 6, lane = tid & 63;
-    const int row16 = lane & 15;       // A row / B col / D col
-    const int kgrp = lane >> 4;        // K sub-group (0..3)
+    const int row16 = lane & 15;
+    const int kgrp = lane >> 4;
 
-    {   // stage A tile: 16 rows x 256 bf16 = 8 KB, 16 B per thread x2
+    {   // stage A tile: 16 rows x 256 bf16 = 8 KB
         const uint4* g = reinterpret_cast<const uint4*>(
             a16 + (size_t)k0 * P_POINTS);
         uint4* d = reinterpret_cast<uint4*>(a_lds);
@@ -169,8 +173,7 @@ pair_gemm_entropy_kernel(const hip_bfloat16* __restrict__ a16,  // (K, P)
     }
     __syncthreads();
 
-    // epilogue: 4 pairs per wave; v-select + normalize + entropy
-    for (int pi = wave; pi < PAIR_TILE; pi += 4) {
+    for (int pi = wave; pi < 16; pi += 4) {
         const int k = k0 + pi;
         const int b = pair_b[k];
         const float* mrow = m_tile + (size_t)pi * mstride;
@@ -194,6 +197,155 @@ pair_gemm_entropy_kernel(const hip_bfloat16* __restrict__ a16,  // (K, P)
         ent = wave_reduce(ent);
         if (lane == 0) h_after[k] = ent;
     }
+}
+
+// ---------------------------------------------------------------------
+// 64-pair tile GEMM+entropy (2H <= 512): A AND B staged in LDS, each
+// egw[c] chunk read once per 64 pairs. JT = #16-col tiles is a template
+// parameter so the accumulator array stays in registers.
+// ---------------------------------------------------------------------
+#define ASTRIDE (P_POINTS + 8)   // bf16 elems; +8 breaks bank alignment
+#define BSTRIDE (32 + 8)
+
+template <int JT>
+__global__ void __launch_bounds__(BLOCK)
+pair_gemm_entropy64_kernel(const hip_bfloat16* __restrict__ a16,
+                           const hip_bfloat16* __restrict__ egw,
+                           const int* __restrict__ pair_b,
+                           const int* __restrict__ pair_c,
+                           const int* __restrict__ cls,
+                           const float* __restrict__ pi_hat,
+                           const float* __restrict__ pbest_before,
+                           const float* __restrict__ mixture0,
+                           float* __restrict__ h_after,
+                           int H, int mstride) {
+    extern __shared__ char smem[];
+    hip_bfloat16* a_lds = reinterpret_cast<hip_bfloat16*>(smem);
+    hip_bfloat16* b_lds = a_lds + 64 * ASTRIDE;
+
+    const int k0 = blockIdx.x * 64;
+    const int c = pair_c[k0];
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6, lane = tid & 63;
+    const int row16 = lane & 15;
+    const int kgrp = lane >> 4;
+    const int twoH = 2 * H;
+
+    {   // stage A: 64 rows x 256 bf16 (padded rows); 128 B per thread
+        const int row = tid >> 2, quarter = tid & 3;
+        const uint4* g = reinterpret_cast<const uint4*>(
+            a16 + (size_t)(k0 + row) * P_POINTS + quarter * 64);
+        uint4* d = reinterpret_cast<uint4*>(
+            reinterpret_cast<char*>(a_lds)
+            + (size_t)row * ASTRIDE * 2 + quarter * 128);
+        d[0] = g[0]; d[1] = g[1]; d[2] = g[2]; d[3] = g[3];
+    }
+
+    f32x4 acc[JT];
+#pragma unroll
+    for (int jt = 0; jt < JT; ++jt) acc[jt] = {0.f, 0.f, 0.f, 0.f};
+
+    for (int kk = 0; kk < P_POINTS; kk += 32) {
+        __syncthreads();
+        // stage the B chunk: egw[c, j, kk..kk+31] for all j (64 B/row)
+        for (int j = tid; j < twoH; j += BLOCK) {
+            const uint4* g = reinterpret_cast<const uint4*>(
+                egw + ((size_t)c * twoH + j) * P_POINTS + kk);
+            uint4* d = reinterpret_cast<uint4*>(
+                reinterpret_cast<char*>(b_lds) + (size_t)j * BSTRIDE * 2);
+            d[0] = g[0]; d[1] = g[1]; d[2] = g[2]; d[3] = g[3];
+        }
+        __syncthreads();
+        const bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
+            a_lds + (size_t)(wave * 16 + row16) * ASTRIDE + kk + kgrp * 8);
+#pragma unroll
+        for (int jt = 0; jt < JT; ++jt) {
+            const int j = jt * 16 + row16;
+            bf16x8 bfrag = {};
+            if (j < twoH)
+                bfrag = *reinterpret_cast<const bf16x8*>(
+                    b_lds + (size_t)j * BSTRIDE + kgrp * 8);
+            acc[jt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afrag, bfrag, acc[jt], 0, 0, 0);
+        }
+    }
+    __syncthreads();
+
+    // phase 2: spill accumulators to LDS (overlapping the A/B buffers)
+    float* m_tile = reinterpret_cast<float*>(smem);
+#pragma unroll
+    for (int jt = 0; jt < JT; ++jt) {
+        const int j = jt * 16 + row16;
+        if (j < twoH) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r)
+                m_tile[(size_t)(wave * 16 + kgrp * 4 + r) * mstride + j]
+                    = acc[jt][r];
+        }
+    }
+    __syncthreads();
+
+    for (int pi = wave; pi < 64; pi += 4) {
+        const int k = k0 + pi;
+        const int b = pair_b[k];
+        const float* mrow = m_tile + (size_t)pi * mstride;
+        float tot = 0.f;
+        for (int h = lane; h < H; h += 64) {
+            const int v = (b >= 0 && cls[(size_t)b * H + h] == c) ? 1 : 0;
+            tot += mrow[2 * h + v];
+        }
+        tot = wave_reduce(tot);
+        const float inv = 1.0f / fmaxf(tot, 1e-30f);
+        const float pic = pi_hat[c];
+        float ent = 0.f;
+        for (int h = lane; h < H; h += 64) {
+            const int v = (b >= 0 && cls[(size_t)b * H + h] == c) ? 1 : 0;
+            const float pb = mrow[2 * h + v] * inv;
+            const float mm = fmaxf(
+                mixture0[h] + pic * (pb - pbest_before[(size_t)c * H + h]),
+                1e-12f);
+            ent += -mm * __log2f(mm);
+        }
+        ent = wave_reduce(ent);
+        if (lane == 0) h_after[k] = ent;
+    }
+}
+
+// ---------------------------------------------------------------------
+// Finalize: q[b] = H_before - (1/rowsum) * (arow . h_base
+//                 + sum_{pairs of b} arow[c]*(h_after - h_base[c]))
+// One wave per candidate; fixed reduction order (deterministic).
+// ---------------------------------------------------------------------
+__global__ void __launch_bounds__(BLOCK)
+pair_eig_finalize_kernel(const float* __restrict__ h_after,   // (K,)
+                         const int* __restrict__ pair_c,      // (K,)
+                         const int* __restrict__ base_pos,    // (C,)
+                         const int* __restrict__ cand_off,    // (B+1,)
+                         const int* __restrict__ cand_pairs,  // (n_real,)
+                         const long* __restrict__ cand_ids,   // (B,)
+                         const float* __restrict__ adjusted,  // (N, C)
+                         const float* __restrict__ row_sums,  // (N,)
+                         float H_before,
+                         float* __restrict__ q,               // (B,)
+                         int B, int C) {
+    const int b = blockIdx.x * 4 + (threadIdx.x >> 6);
+    if (b >= B) return;
+    const int lane = threadIdx.x & 63;
+    const long id = cand_ids[b];
+    const float inv = 1.0f / fmaxf(row_sums[id], 1e-12f);
+    const float* arow = adjusted + (size_t)id * C;
+    float base = 0.f;
+    for (int cc = lane; cc < C; cc += 64)
+        base += arow[cc] * h_after[base_pos[cc]];
+    float corr = 0.f;
+    const int s1 = cand_off[b + 1];
+    for (int s = cand_off[b] + lane; s < s1; s += 64) {
+        const int k = cand_pairs[s];
+        const int cc = pair_c[k];
+        corr += arow[cc] * (h_after[k] - h_after[base_pos[cc]]);
+    }
+    const float tot = wave_reduce(base + corr);
+    if (lane == 0) q[b] = H_before - tot * inv;
 }
 
 // MFMA layout probe (correctness insurance, not a production op):
@@ -224,20 +376,21 @@ __global__ void mfma_probe_kernel(const float* __restrict__ a,   // (16,32)
 // Host bindings
 // ---------------------------------------------------------------------
 
-torch::Tensor pair_dsum_es(torch::Tensor delta, torch::Tensor pair_c,
+torch::Tensor pair_dsum_es(torch::Tensor delta16, torch::Tensor pair_c,
                            torch::Tensor seg_off, torch::Tensor seg_h) {
-    TORCH_CHECK(delta.is_cuda() && delta.dtype() == torch::kFloat32);
-    TORCH_CHECK(delta.size(-1) == P_POINTS);
-    const int C = delta.size(0), H = delta.size(1);
+    TORCH_CHECK(delta16.is_cuda() && delta16.dtype() == torch::kFloat16);
+    TORCH_CHECK(delta16.size(-1) == P_POINTS);
+    const int H = delta16.size(1);
     const int K = pair_c.size(0);
-    (void)C;
     auto a16 = torch::empty({K, P_POINTS},
-                            delta.options().dtype(torch::kBFloat16));
+                            delta16.options().dtype(torch::kBFloat16));
     auto stream = c10::hip::getCurrentHIPStream();
     dim3 grid((K + 3) / 4);
     hipLaunchKernelGGL(pairops::pair_dsum_es_kernel, grid, dim3(BLOCK), 0,
                        stream.stream(),
-                       delta.data_ptr<float>(), pair_c.data_ptr<int>(),
+                       reinterpret_cast<const _Float16*>(
+                           delta16.data_ptr()),
+                       pair_c.data_ptr<int>(),
                        seg_off.data_ptr<int>(), seg_h.data_ptr<int>(),
                        reinterpret_cast<hip_bfloat16*>(a16.data_ptr()),
                        K, H);
@@ -248,33 +401,85 @@ torch::Tensor pair_gemm_entropy(torch::Tensor a16, torch::Tensor egw,
                                 torch::Tensor pair_b, torch::Tensor pair_c,
                                 torch::Tensor cls, torch::Tensor pi_hat,
                                 torch::Tensor pbest_before,
-                                torch::Tensor mixture0) {
+                                torch::Tensor mixture0, int64_t tile) {
     TORCH_CHECK(a16.is_cuda() && a16.dtype() == torch::kBFloat16);
     TORCH_CHECK(egw.dtype() == torch::kBFloat16);
     const int K = a16.size(0);
     const int H = mixture0.size(0);
-    TORCH_CHECK(K % PAIR_TILE == 0, "pair count must be tile-padded");
+    TORCH_CHECK(K % tile == 0, "pair count must be tile-padded");
     TORCH_CHECK(egw.size(1) == 2 * H);
     const int mstride = 2 * H + 4;  // LDS row pad against bank conflicts
-    const size_t shmem = PAIR_TILE * P_POINTS * sizeof(hip_bfloat16)
-                       + (size_t)PAIR_TILE * mstride * sizeof(float);
-    TORCH_CHECK(shmem <= 160 * 1024, "H too large for the fused pair "
-                "kernel (use the table engine beyond H=1024)");
     auto h_after = torch::empty({K}, pi_hat.options());
     auto stream = c10::hip::getCurrentHIPStream();
-    hipLaunchKernelGGL(pairops::pair_gemm_entropy_kernel,
-                       dim3(K / PAIR_TILE), dim3(BLOCK), shmem,
-                       stream.stream(),
-                       reinterpret_cast<const hip_bfloat16*>(
-                           a16.data_ptr()),
-                       reinterpret_cast<const hip_bfloat16*>(
-                           egw.data_ptr()),
-                       pair_b.data_ptr<int>(), pair_c.data_ptr<int>(),
-                       cls.data_ptr<int>(), pi_hat.data_ptr<float>(),
-                       pbest_before.data_ptr<float>(),
-                       mixture0.data_ptr<float>(),
-                       h_after.data_ptr<float>(), H, mstride);
+    const auto ab16 = reinterpret_cast<const hip_bfloat16*>(
+        a16.data_ptr());
+    const auto eb16 = reinterpret_cast<const hip_bfloat16*>(
+        egw.data_ptr());
+    if (tile == 64) {
+        TORCH_CHECK(2 * H <= 512, "64-pair tile needs 2H <= 512");
+        const size_t phase1 = 64 * ASTRIDE * sizeof(hip_bfloat16)
+                            + (size_t)2 * H * BSTRIDE
+                                * sizeof(hip_bfloat16);
+        const size_t phase2 = (size_t)64 * mstride * sizeof(float);
+        const size_t shmem = std::max(phase1, phase2);
+        const int JT = (2 * H + 15) / 16;
+        auto launch = [&](auto kern) {
+            hipLaunchKernelGGL(kern, dim3(K / 64), dim3(BLOCK), shmem,
+                               stream.stream(), ab16, eb16,
+                               pair_b.data_ptr<int>(),
+                               pair_c.data_ptr<int>(),
+                               cls.data_ptr<int>(),
+                               pi_hat.data_ptr<float>(),
+                               pbest_before.data_ptr<float>(),
+                               mixture0.data_ptr<float>(),
+                               h_after.data_ptr<float>(), H, mstride);
+        };
+        if (JT <= 4) launch(pairops::pair_gemm_entropy64_kernel<4>);
+        else if (JT <= 8) launch(pairops::pair_gemm_entropy64_kernel<8>);
+        else if (JT <= 16)
+            launch(pairops::pair_gemm_entropy64_kernel<16>);
+        else launch(pairops::pair_gemm_entropy64_kernel<32>);
+    } else {
+        TORCH_CHECK(tile == 16, "tile must be 16 or 64");
+        const size_t shmem = 16 * P_POINTS * sizeof(hip_bfloat16)
+                           + (size_t)16 * mstride * sizeof(float);
+        TORCH_CHECK(shmem <= 160 * 1024, "H too large for the fused "
+                    "pair kernel (use the table engine beyond H=1024)");
+        hipLaunchKernelGGL(pairops::pair_gemm_entropy16_kernel,
+                           dim3(K / 16), dim3(BLOCK), shmem,
+                           stream.stream(), ab16, eb16,
+                           pair_b.data_ptr<int>(), pair_c.data_ptr<int>(),
+                           cls.data_ptr<int>(), pi_hat.data_ptr<float>(),
+                           pbest_before.data_ptr<float>(),
+                           mixture0.data_ptr<float>(),
+                           h_after.data_ptr<float>(), H, mstride);
+    }
     return h_after;
+}
+
+torch::Tensor pair_eig_finalize(torch::Tensor h_after,
+                                torch::Tensor pair_c,
+                                torch::Tensor base_pos,
+                                torch::Tensor cand_off,
+                                torch::Tensor cand_pairs,
+                                torch::Tensor cand_ids,
+                                torch::Tensor adjusted,
+                                torch::Tensor row_sums,
+                                double H_before) {
+    const int B = cand_ids.size(0);
+    const int C = adjusted.size(1);
+    auto q = torch::empty({B}, adjusted.options());
+    auto stream = c10::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(pairops::pair_eig_finalize_kernel,
+                       dim3((B + 3) / 4), dim3(BLOCK), 0, stream.stream(),
+                       h_after.data_ptr<float>(), pair_c.data_ptr<int>(),
+                       base_pos.data_ptr<int>(), cand_off.data_ptr<int>(),
+                       cand_pairs.data_ptr<int>(),
+                       cand_ids.data_ptr<long>(),
+                       adjusted.data_ptr<float>(),
+                       row_sums.data_ptr<float>(),
+                       (float)H_before, q.data_ptr<float>(), B, C);
+    return q;
 }
 
 torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor bt) {
@@ -293,6 +498,8 @@ void register_pair_ops(pybind11::module_& m) {
           "v3 pair A-operand: 2^(summed delta curves) -> (K, P) bf16");
     m.def("pair_gemm_entropy", &pair_gemm_entropy,
           "v3 fused pairing MFMA GEMM + entropy epilogue -> (K,)");
+    m.def("pair_eig_finalize", &pair_eig_finalize,
+          "v3 per-candidate EIG assembly (deterministic) -> (B,)");
     m.def("mfma_probe", &mfma_probe,
           "16x16x32 bf16 MFMA fragment-layout probe");
 }

