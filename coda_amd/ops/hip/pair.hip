@@ -238,7 +238,8 @@ pair_gemm_entropy64_kernel(const hip_bfloat16* __restrict__ a16,
         uint4* d = reinterpret_cast<uint4*>(
             reinterpret_cast<char*>(a_lds)
             + (size_t)row * ASTRIDE * 2 + quarter * 128);
-        d[0] = g[0]; d[1] = g[1]; d[2] = g[2]; d[3] = g[3];
+#pragma unroll
+        for (int i = 0; i < 8; ++i) d[i] = g[i];
     }
 
     f32x4 acc[JT];

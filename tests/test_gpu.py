@@ -508,14 +508,20 @@ class TestPairKernels:
                 H_before)[ps.cand_ids]
             # bf16 GEMM operands; EIG differences are where the signal
             # (h_after - h_base) loses a digit to bf16 rounding, so the
-            # tolerance is at the EIG-value scale, not elementwise
-            scale = eig_e.abs().max()
-            assert float((eig_k - eig_e).abs().max()) < 2e-2 * scale
+            # tolerance is at the EIG-value scale with an absolute floor
+            # for near-zero-EIG shapes (round-1 calibration: max rel
+            # ~6e-4 at O(1) scales, abs ~2e-4 worst case)
+            scale = float(eig_e.abs().max())
+            err = float((eig_k - eig_e).abs().max())
+            assert err < max(5e-3 * scale, 3e-4), (H, N, C, err, scale)
 
     def test_pair_trajectory_matches_table(self, dev):
         """Full selector on cuda: eig_impl='pair' (the GPU default) vs
-        'table' - same selections over 8 steps of the no-prefilter
-        acquisition."""
+        'table' over the no-prefilter acquisition. Both run bf16
+        pairing GEMMs with different factorizations, so candidates
+        whose EIG values are within the bf16 noise band may swap rank -
+        selections must agree except at such near-ties, and the EIG
+        vectors themselves must agree to GEMM tolerance."""
         import bench
         from coda_amd import CODA, Oracle
         from coda_amd.datasets import Dataset
@@ -528,14 +534,29 @@ class TestPairKernels:
         def run(impl):
             random.seed(0); torch.manual_seed(0)
             sel = CODA(ds, eig_impl=impl)
-            traj = []
+            traj, qs, cands = [], [], []
             for _ in range(8):
+                q_vals, cand = sel.eig_batched()
+                qs.append(q_vals.detach().clone())
+                cands.append(list(cand))
                 i, q = sel.get_next_item_to_label()
                 sel.add_label(i, oracle(int(i)), q)
-                traj.append((int(i), int(sel.get_best_model_prediction())))
-            return traj, sel.get_pbest().cpu()
+                traj.append(int(i))
+            return traj, qs, cands
 
-        t_t, p_t = run("table")
-        t_p, p_p = run("pair")
-        assert t_t == t_p
-        torch.testing.assert_close(p_t, p_p, rtol=1e-3, atol=1e-5)
+        t_t, q_t, c_t = run("table")
+        t_p, q_p, c_p = run("pair")
+        for s in range(8):
+            if t_t[s] == t_p[s]:
+                assert c_t[s] == c_p[s]
+                scale = float(q_t[s].max())
+                assert float((q_t[s] - q_p[s]).abs().max()) < 3e-3 * scale
+                continue
+            # first divergence: must be a near-tie in the table engine's
+            # own values; states differ beyond this step - stop
+            pos_p = c_t[s].index(t_p[s])
+            pos_t = c_t[s].index(t_t[s])
+            gap = float((q_t[s][pos_t] - q_t[s][pos_p]).abs())
+            assert gap < 2e-3 * float(q_t[s].max()), \
+                (s, t_t[s], t_p[s], gap)
+            break
