@@ -50,12 +50,35 @@ typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
 typedef _Float16 half4v __attribute__((ext_vector_type(4)));
 
+#ifdef CODA_DPP_SCAN
+// Wave64 sum via the GCN DPP row_shr/row_bcast ladder: 6 VALU ops at
+// 1-2 cycle dependent latency each, vs ~7 dependent ds_bpermute rounds
+// (~30+ cycles each) for the __shfl ladder. Same ladder as pbest.hip.
+template <int CTRL, int ROW_MASK>
+__device__ __forceinline__ float dpp_add(float x) {
+    int moved = __builtin_amdgcn_update_dpp(0, __float_as_int(x), CTRL,
+                                            ROW_MASK, 0xf, true);
+    return x + __int_as_float(moved);
+}
+
+__device__ __forceinline__ float wave_reduce(float v) {
+    v = dpp_add<0x111, 0xf>(v);  // row_shr:1
+    v = dpp_add<0x112, 0xf>(v);  // row_shr:2
+    v = dpp_add<0x114, 0xf>(v);  // row_shr:4
+    v = dpp_add<0x118, 0xf>(v);  // row_shr:8
+    v = dpp_add<0x142, 0xa>(v);  // row_bcast:15
+    v = dpp_add<0x143, 0xc>(v);  // row_bcast:31
+    return __int_as_float(__builtin_amdgcn_readlane(__float_as_int(v),
+                                                    63));
+}
+#else
 __device__ __forceinline__ float wave_reduce(float v) {
 #pragma unroll
     for (int off = 32; off > 0; off >>= 1)
         v += __shfl_down(v, off, 64);
     return __shfl(v, 0, 64);
 }
+#endif
 
 // ---------------------------------------------------------------------
 // A-operand build: a16[k, p] = 2^(sum_{h in seg(k)} delta16[c_k, h, p])
@@ -220,8 +243,10 @@ pair_gemm_entropy64_kernel(const hip_bfloat16* __restrict__ a16,
                            float* __restrict__ h_after,
                            int H, int mstride) {
     extern __shared__ char smem[];
+    const int twoH = 2 * H;
     hip_bfloat16* a_lds = reinterpret_cast<hip_bfloat16*>(smem);
-    hip_bfloat16* b_lds = a_lds + 64 * ASTRIDE;
+    hip_bfloat16* b_lds[2] = {a_lds + 64 * ASTRIDE,
+                              a_lds + 64 * ASTRIDE + twoH * BSTRIDE};
 
     const int k0 = blockIdx.x * 64;
     const int c = pair_c[k0];
@@ -229,7 +254,7 @@ pair_gemm_entropy64_kernel(const hip_bfloat16* __restrict__ a16,
     const int wave = tid >> 6, lane = tid & 63;
     const int row16 = lane & 15;
     const int kgrp = lane >> 4;
-    const int twoH = 2 * H;
+    const hip_bfloat16* egw_c = egw + (size_t)c * twoH * P_POINTS;
 
     {   // stage A: 64 rows x 256 bf16 (padded rows); 128 B per thread
         const int row = tid >> 2, quarter = tid & 3;
@@ -241,22 +266,36 @@ pair_gemm_entropy64_kernel(const hip_bfloat16* __restrict__ a16,
 #pragma unroll
         for (int i = 0; i < 8; ++i) d[i] = g[i];
     }
+    // prologue: chunk 0 into buffer 0
+    for (int j = tid; j < twoH; j += BLOCK) {
+        const uint4* g = reinterpret_cast<const uint4*>(
+            egw_c + (size_t)j * P_POINTS);
+        uint4* d = reinterpret_cast<uint4*>(
+            reinterpret_cast<char*>(b_lds[0]) + (size_t)j * BSTRIDE * 2);
+        d[0] = g[0]; d[1] = g[1]; d[2] = g[2]; d[3] = g[3];
+    }
+    __syncthreads();
 
     f32x4 acc[JT];
 #pragma unroll
     for (int jt = 0; jt < JT; ++jt) acc[jt] = {0.f, 0.f, 0.f, 0.f};
 
+    // double-buffered K loop: one barrier per chunk, next-chunk loads
+    // in flight under the MFMAs
     for (int kk = 0; kk < P_POINTS; kk += 32) {
-        __syncthreads();
-        // stage the B chunk: egw[c, j, kk..kk+31] for all j (64 B/row)
-        for (int j = tid; j < twoH; j += BLOCK) {
-            const uint4* g = reinterpret_cast<const uint4*>(
-                egw + ((size_t)c * twoH + j) * P_POINTS + kk);
-            uint4* d = reinterpret_cast<uint4*>(
-                reinterpret_cast<char*>(b_lds) + (size_t)j * BSTRIDE * 2);
-            d[0] = g[0]; d[1] = g[1]; d[2] = g[2]; d[3] = g[3];
+        const int cur = (kk >> 5) & 1;
+        if (kk + 32 < P_POINTS) {
+            const hip_bfloat16* src = egw_c + kk + 32;
+            hip_bfloat16* dst = b_lds[cur ^ 1];
+            for (int j = tid; j < twoH; j += BLOCK) {
+                const uint4* g = reinterpret_cast<const uint4*>(
+                    src + (size_t)j * P_POINTS);
+                uint4* d = reinterpret_cast<uint4*>(
+                    reinterpret_cast<char*>(dst)
+                    + (size_t)j * BSTRIDE * 2);
+                d[0] = g[0]; d[1] = g[1]; d[2] = g[2]; d[3] = g[3];
+            }
         }
-        __syncthreads();
         const bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
             a_lds + (size_t)(wave * 16 + row16) * ASTRIDE + kk + kgrp * 8);
 #pragma unroll
@@ -265,12 +304,12 @@ pair_gemm_entropy64_kernel(const hip_bfloat16* __restrict__ a16,
             bf16x8 bfrag = {};
             if (j < twoH)
                 bfrag = *reinterpret_cast<const bf16x8*>(
-                    b_lds + (size_t)j * BSTRIDE + kgrp * 8);
+                    b_lds[cur] + (size_t)j * BSTRIDE + kgrp * 8);
             acc[jt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 afrag, bfrag, acc[jt], 0, 0, 0);
         }
+        __syncthreads();
     }
-    __syncthreads();
 
     // phase 2: spill accumulators to LDS (overlapping the A/B buffers)
     float* m_tile = reinterpret_cast<float*>(smem);
@@ -419,7 +458,7 @@ torch::Tensor pair_gemm_entropy(torch::Tensor a16, torch::Tensor egw,
     if (tile == 64) {
         TORCH_CHECK(2 * H <= 512, "64-pair tile needs 2H <= 512");
         const size_t phase1 = 64 * ASTRIDE * sizeof(hip_bfloat16)
-                            + (size_t)2 * H * BSTRIDE
+                            + (size_t)2 * (2 * H) * BSTRIDE
                                 * sizeof(hip_bfloat16);
         const size_t phase2 = (size_t)64 * mstride * sizeof(float);
         const size_t shmem = std::max(phase1, phase2);

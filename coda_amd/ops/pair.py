@@ -83,17 +83,17 @@ class PairStructure(NamedTuple):
 
 
 def build_pairs(cls_rows: torch.Tensor, cand_ids: torch.Tensor,
-                C: int) -> PairStructure:
+                C: int, tile: int = 0) -> PairStructure:
     """Build the hit structure from candidate argmax classes.
 
     cls_rows: (B, H) — argmax class of every (global) model on each
     candidate; cand_ids: (B,) point ids. All work is batched torch ops
     (one sort over B*H keys), so the one-off cost at N=50k, H=128 is a
-    few ms on device.
+    few ms on device. tile overrides the MFMA tile height (benchmarks).
     """
     device = cls_rows.device
     B, H = cls_rows.shape
-    tile = tile_for(H, B)
+    tile = tile or tile_for(H, B)
     cls_l = cls_rows.long()
     # sort all (b, h) entries by (class, candidate) so pairs come out
     # grouped by class, and each pair's model segment is contiguous

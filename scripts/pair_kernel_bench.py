@@ -1,0 +1,106 @@
+"""Micro-bench the v3 pair kernels at the headline shape (GPU).
+
+Times each kernel in isolation with CUDA events over interleaved
+rounds (guide rule 24), at the bench's real candidate/hit distribution.
+Usage: python scripts/pair_kernel_bench.py [--n 50000] [--h 128]
+"""
+import argparse
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, ".")
+import coda_amd.ops as O
+from coda_amd.ops import pair as pops
+from coda_amd.ops import table as tops
+from coda_amd.ops import reference as R
+import bench
+
+ap = argparse.ArgumentParser()
+ap.add_argument("--n", type=int, default=50_000)
+ap.add_argument("--h", type=int, default=128)
+ap.add_argument("--c", type=int, default=1000)
+ap.add_argument("--rounds", type=int, default=30)
+args = ap.parse_args()
+
+assert O.hip_available()
+dev = torch.device("cuda:0")
+H, N, C = args.h, args.n, args.c
+
+preds, labels = bench.synth_preds(list(range(H)), N, C, dev)
+cls = preds.argmax(-1)                                   # (H, N)
+g = torch.Generator().manual_seed(0)
+dirichlets = (torch.rand(H, C, C, generator=g) * 2 + 0.5).to(dev)
+alpha_cc, beta_cc = R.dirichlet_to_beta(dirichlets)
+tables = tops.table_precompute(alpha_cc, beta_cc)
+tables = pops.attach_pair_tables(tables)
+pbest_before = R.pbest_from_beta(alpha_cc.t().contiguous(),
+                                 beta_cc.t().contiguous())
+pi_hat = torch.rand(C, generator=g).to(dev)
+pi_hat /= pi_hat.sum()
+mixture0, H_before = R.mixture_entropy(pbest_before, pi_hat)
+adjusted = (torch.rand(N, C, generator=g) + 0.1).to(dev)
+row_sums = adjusted.sum(-1)
+
+ids = torch.arange(N, device=dev)
+cls_rows = cls[:, ids].t().to(torch.int32).contiguous()
+
+results = {}
+
+
+def timeit(name, fn, rounds=args.rounds):
+    fn()
+    torch.cuda.synchronize()
+    evs = [(torch.cuda.Event(True), torch.cuda.Event(True))
+           for _ in range(rounds)]
+    for s, e in evs:
+        s.record()
+        fn()
+        e.record()
+    torch.cuda.synchronize()
+    ts = sorted(s.elapsed_time(e) for s, e in evs)
+    med = ts[len(ts) // 2]
+    results.setdefault(name, []).append(med)
+    print(f"{name:34s} med={med*1000:9.1f}us min={ts[0]*1000:9.1f}us")
+
+
+for tile in (16, 64):
+    ps = pops.build_pairs(cls_rows, ids, C, tile=tile)
+    print(f"--- tile={tile}  K={ps.K} n_real={ps.n_real} "
+          f"avg_seg={ps.seg_h.numel()/max(ps.n_real,1):.1f}")
+    A16 = O._ext.pair_dsum_es(tables.delta16, ps.pair_c, ps.seg_off,
+                              ps.seg_h)
+    timeit(f"dsum_es[t{tile}]",
+           lambda: O._ext.pair_dsum_es(tables.delta16, ps.pair_c,
+                                       ps.seg_off, ps.seg_h))
+    h_after = O._ext.pair_gemm_entropy(
+        A16, tables.egw, ps.pair_b, ps.pair_c, cls_rows,
+        pi_hat.contiguous(), pbest_before.contiguous(),
+        mixture0.contiguous(), tile)
+    timeit(f"gemm_entropy[t{tile}]",
+           lambda: O._ext.pair_gemm_entropy(
+               A16, tables.egw, ps.pair_b, ps.pair_c, cls_rows,
+               pi_hat.contiguous(), pbest_before.contiguous(),
+               mixture0.contiguous(), tile))
+    timeit(f"finalize[t{tile}]",
+           lambda: O._ext.pair_eig_finalize(
+               h_after, ps.pair_c, ps.base_pos.to(torch.int32),
+               ps.cand_off, ps.cand_pairs, ps.cand_ids, adjusted,
+               row_sums, float(H_before)))
+
+# numerics cross-check tile16 vs tile64 h_after on the base rows
+ps16 = pops.build_pairs(cls_rows[:4096], ids[:4096], C, tile=16)
+ps64 = pops.build_pairs(cls_rows[:4096], ids[:4096], C, tile=64)
+for ps, name in ((ps16, "t16"), (ps64, "t64")):
+    A = O._ext.pair_dsum_es(tables.delta16, ps.pair_c, ps.seg_off,
+                            ps.seg_h)
+    h = O._ext.pair_gemm_entropy(A, tables.egw, ps.pair_b, ps.pair_c,
+                                 cls_rows[:4096], pi_hat.contiguous(),
+                                 pbest_before.contiguous(),
+                                 mixture0.contiguous(), ps.tile)
+    q = O._ext.pair_eig_finalize(h, ps.pair_c,
+                                 ps.base_pos.to(torch.int32),
+                                 ps.cand_off, ps.cand_pairs, ps.cand_ids,
+                                 adjusted, row_sums, float(H_before))
+    print(name, "q[:4]", q[:4].tolist())
