@@ -90,48 +90,40 @@ pair_dsum_es_kernel(const _Float16* __restrict__ delta16,  // (C, H, P)
                     const int* __restrict__ seg_h,         // (S,)
                     hip_bfloat16* __restrict__ a16,        // (K, P)
                     int K, int H) {
-    const int k = blockIdx.x * 4 + (threadIdx.x >> 6);
+    // 16 lanes per pair (the hit distribution is singleton-heavy:
+    // avg segment ~2 models - a wave per pair would mostly idle on one
+    // 512-B load). Each lane owns 16 grid points as 4x half4v loads.
+    const int k = blockIdx.x * 16 + (threadIdx.x >> 4);
     if (k >= K) return;
-    const int lane = threadIdx.x & 63;
-    const int p0 = lane * 4;
+    const int sublane = threadIdx.x & 15;
+    const int p0 = sublane * 16;
     const int c = pair_c[k];
     const int s0 = seg_off[k], s1 = seg_off[k + 1];
     const size_t dbase = (size_t)c * H * P_POINTS + p0;
 
-    // 4 accumulator sets keep 4 delta-row loads in flight (fp32 adds
-    // are not reassociated by the compiler)
-    float a0[4] = {0.f, 0.f, 0.f, 0.f}, a1[4] = {0.f, 0.f, 0.f, 0.f},
-          a2[4] = {0.f, 0.f, 0.f, 0.f}, a3[4] = {0.f, 0.f, 0.f, 0.f};
-    int s = s0;
-    for (; s + 3 < s1; s += 4) {
-        const half4v da = *reinterpret_cast<const half4v*>(
-            delta16 + dbase + (size_t)seg_h[s] * P_POINTS);
-        const half4v db = *reinterpret_cast<const half4v*>(
-            delta16 + dbase + (size_t)seg_h[s + 1] * P_POINTS);
-        const half4v dc = *reinterpret_cast<const half4v*>(
-            delta16 + dbase + (size_t)seg_h[s + 2] * P_POINTS);
-        const half4v dd = *reinterpret_cast<const half4v*>(
-            delta16 + dbase + (size_t)seg_h[s + 3] * P_POINTS);
+    float acc[16] = {};
+    for (int s = s0; s < s1; ++s) {
+        const _Float16* row = delta16 + dbase + (size_t)seg_h[s] * P_POINTS;
+        const half4v d0 = *reinterpret_cast<const half4v*>(row);
+        const half4v d1 = *reinterpret_cast<const half4v*>(row + 4);
+        const half4v d2 = *reinterpret_cast<const half4v*>(row + 8);
+        const half4v d3 = *reinterpret_cast<const half4v*>(row + 12);
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
-            a0[j] += (float)da[j]; a1[j] += (float)db[j];
-            a2[j] += (float)dc[j]; a3[j] += (float)dd[j];
+            acc[j] += (float)d0[j];
+            acc[4 + j] += (float)d1[j];
+            acc[8 + j] += (float)d2[j];
+            acc[12 + j] += (float)d3[j];
         }
     }
-    for (; s < s1; ++s) {
-        const half4v d = *reinterpret_cast<const half4v*>(
-            delta16 + dbase + (size_t)seg_h[s] * P_POINTS);
+    ushort4 out[4];
+    unsigned short* o = reinterpret_cast<unsigned short*>(out);
 #pragma unroll
-        for (int j = 0; j < 4; ++j) a0[j] += (float)d[j];
-    }
-    ushort4 out;
-    unsigned short* o = reinterpret_cast<unsigned short*>(&out);
-#pragma unroll
-    for (int j = 0; j < 4; ++j) {
-        const float v = exp2f(a0[j] + (a1[j] + a2[j]) + a3[j]);
-        o[j] = hip_bfloat16(v).data;
-    }
-    *reinterpret_cast<ushort4*>(a16 + (size_t)k * P_POINTS + p0) = out;
+    for (int j = 0; j < 16; ++j)
+        o[j] = hip_bfloat16(exp2f(acc[j])).data;
+    ushort4* dst = reinterpret_cast<ushort4*>(
+        a16 + (size_t)k * P_POINTS + p0);
+    dst[0] = out[0]; dst[1] = out[1]; dst[2] = out[2]; dst[3] = out[3];
 }
 
 // ---------------------------------------------------------------------
@@ -223,32 +215,33 @@ pair_gemm_entropy16_kernel(const hip_bfloat16* __restrict__ a16,
 }
 
 // ---------------------------------------------------------------------
-// 64-pair tile GEMM+entropy (2H <= 512): A AND B staged in LDS, each
-// egw[c] chunk read once per 64 pairs. JT = #16-col tiles is a template
-// parameter so the accumulator array stays in registers.
+// 128-pair B-resident tile GEMM+entropy (2H <= 288): the tile's whole
+// egw[c] table (2H x P bf16) is staged into LDS ONCE with a coalesced
+// row-major copy, then 8 waves run the full K loop against it with no
+// further barriers - B traffic drops to K/128 reads of 131 KB. The A
+// operand (8 KB/wave) streams from L2. JT = #16-col tiles is a
+// template parameter so the accumulator array stays in registers.
 // ---------------------------------------------------------------------
-#define ASTRIDE (P_POINTS + 8)   // bf16 elems; +8 breaks bank alignment
-#define BSTRIDE (32 + 8)
+#define BSTRIDE (P_POINTS + 8)   // bf16 elems; +8 de-aliases banks
+#define BLOCK2 512
 
 template <int JT>
-__global__ void __launch_bounds__(BLOCK)
-pair_gemm_entropy64_kernel(const hip_bfloat16* __restrict__ a16,
-                           const hip_bfloat16* __restrict__ egw,
-                           const int* __restrict__ pair_b,
-                           const int* __restrict__ pair_c,
-                           const int* __restrict__ cls,
-                           const float* __restrict__ pi_hat,
-                           const float* __restrict__ pbest_before,
-                           const float* __restrict__ mixture0,
-                           float* __restrict__ h_after,
-                           int H, int mstride) {
+__global__ void __launch_bounds__(BLOCK2)
+pair_gemm_entropy128_kernel(const hip_bfloat16* __restrict__ a16,
+                            const hip_bfloat16* __restrict__ egw,
+                            const int* __restrict__ pair_b,
+                            const int* __restrict__ pair_c,
+                            const int* __restrict__ cls,
+                            const float* __restrict__ pi_hat,
+                            const float* __restrict__ pbest_before,
+                            const float* __restrict__ mixture0,
+                            float* __restrict__ h_after,
+                            int H, int mstride) {
     extern __shared__ char smem[];
     const int twoH = 2 * H;
-    hip_bfloat16* a_lds = reinterpret_cast<hip_bfloat16*>(smem);
-    hip_bfloat16* b_lds[2] = {a_lds + 64 * ASTRIDE,
-                              a_lds + 64 * ASTRIDE + twoH * BSTRIDE};
+    hip_bfloat16* b_lds = reinterpret_cast<hip_bfloat16*>(smem);
 
-    const int k0 = blockIdx.x * 64;
+    const int k0 = blockIdx.x * 128;
     const int c = pair_c[k0];
     const int tid = threadIdx.x;
     const int wave = tid >> 6, lane = tid & 63;
@@ -256,23 +249,17 @@ pair_gemm_entropy64_kernel(const hip_bfloat16* __restrict__ a16,
     const int kgrp = lane >> 4;
     const hip_bfloat16* egw_c = egw + (size_t)c * twoH * P_POINTS;
 
-    {   // stage A: 64 rows x 256 bf16 (padded rows); 128 B per thread
-        const int row = tid >> 2, quarter = tid & 3;
+    // stage the whole B table: row j (512 B) split in 2 halves; thread
+    // t copies half (t&1) of row (t>>1) - fully coalesced 16-B loads
+    for (int t = tid; t < 2 * twoH; t += BLOCK2) {
+        const int j = t >> 1, half = t & 1;
         const uint4* g = reinterpret_cast<const uint4*>(
-            a16 + (size_t)(k0 + row) * P_POINTS + quarter * 64);
+            egw_c + (size_t)j * P_POINTS + half * 128);
         uint4* d = reinterpret_cast<uint4*>(
-            reinterpret_cast<char*>(a_lds)
-            + (size_t)row * ASTRIDE * 2 + quarter * 128);
+            reinterpret_cast<char*>(b_lds)
+            + (size_t)j * BSTRIDE * 2 + half * 256);
 #pragma unroll
-        for (int i = 0; i < 8; ++i) d[i] = g[i];
-    }
-    // prologue: chunk 0 into buffer 0
-    for (int j = tid; j < twoH; j += BLOCK) {
-        const uint4* g = reinterpret_cast<const uint4*>(
-            egw_c + (size_t)j * P_POINTS);
-        uint4* d = reinterpret_cast<uint4*>(
-            reinterpret_cast<char*>(b_lds[0]) + (size_t)j * BSTRIDE * 2);
-        d[0] = g[0]; d[1] = g[1]; d[2] = g[2]; d[3] = g[3];
+        for (int i = 0; i < 16; ++i) d[i] = g[i];
     }
     __syncthreads();
 
@@ -280,38 +267,26 @@ pair_gemm_entropy64_kernel(const hip_bfloat16* __restrict__ a16,
 #pragma unroll
     for (int jt = 0; jt < JT; ++jt) acc[jt] = {0.f, 0.f, 0.f, 0.f};
 
-    // double-buffered K loop: one barrier per chunk, next-chunk loads
-    // in flight under the MFMAs
+    const hip_bfloat16* arow =
+        a16 + (size_t)(k0 + wave * 16 + row16) * P_POINTS;
+#pragma unroll
     for (int kk = 0; kk < P_POINTS; kk += 32) {
-        const int cur = (kk >> 5) & 1;
-        if (kk + 32 < P_POINTS) {
-            const hip_bfloat16* src = egw_c + kk + 32;
-            hip_bfloat16* dst = b_lds[cur ^ 1];
-            for (int j = tid; j < twoH; j += BLOCK) {
-                const uint4* g = reinterpret_cast<const uint4*>(
-                    src + (size_t)j * P_POINTS);
-                uint4* d = reinterpret_cast<uint4*>(
-                    reinterpret_cast<char*>(dst)
-                    + (size_t)j * BSTRIDE * 2);
-                d[0] = g[0]; d[1] = g[1]; d[2] = g[2]; d[3] = g[3];
-            }
-        }
         const bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
-            a_lds + (size_t)(wave * 16 + row16) * ASTRIDE + kk + kgrp * 8);
+            arow + kk + kgrp * 8);
 #pragma unroll
         for (int jt = 0; jt < JT; ++jt) {
             const int j = jt * 16 + row16;
             bf16x8 bfrag = {};
             if (j < twoH)
                 bfrag = *reinterpret_cast<const bf16x8*>(
-                    b_lds[cur] + (size_t)j * BSTRIDE + kgrp * 8);
+                    b_lds + (size_t)j * BSTRIDE + kk + kgrp * 8);
             acc[jt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 afrag, bfrag, acc[jt], 0, 0, 0);
         }
-        __syncthreads();
     }
+    __syncthreads();
 
-    // phase 2: spill accumulators to LDS (overlapping the A/B buffers)
+    // phase 2: spill accumulators to LDS (overlapping the B buffer)
     float* m_tile = reinterpret_cast<float*>(smem);
 #pragma unroll
     for (int jt = 0; jt < JT; ++jt) {
@@ -325,7 +300,7 @@ pair_gemm_entropy64_kernel(const hip_bfloat16* __restrict__ a16,
     }
     __syncthreads();
 
-    for (int pi = wave; pi < 64; pi += 4) {
+    for (int pi = wave; pi < 128; pi += 8) {
         const int k = k0 + pi;
         const int b = pair_b[k];
         const float* mrow = m_tile + (size_t)pi * mstride;
@@ -455,16 +430,15 @@ torch::Tensor pair_gemm_entropy(torch::Tensor a16, torch::Tensor egw,
         a16.data_ptr());
     const auto eb16 = reinterpret_cast<const hip_bfloat16*>(
         egw.data_ptr());
-    if (tile == 64) {
-        TORCH_CHECK(2 * H <= 512, "64-pair tile needs 2H <= 512");
-        const size_t phase1 = 64 * ASTRIDE * sizeof(hip_bfloat16)
-                            + (size_t)2 * (2 * H) * BSTRIDE
-                                * sizeof(hip_bfloat16);
-        const size_t phase2 = (size_t)64 * mstride * sizeof(float);
+    if (tile == 128) {
+        TORCH_CHECK(2 * H <= 288, "128-pair tile needs 2H <= 288");
+        const size_t phase1 = (size_t)2 * H * BSTRIDE
+                            * sizeof(hip_bfloat16);
+        const size_t phase2 = (size_t)128 * mstride * sizeof(float);
         const size_t shmem = std::max(phase1, phase2);
         const int JT = (2 * H + 15) / 16;
         auto launch = [&](auto kern) {
-            hipLaunchKernelGGL(kern, dim3(K / 64), dim3(BLOCK), shmem,
+            hipLaunchKernelGGL(kern, dim3(K / 128), dim3(BLOCK2), shmem,
                                stream.stream(), ab16, eb16,
                                pair_b.data_ptr<int>(),
                                pair_c.data_ptr<int>(),
@@ -474,11 +448,12 @@ torch::Tensor pair_gemm_entropy(torch::Tensor a16, torch::Tensor egw,
                                mixture0.data_ptr<float>(),
                                h_after.data_ptr<float>(), H, mstride);
         };
-        if (JT <= 4) launch(pairops::pair_gemm_entropy64_kernel<4>);
-        else if (JT <= 8) launch(pairops::pair_gemm_entropy64_kernel<8>);
+        if (JT <= 4) launch(pairops::pair_gemm_entropy128_kernel<4>);
+        else if (JT <= 8)
+            launch(pairops::pair_gemm_entropy128_kernel<8>);
         else if (JT <= 16)
-            launch(pairops::pair_gemm_entropy64_kernel<16>);
-        else launch(pairops::pair_gemm_entropy64_kernel<32>);
+            launch(pairops::pair_gemm_entropy128_kernel<16>);
+        else launch(pairops::pair_gemm_entropy128_kernel<18>);
     } else {
         TORCH_CHECK(tile == 16, "tile must be 16 or 64");
         const size_t shmem = 16 * P_POINTS * sizeof(hip_bfloat16)

@@ -50,12 +50,12 @@ PAIR_TILE = 16  # minimum tile; build_pairs may pad to 64 (see below)
 def tile_for(H: int, B: int) -> int:
     """MFMA tile height (pairs per class-uniform tile).
 
-    64-pair tiles quarter the B-operand traffic (the dominant cost of
-    the pairing GEMM) but need 2H <= 512 to fit the LDS epilogue and
-    only pay off when the padding (<= tile-1 pairs per class) is small
-    against the real pair count.
+    128-pair tiles keep the whole egw[c] B operand resident in LDS
+    (divides the dominant GEMM traffic by 8 vs 16-pair tiles) but need
+    2H <= 288 to fit, and only pay off when the padding (<= tile-1
+    pairs per class) is small against the real pair count.
     """
-    return 64 if (H <= 256 and B >= 4096) else PAIR_TILE
+    return 128 if (H <= 144 and B >= 4096) else PAIR_TILE
 
 
 class PairStructure(NamedTuple):

@@ -65,7 +65,7 @@ def timeit(name, fn, rounds=args.rounds):
     print(f"{name:34s} med={med*1000:9.1f}us min={ts[0]*1000:9.1f}us")
 
 
-for tile in (16, 64):
+for tile in (16, 128):
     ps = pops.build_pairs(cls_rows, ids, C, tile=tile)
     print(f"--- tile={tile}  K={ps.K} n_real={ps.n_real} "
           f"avg_seg={ps.seg_h.numel()/max(ps.n_real,1):.1f}")
@@ -91,7 +91,7 @@ for tile in (16, 64):
 
 # numerics cross-check tile16 vs tile64 h_after on the base rows
 ps16 = pops.build_pairs(cls_rows[:4096], ids[:4096], C, tile=16)
-ps64 = pops.build_pairs(cls_rows[:4096], ids[:4096], C, tile=64)
+ps64 = pops.build_pairs(cls_rows[:4096], ids[:4096], C, tile=128)
 for ps, name in ((ps16, "t16"), (ps64, "t64")):
     A = O._ext.pair_dsum_es(tables.delta16, ps.pair_c, ps.seg_off,
                             ps.seg_h)
