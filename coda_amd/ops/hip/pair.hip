@@ -132,9 +132,9 @@ pair_dsum_es_kernel(const _Float16* __restrict__ delta16,  // (C, H, P)
 __global__ void __launch_bounds__(BLOCK)
 pair_gemm_entropy16_kernel(const hip_bfloat16* __restrict__ a16,
                            const hip_bfloat16* __restrict__ egw,
-                           const int* __restrict__ pair_b,
+                           const unsigned* __restrict__ vmask,
                            const int* __restrict__ pair_c,
-                           const int* __restrict__ cls,
+                           int W,
                            const float* __restrict__ pi_hat,
                            const float* __restrict__ pbest_before,
                            const float* __restrict__ mixture0,
@@ -190,11 +190,11 @@ pair_gemm_entropy16_kernel(const hip_bfloat16* __restrict__ a16,
 
     for (int pi = wave; pi < 16; pi += 4) {
         const int k = k0 + pi;
-        const int b = pair_b[k];
+        const unsigned* vm = vmask + (size_t)k * W;
         const float* mrow = m_tile + (size_t)pi * mstride;
         float tot = 0.f;
         for (int h = lane; h < H; h += 64) {
-            const int v = (b >= 0 && cls[(size_t)b * H + h] == c) ? 1 : 0;
+            const int v = (vm[h >> 5] >> (h & 31)) & 1;
             tot += mrow[2 * h + v];
         }
         tot = wave_reduce(tot);
@@ -202,7 +202,7 @@ pair_gemm_entropy16_kernel(const hip_bfloat16* __restrict__ a16,
         const float pic = pi_hat[c];
         float ent = 0.f;
         for (int h = lane; h < H; h += 64) {
-            const int v = (b >= 0 && cls[(size_t)b * H + h] == c) ? 1 : 0;
+            const int v = (vm[h >> 5] >> (h & 31)) & 1;
             const float pb = mrow[2 * h + v] * inv;
             const float mm = fmaxf(
                 mixture0[h] + pic * (pb - pbest_before[(size_t)c * H + h]),
@@ -225,13 +225,13 @@ pair_gemm_entropy16_kernel(const hip_bfloat16* __restrict__ a16,
 #define BSTRIDE (P_POINTS + 8)   // bf16 elems; +8 de-aliases banks
 #define BLOCK2 512
 
-template <int JT>
+template <int JT, int ABL = 0>   // ABL: 1=skip epilogue, 2=skip MFMA, 3=no cls reads
 __global__ void __launch_bounds__(BLOCK2)
 pair_gemm_entropy128_kernel(const hip_bfloat16* __restrict__ a16,
                             const hip_bfloat16* __restrict__ egw,
-                            const int* __restrict__ pair_b,
+                            const unsigned* __restrict__ vmask,
                             const int* __restrict__ pair_c,
-                            const int* __restrict__ cls,
+                            int W,
                             const float* __restrict__ pi_hat,
                             const float* __restrict__ pbest_before,
                             const float* __restrict__ mixture0,
@@ -270,7 +270,7 @@ pair_gemm_entropy128_kernel(const hip_bfloat16* __restrict__ a16,
     const hip_bfloat16* arow =
         a16 + (size_t)(k0 + wave * 16 + row16) * P_POINTS;
 #pragma unroll
-    for (int kk = 0; kk < P_POINTS; kk += 32) {
+    for (int kk = 0; kk < (ABL == 2 ? 0 : P_POINTS); kk += 32) {
         const bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
             arow + kk + kgrp * 8);
 #pragma unroll
@@ -300,22 +300,29 @@ pair_gemm_entropy128_kernel(const hip_bfloat16* __restrict__ a16,
     }
     __syncthreads();
 
+    if (ABL == 1) {
+        if (tid < 128) h_after[k0 + tid] = m_tile[tid];
+        return;
+    }
     for (int pi = wave; pi < 128; pi += 8) {
         const int k = k0 + pi;
-        const int b = pair_b[k];
+        const unsigned* vm = vmask + (size_t)k * W;
         const float* mrow = m_tile + (size_t)pi * mstride;
+        float val[3];          // pass-1 cache (H <= 144 -> <= 3 / lane)
         float tot = 0.f;
-        for (int h = lane; h < H; h += 64) {
-            const int v = (b >= 0 && cls[(size_t)b * H + h] == c) ? 1 : 0;
-            tot += mrow[2 * h + v];
+        int i = 0;
+        for (int h = lane; h < H; h += 64, ++i) {
+            const int v = ABL == 3 ? 0 : (vm[h >> 5] >> (h & 31)) & 1;
+            val[i] = mrow[2 * h + v];
+            tot += val[i];
         }
         tot = wave_reduce(tot);
         const float inv = 1.0f / fmaxf(tot, 1e-30f);
         const float pic = pi_hat[c];
         float ent = 0.f;
-        for (int h = lane; h < H; h += 64) {
-            const int v = (b >= 0 && cls[(size_t)b * H + h] == c) ? 1 : 0;
-            const float pb = mrow[2 * h + v] * inv;
+        i = 0;
+        for (int h = lane; h < H; h += 64, ++i) {
+            const float pb = val[i] * inv;
             const float mm = fmaxf(
                 mixture0[h] + pic * (pb - pbest_before[(size_t)c * H + h]),
                 1e-12f);
@@ -413,10 +420,11 @@ torch::Tensor pair_dsum_es(torch::Tensor delta16, torch::Tensor pair_c,
 }
 
 torch::Tensor pair_gemm_entropy(torch::Tensor a16, torch::Tensor egw,
-                                torch::Tensor pair_b, torch::Tensor pair_c,
-                                torch::Tensor cls, torch::Tensor pi_hat,
+                                torch::Tensor vmask, torch::Tensor pair_c,
+                                torch::Tensor pi_hat,
                                 torch::Tensor pbest_before,
-                                torch::Tensor mixture0, int64_t tile) {
+                                torch::Tensor mixture0, int64_t tile,
+                                int64_t ablate) {
     TORCH_CHECK(a16.is_cuda() && a16.dtype() == torch::kBFloat16);
     TORCH_CHECK(egw.dtype() == torch::kBFloat16);
     const int K = a16.size(0);
@@ -440,9 +448,10 @@ torch::Tensor pair_gemm_entropy(torch::Tensor a16, torch::Tensor egw,
         auto launch = [&](auto kern) {
             hipLaunchKernelGGL(kern, dim3(K / 128), dim3(BLOCK2), shmem,
                                stream.stream(), ab16, eb16,
-                               pair_b.data_ptr<int>(),
+                               reinterpret_cast<const unsigned*>(
+                                   vmask.data_ptr<int>()),
                                pair_c.data_ptr<int>(),
-                               cls.data_ptr<int>(),
+                               (int)vmask.size(1),
                                pi_hat.data_ptr<float>(),
                                pbest_before.data_ptr<float>(),
                                mixture0.data_ptr<float>(),
@@ -451,9 +460,15 @@ torch::Tensor pair_gemm_entropy(torch::Tensor a16, torch::Tensor egw,
         if (JT <= 4) launch(pairops::pair_gemm_entropy128_kernel<4>);
         else if (JT <= 8)
             launch(pairops::pair_gemm_entropy128_kernel<8>);
-        else if (JT <= 16)
-            launch(pairops::pair_gemm_entropy128_kernel<16>);
-        else launch(pairops::pair_gemm_entropy128_kernel<18>);
+        else if (JT <= 16) {
+            if (ablate == 1)
+                launch(pairops::pair_gemm_entropy128_kernel<16, 1>);
+            else if (ablate == 2)
+                launch(pairops::pair_gemm_entropy128_kernel<16, 2>);
+            else if (ablate == 3)
+                launch(pairops::pair_gemm_entropy128_kernel<16, 3>);
+            else launch(pairops::pair_gemm_entropy128_kernel<16>);
+        } else launch(pairops::pair_gemm_entropy128_kernel<18>);
     } else {
         TORCH_CHECK(tile == 16, "tile must be 16 or 64");
         const size_t shmem = 16 * P_POINTS * sizeof(hip_bfloat16)
@@ -463,8 +478,11 @@ torch::Tensor pair_gemm_entropy(torch::Tensor a16, torch::Tensor egw,
         hipLaunchKernelGGL(pairops::pair_gemm_entropy16_kernel,
                            dim3(K / 16), dim3(BLOCK), shmem,
                            stream.stream(), ab16, eb16,
-                           pair_b.data_ptr<int>(), pair_c.data_ptr<int>(),
-                           cls.data_ptr<int>(), pi_hat.data_ptr<float>(),
+                           reinterpret_cast<const unsigned*>(
+                               vmask.data_ptr<int>()),
+                           pair_c.data_ptr<int>(),
+                           (int)vmask.size(1),
+                           pi_hat.data_ptr<float>(),
                            pbest_before.data_ptr<float>(),
                            mixture0.data_ptr<float>(),
                            h_after.data_ptr<float>(), H, mstride);
@@ -512,7 +530,12 @@ void register_pair_ops(pybind11::module_& m) {
     m.def("pair_dsum_es", &pair_dsum_es,
           "v3 pair A-operand: 2^(summed delta curves) -> (K, P) bf16");
     m.def("pair_gemm_entropy", &pair_gemm_entropy,
-          "v3 fused pairing MFMA GEMM + entropy epilogue -> (K,)");
+          "v3 fused pairing MFMA GEMM + entropy epilogue -> (K,)",
+          pybind11::arg("a16"), pybind11::arg("egw"),
+          pybind11::arg("vmask"), pybind11::arg("pair_c"),
+          pybind11::arg("pi_hat"),
+          pybind11::arg("pbest_before"), pybind11::arg("mixture0"),
+          pybind11::arg("tile"), pybind11::arg("ablate") = 0);
     m.def("pair_eig_finalize", &pair_eig_finalize,
           "v3 per-candidate EIG assembly (deterministic) -> (B,)");
     m.def("mfma_probe", &mfma_probe,

@@ -73,9 +73,13 @@ class PairStructure(NamedTuple):
     seg_h: torch.Tensor      # (S,) int32 — models with argmax c on point b
     base_pos: torch.Tensor   # (C,) long — position of class c's base pair
     n_real: int              # real (non-base, non-pad) pair count
-    tile: int                # pairs per kernel tile (16 or 64)
+    tile: int                # pairs per kernel tile (16 or 128)
     cand_off: torch.Tensor = None    # (B+1,) int32 — per-candidate CSR
     cand_pairs: torch.Tensor = None  # (n_real,) int32 — pair ids by cand
+    # (K, ceil(H/32)) int32 — bit h set iff model h hits pair k (the
+    # static v-select; reading cls rows per pair costs 1.6 ms/step at
+    # the headline shape, the bitmask 0.05 ms)
+    vmask: torch.Tensor = None
 
     @property
     def K(self) -> int:
@@ -141,11 +145,22 @@ def build_pairs(cls_rows: torch.Tensor, cand_ids: torch.Tensor,
     cand_off = torch.zeros(B + 1, dtype=torch.int32, device=device)
     cand_off[1:] = b_counts.cumsum(0).to(torch.int32)
 
+    # static v-select bitmask: bit h of vmask[k] iff model h in seg(k)
+    # (disjoint bits, so scatter-add == bitwise-or; base/pad rows stay 0)
+    W = (H + 31) // 32
+    seg_pair_pos = torch.repeat_interleave(pos, seg_len_real)
+    word = seg_pair_pos * W + (seg_h.long() >> 5)
+    bit = torch.bitwise_left_shift(
+        torch.ones_like(seg_h, dtype=torch.int32), seg_h & 31)
+    vmask = torch.zeros(K * W, dtype=torch.int32, device=device)
+    vmask.index_put_((word,), bit, accumulate=True)
+    vmask = vmask.view(K, W)
+
     return PairStructure(cand_ids=cand_ids.long(), pair_b=pair_b,
                          pair_c=pair_c, seg_off=seg_off, seg_h=seg_h,
                          base_pos=run_off[:-1].clone(), n_real=K_real,
                          tile=tile, cand_off=cand_off,
-                         cand_pairs=cand_pairs)
+                         cand_pairs=cand_pairs, vmask=vmask)
 
 
 def pair_h_after(tables, ps: PairStructure, cls_rows: torch.Tensor,
@@ -231,8 +246,7 @@ def eig_pairs(tables, ps: PairStructure, cls_rows: torch.Tensor,
         A16 = O._ext.pair_dsum_es(tables.delta16, ps.pair_c, ps.seg_off,
                                   ps.seg_h)                 # (K, P) bf16
         h_after = O._ext.pair_gemm_entropy(
-            A16, tables.egw, ps.pair_b, ps.pair_c,
-            cls_rows.to(torch.int32).contiguous(),
+            A16, tables.egw, ps.vmask, ps.pair_c,
             pi_hat.contiguous(), pbest_before.contiguous(),
             mixture0.contiguous(), ps.tile)                 # (K,)
         q = O._ext.pair_eig_finalize(

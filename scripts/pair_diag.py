@@ -36,7 +36,7 @@ for H, N, C in [(10, 120, 7), (3, 60, 2), (16, 200, 126), (128, 500, 50),
     h_k = O._ext.pair_gemm_entropy(
         O._ext.pair_dsum_es(tables.delta16, ps.pair_c, ps.seg_off,
                             ps.seg_h),
-        tables.egw, ps.pair_b, ps.pair_c, cls_rows,
+        tables.egw, ps.vmask, ps.pair_c,
         pi_hat.to(dev).contiguous(), pbest_before.contiguous(),
         mixture0.contiguous(), ps.tile)
     eig_e = pops.eig_from_pairs(h_e, ps, adjusted.to(dev),

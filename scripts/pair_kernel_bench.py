@@ -75,12 +75,12 @@ for tile in (16, 128):
            lambda: O._ext.pair_dsum_es(tables.delta16, ps.pair_c,
                                        ps.seg_off, ps.seg_h))
     h_after = O._ext.pair_gemm_entropy(
-        A16, tables.egw, ps.pair_b, ps.pair_c, cls_rows,
+        A16, tables.egw, ps.vmask, ps.pair_c,
         pi_hat.contiguous(), pbest_before.contiguous(),
         mixture0.contiguous(), tile)
     timeit(f"gemm_entropy[t{tile}]",
            lambda: O._ext.pair_gemm_entropy(
-               A16, tables.egw, ps.pair_b, ps.pair_c, cls_rows,
+               A16, tables.egw, ps.vmask, ps.pair_c,
                pi_hat.contiguous(), pbest_before.contiguous(),
                mixture0.contiguous(), tile))
     timeit(f"finalize[t{tile}]",
@@ -95,8 +95,8 @@ ps64 = pops.build_pairs(cls_rows[:4096], ids[:4096], C, tile=128)
 for ps, name in ((ps16, "t16"), (ps64, "t64")):
     A = O._ext.pair_dsum_es(tables.delta16, ps.pair_c, ps.seg_off,
                             ps.seg_h)
-    h = O._ext.pair_gemm_entropy(A, tables.egw, ps.pair_b, ps.pair_c,
-                                 cls_rows[:4096], pi_hat.contiguous(),
+    h = O._ext.pair_gemm_entropy(A, tables.egw, ps.vmask, ps.pair_c,
+                                 pi_hat.contiguous(),
                                  pbest_before.contiguous(),
                                  mixture0.contiguous(), ps.tile)
     q = O._ext.pair_eig_finalize(h, ps.pair_c,
@@ -104,3 +104,14 @@ for ps, name in ((ps16, "t16"), (ps64, "t64")):
                                  ps.cand_off, ps.cand_pairs, ps.cand_ids,
                                  adjusted, row_sums, float(H_before))
     print(name, "q[:4]", q[:4].tolist())
+
+# phase ablation at tile=128 (guide: symptom -> diagnosis loop)
+ps = pops.build_pairs(cls_rows, ids, C, tile=128)
+A16 = O._ext.pair_dsum_es(tables.delta16, ps.pair_c, ps.seg_off, ps.seg_h)
+for abl, name in [(0, "full"), (1, "no-epilogue"), (2, "no-mfma"),
+                  (3, "no-cls-reads")]:
+    timeit(f"gemm128[{name}]",
+           lambda: O._ext.pair_gemm_entropy(
+               A16, tables.egw, ps.vmask, ps.pair_c,
+               pi_hat.contiguous(), pbest_before.contiguous(),
+               mixture0.contiguous(), 128, abl))
