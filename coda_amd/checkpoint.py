@@ -75,10 +75,23 @@ def load_state_dict(selector, state: Dict[str, Any]):
         selector._posterior_version += 1
         selector._pbest_rows_cache = (-1, None)
         selector._label_graph = None  # graph buffers alias replaced state
+        selector._pairs_static = None  # hit structure covers a stale set
+        selector._pair_row_of = None
+        selector._active_mask = None
         from sortedcontainers import SortedList
         selector._active_candidates = SortedList(
             i for i in selector.unlabeled_idxs
             if selector._disagreement_host[i])
+        if getattr(selector, "_replicated", False):
+            from coda_amd import ops
+            a_l, b_l = ops.dirichlet_to_beta(selector.dirichlets)
+            sizes = selector.comm.shard_sizes(selector.H)
+            order = selector.comm.unshard_order(selector.H) \
+                .to(selector.device)
+            selector._alpha_g = selector.comm.all_gather_cat(
+                a_l, dim=0, sizes=sizes)[order].contiguous()
+            selector._beta_g = selector.comm.all_gather_cat(
+                b_l, dim=0, sizes=sizes)[order].contiguous()
         selector.update_pi_hat()
     return selector
 

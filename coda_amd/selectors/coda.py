@@ -93,6 +93,30 @@ class CODA(ModelSelector):
         # consensus sum (storage may be fp32/bf16/fp8)
         self.classes, ens_sum = ops.init_model_stats(preds)
 
+        # Sharded pair engine = "replicated Betas, sharded candidates":
+        # every rank keeps the global (H, C) diagonal-Beta view + the
+        # global argmax classes (gathered ONCE - they never change), so
+        # tables/pbest/mixture are computed locally and bit-identically
+        # on all ranks, EIG work is split by candidate, and the only
+        # per-step wire is the (N,) pi_hat delta + a (B,) EIG gather.
+        # This replaces the round-1 design that gathered (B, H, P) delta
+        # curves per chunk (268 MB/chunk at the 8-GPU config;
+        # VERDICT.md round-1 finding #1).
+        self._replicated = (
+            self.comm.is_distributed and self.H <= 1024
+            and self.C <= 32767 and eig_impl in ("auto", "pair"))
+        self.classes_global = None
+        self._alpha_g = self._beta_g = None
+        if self._replicated:
+            # gather in int32 (gloo has no int16 collectives), store
+            # int16 - at the 8-GPU 1024x1M config that's 2 GB resident
+            cls32 = self.classes.to(torch.int32)
+            gathered = self.comm.all_gather_cat(
+                cls32, dim=0, sizes=self.comm.shard_sizes(self.H))
+            order = self.comm.unshard_order(self.H).to(gathered.device)
+            self.classes_global = gathered[order].to(torch.int16) \
+                .contiguous()                               # (H, N)
+
         # pi_hat compute dtype: bf16 MFMA on GPU (one packed (N, H*C) GEMM,
         # ~16x the f32 matrix rate, f32 accumulation), fp32 elsewhere.
         if pi_hat_precision == "auto":
@@ -108,17 +132,33 @@ class CODA(ModelSelector):
         soft_conf = ops.confusion_prior(pseudo, preds)    # (Hl, C, C)
         self.dirichlets = ops.init_dirichlets(
             soft_conf, self.prior_strength, disable_diag_prior, multiplier)
+        if self._replicated:
+            # replicated global diagonal-Beta view (gathered once; kept
+            # current locally from classes_global on every label)
+            a_l, b_l = ops.dirichlet_to_beta(self.dirichlets)  # (Hl, C)
+            sizes = self.comm.shard_sizes(self.H)
+            order = self.comm.unshard_order(self.H).to(self.device)
+            self._alpha_g = self.comm.all_gather_cat(
+                a_l, dim=0, sizes=sizes)[order].contiguous()
+            self._beta_g = self.comm.all_gather_cat(
+                b_l, dim=0, sizes=sizes)[order].contiguous()
         self.update_pi_hat()
 
         # static disagreement mask (K12): "not all models agree".
-        # Cross-shard: compare against global model 0's classes (it lives on
-        # rank 0 under strided sharding), then OR-reduce.
-        ref_row = self.classes[0].clone() if self.comm.rank == 0 \
-            else torch.empty_like(self.classes[0])
-        self.comm.broadcast_(ref_row, src=0)
-        dis = (self.classes != ref_row.unsqueeze(0)).any(dim=0).to(torch.float32)
-        self.comm.all_reduce_(dis)
-        self._disagreement = dis > 0                      # (N,) bool
+        if self._replicated:
+            ref = self.classes_global[0]
+            self._disagreement = (self.classes_global
+                                  != ref.unsqueeze(0)).any(dim=0)
+        else:
+            # Cross-shard: compare against global model 0's classes (it
+            # lives on rank 0 under strided sharding), then OR-reduce.
+            ref_row = self.classes[0].clone() if self.comm.rank == 0 \
+                else torch.empty_like(self.classes[0])
+            self.comm.broadcast_(ref_row, src=0)
+            dis = (self.classes != ref_row.unsqueeze(0)).any(dim=0) \
+                .to(torch.float32)
+            self.comm.all_reduce_(dis)
+            self._disagreement = dis > 0                  # (N,) bool
         # host-side copy for the per-step candidate filter (indexing a GPU
         # tensor point-by-point would be one device sync per point)
         self._disagreement_host = self._disagreement.cpu().tolist()
@@ -224,15 +264,22 @@ class CODA(ModelSelector):
         ver, cached = self._pbest_rows_cache
         if ver == self._posterior_version and cached is not None:
             return cached
-        alpha_cc, beta_cc = ops.dirichlet_to_beta(self.dirichlets)  # (Hl, C)
-        a, b = alpha_cc.t().contiguous(), beta_cc.t().contiguous()  # (C, Hl)
-        if self.comm.is_distributed:
+        alpha_cc, beta_cc = self._beta_view()
+        a, b = alpha_cc.t().contiguous(), beta_cc.t().contiguous()
+        if self.comm.is_distributed and not self._replicated:
             rows = shops.pbest_from_beta_sharded(a, b, self.comm,
                                                  self.num_points)
         else:
+            # replicated mode computes the GLOBAL (C, H) rows locally
             rows = ops.pbest_from_beta(a, b, self.num_points)
         self._pbest_rows_cache = (self._posterior_version, rows)
         return rows
+
+    def _beta_view(self):
+        """(H?, C) diagonal Betas: global when replicated, else local."""
+        if self._replicated:
+            return self._alpha_g, self._beta_g
+        return ops.dirichlet_to_beta(self.dirichlets)
 
     def _refresh_tables(self, alpha_cc, beta_cc, want_egw: bool):
         """Build or incrementally refresh the per-step curve tables."""
@@ -250,41 +297,76 @@ class CODA(ModelSelector):
         return self._tables
 
     def _eig_pair(self, candidate_ids):
-        """v3 hit-sparse EIG over the candidate set (ops/pair.py)."""
+        """v3 hit-sparse EIG over the candidate set (ops/pair.py).
+
+        Distributed: candidates are split round-robin by their position
+        in the (shared, ascending) candidate list; each rank scores its
+        slice against the replicated global tables, and the (B,) EIG
+        values are all-gathered and re-interleaved - O(B) bytes per
+        step on the wire.
+        """
         from ..ops import pair as pops
-        alpha_cc, beta_cc = ops.dirichlet_to_beta(self.dirichlets)
+        alpha_cc, beta_cc = self._beta_view()
         tables = self._refresh_tables(alpha_cc, beta_cc, want_egw=True)
         pbest_before = self._pbest_rows_before()            # (C, H)
         mixture0, H_before = ops.mixture_entropy(pbest_before, self.pi_hat)
+
+        r, w = self.comm.rank, self.comm.world
+
+        def build(ids_t):
+            mine = ids_t[r::w] if self._replicated else ids_t
+            cls_rows = self._global_classes(mine)
+            return pops.build_pairs(cls_rows, mine, self.C), cls_rows
 
         if candidate_ids is self._active_candidates:
             # full-pool acquisition: the hit structure is static (argmax
             # classes never change); labeled points are masked at gather
             if self._pairs_static is None:
                 ids = torch.tensor(list(candidate_ids), device=self.device)
-                cls_rows = self._global_classes(ids)
-                self._pairs_static = (
-                    pops.build_pairs(cls_rows, ids, self.C), cls_rows)
-                self._pair_row_of = {int(p): r
-                                     for r, p in enumerate(ids.tolist())}
+                self._pairs_static = build(ids)
+                self._pair_row_of = {int(p): i
+                                     for i, p in enumerate(ids.tolist())}
                 self._active_mask = torch.ones(
                     ids.numel(), dtype=torch.bool, device=self.device)
+                self._shard_perm = self._interleave_perm(ids.numel())
             ps, cls_rows = self._pairs_static
-            mask = self._active_mask
+            mask, perm = self._active_mask, self._shard_perm
         else:
             cand = torch.tensor(list(candidate_ids), device=self.device)
-            cls_rows = self._global_classes(cand)
-            ps = pops.build_pairs(cls_rows, cand, self.C)
-            mask = None
-        q = pops.eig_pairs(tables, ps, cls_rows, pbest_before,
-                           self.pi_hat, mixture0, H_before,
-                           self._adjusted, self._row_sums)  # (B,)
+            ps, cls_rows = build(cand)
+            mask, perm = None, self._interleave_perm(cand.numel())
+        if ps.cand_ids.numel():
+            q = pops.eig_pairs(tables, ps, cls_rows, pbest_before,
+                               self.pi_hat, mixture0, H_before,
+                               self._adjusted, self._row_sums)
+        else:  # more ranks than candidates
+            q = torch.empty(0, device=self.device)
+        if self._replicated:
+            sizes = [len(range(rr, (mask.numel() if mask is not None
+                                    else perm.numel()), w))
+                     for rr in range(w)]
+            q = self.comm.all_gather_cat(q, dim=0, sizes=sizes)[perm]
         return (q[mask] if mask is not None else q), candidate_ids
+
+    def _interleave_perm(self, n: int) -> torch.Tensor:
+        """Permutation mapping rank-concatenated round-robin slices back
+        to list order (identity when not distributed)."""
+        if not self._replicated:
+            return torch.arange(n, device=self.device)
+        w = self.comm.world
+        perm = torch.empty(n, dtype=torch.long, device=self.device)
+        off = 0
+        for rr in range(w):
+            cnt = len(range(rr, n, w))
+            perm[rr:n:w] = torch.arange(off, off + cnt, device=self.device)
+            off += cnt
+        return perm
 
     def _global_classes(self, ids: torch.Tensor) -> torch.Tensor:
         """(B, H) int32 argmax classes of every GLOBAL model on the
-        given points (single-device: classes is already global)."""
-        return self.classes[:, ids].t().to(torch.int32).contiguous()
+        given points."""
+        src = self.classes_global if self._replicated else self.classes
+        return src[:, ids].t().to(torch.int32).contiguous()
 
     def eig_batched(self):
         """EIG for every candidate point (reference: coda/coda.py:235-281)."""
@@ -292,10 +374,14 @@ class CODA(ModelSelector):
             or self.unlabeled_idxs
 
         impl0 = self.eig_impl
-        if impl0 == "auto" and self.device.type == "cuda" \
-                and not self.comm.is_distributed and self.H <= 1024:
-            impl0 = "pair"
-        if impl0 == "pair" and not self.comm.is_distributed:
+        if impl0 == "auto":
+            if self._replicated:
+                impl0 = "pair"
+            elif (self.device.type == "cuda"
+                    and not self.comm.is_distributed and self.H <= 1024):
+                impl0 = "pair"
+        if impl0 == "pair" and (self._replicated
+                                or not self.comm.is_distributed):
             return self._eig_pair(candidate_ids)
 
         cand = torch.tensor(list(candidate_ids), device=self.device)
@@ -514,6 +600,15 @@ class CODA(ModelSelector):
         onehot = torch.nn.functional.one_hot(
             self.classes[:, idx], self.C).to(self.dirichlets.dtype)  # (Hl, C)
         self.dirichlets[:, int(true_class)] += self.update_strength * onehot
+        if self._replicated:
+            # keep the replicated global Beta view current - a local
+            # elementwise update from the (gathered-once) global classes,
+            # identical on every rank, no collective
+            hit = self.classes_global[:, idx].long() == int(true_class)
+            self._alpha_g[:, int(true_class)] += \
+                self.update_strength * hit.float()
+            self._beta_g[:, int(true_class)] += \
+                self.update_strength * (~hit).float()
         self._tables_dirty.add(int(true_class))
         self._posterior_version += 1
         delta = ops.pi_hat_delta(self.dataset.preds, self.classes[:, idx])
@@ -549,9 +644,9 @@ class CODA(ModelSelector):
     # ------------------------------------------------------------------
     def get_pbest(self):
         """Marginal P(best) over models (K15): (H,) in GLOBAL model order."""
-        rows = self._pbest_rows_before()                   # (C, Hl)
-        marg_local = (rows * self.pi_hat.view(-1, 1)).sum(0)  # (Hl,)
-        if self.comm.is_distributed:
+        rows = self._pbest_rows_before()     # (C, Hl) / (C, H) replicated
+        marg_local = (rows * self.pi_hat.view(-1, 1)).sum(0)
+        if self.comm.is_distributed and not self._replicated:
             gathered = self.comm.all_gather_cat(marg_local, dim=0)
             pbest = gathered[self.comm.unshard_order(self.H).to(gathered.device)]
         else:
