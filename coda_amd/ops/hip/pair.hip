@@ -334,6 +334,144 @@ pair_gemm_entropy128_kernel(const hip_bfloat16* __restrict__ a16,
 }
 
 // ---------------------------------------------------------------------
+// Wide-H pipeline (288 < 2H <= 4096, i.e. the multi-GPU H=256..1024
+// pools): the fused B-resident tile cannot hold egw[c] or the M tile in
+// LDS, so the pairing GEMM writes M (K, 2H) bf16 to global with a
+// standard 128x128 double-buffered MFMA tile (B traffic divided by 128
+// vs the 16-pair fused tile), and a wave-per-pair entropy kernel
+// consumes it with the same vmask epilogue.
+// ---------------------------------------------------------------------
+#define WSTRIDE 40   // 32 k-elems + 8 pad (bf16) per LDS row
+
+__global__ void __launch_bounds__(BLOCK)
+pair_gemm_wide_kernel(const hip_bfloat16* __restrict__ a16,  // (K, P)
+                      const hip_bfloat16* __restrict__ egw,  // (C, 2H, P)
+                      const int* __restrict__ pair_c,        // (K,)
+                      hip_bfloat16* __restrict__ mout,       // (K, 2H)
+                      int twoH) {
+    extern __shared__ char smem[];
+    // [2 buffers][128 rows][WSTRIDE] for A and B chunks
+    hip_bfloat16* a_lds = reinterpret_cast<hip_bfloat16*>(smem);
+    hip_bfloat16* b_lds = a_lds + 2 * 128 * WSTRIDE;
+
+    const int k0 = blockIdx.x * 128;
+    const int j0 = blockIdx.y * 128;
+    const int c = pair_c[k0];
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6, lane = tid & 63;
+    const int row16 = lane & 15, kgrp = lane >> 4;
+    const hip_bfloat16* egw_c = egw + (size_t)c * twoH * P_POINTS;
+
+    // stage one 128x32 chunk of A or B: thread t loads 16 B (8 elems)
+    auto stage = [&](hip_bfloat16* dst, const hip_bfloat16* src,
+                     int kk, bool valid_rows) {
+        const int row = tid >> 1, half = tid & 1;
+        const uint4* g = reinterpret_cast<const uint4*>(
+            src + (size_t)row * P_POINTS + kk + half * 16);
+        uint4* d = reinterpret_cast<uint4*>(
+            reinterpret_cast<char*>(dst)
+            + (size_t)row * WSTRIDE * 2 + half * 32);
+        if (valid_rows) { d[0] = g[0]; d[1] = g[1]; }
+    };
+
+    stage(a_lds, a16 + (size_t)k0 * P_POINTS, 0, true);
+    stage(b_lds, egw_c + (size_t)j0 * P_POINTS, 0,
+          j0 + (tid >> 1) < twoH);
+    __syncthreads();
+
+    f32x4 acc[2][8];
+#pragma unroll
+    for (int r = 0; r < 2; ++r)
+#pragma unroll
+        for (int jt = 0; jt < 8; ++jt) acc[r][jt] = {0.f, 0.f, 0.f, 0.f};
+
+    for (int kk = 0; kk < P_POINTS; kk += 32) {
+        const int cur = (kk >> 5) & 1;
+        if (kk + 32 < P_POINTS) {
+            stage(a_lds + (cur ^ 1) * 128 * WSTRIDE,
+                  a16 + (size_t)k0 * P_POINTS, kk + 32, true);
+            stage(b_lds + (cur ^ 1) * 128 * WSTRIDE,
+                  egw_c + (size_t)j0 * P_POINTS, kk + 32,
+                  j0 + (threadIdx.x >> 1) < twoH);
+        }
+        const hip_bfloat16* ab = a_lds + cur * 128 * WSTRIDE;
+        const hip_bfloat16* bb = b_lds + cur * 128 * WSTRIDE;
+#pragma unroll
+        for (int r = 0; r < 2; ++r) {
+            // wave w owns pair rows [wave*32 + r*16 + row16]
+            const bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
+                ab + (size_t)(wave * 32 + r * 16 + row16) * WSTRIDE
+                + kgrp * 8);
+#pragma unroll
+            for (int jt = 0; jt < 8; ++jt) {
+                const bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
+                    bb + (size_t)(jt * 16 + row16) * WSTRIDE + kgrp * 8);
+                acc[r][jt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    afrag, bfrag, acc[r][jt], 0, 0, 0);
+            }
+        }
+        __syncthreads();
+    }
+
+    // write M tile: D col = row16 (the B row j), D rows = kgrp*4+e
+#pragma unroll
+    for (int r = 0; r < 2; ++r) {
+#pragma unroll
+        for (int jt = 0; jt < 8; ++jt) {
+            const int j = j0 + jt * 16 + row16;
+            if (j >= twoH) continue;
+#pragma unroll
+            for (int e = 0; e < 4; ++e) {
+                const int k = k0 + wave * 32 + r * 16 + kgrp * 4 + e;
+                mout[(size_t)k * twoH + j] =
+                    hip_bfloat16(acc[r][jt][e]);
+            }
+        }
+    }
+}
+
+__global__ void __launch_bounds__(BLOCK)
+pair_entropy_wide_kernel(const hip_bfloat16* __restrict__ m,  // (K, 2H)
+                         const unsigned* __restrict__ vmask,  // (K, W)
+                         const int* __restrict__ pair_c,      // (K,)
+                         int W,
+                         const float* __restrict__ pi_hat,
+                         const float* __restrict__ pbest_before,
+                         const float* __restrict__ mixture0,
+                         float* __restrict__ h_after,         // (K,)
+                         int K, int H) {
+    const int k = blockIdx.x * 4 + (threadIdx.x >> 6);
+    if (k >= K) return;
+    const int lane = threadIdx.x & 63;
+    const int c = pair_c[k];
+    const unsigned* vm = vmask + (size_t)k * W;
+    const hip_bfloat16* mrow = m + (size_t)k * 2 * H;
+
+    float val[32];   // H <= 2048 -> <= 32 h per lane
+    float tot = 0.f;
+    int i = 0;
+    for (int h = lane; h < H; h += 64, ++i) {
+        const int v = (vm[h >> 5] >> (h & 31)) & 1;
+        val[i] = (float)mrow[2 * h + v];
+        tot += val[i];
+    }
+    tot = wave_reduce(tot);
+    const float inv = 1.0f / fmaxf(tot, 1e-30f);
+    const float pic = pi_hat[c];
+    float ent = 0.f;
+    i = 0;
+    for (int h = lane; h < H; h += 64, ++i) {
+        const float pb = val[i] * inv;
+        const float mm = fmaxf(
+            mixture0[h] + pic * (pb - pbest_before[(size_t)c * H + h]),
+            1e-12f);
+        ent += -mm * __log2f(mm);
+    }
+    ent = wave_reduce(ent);
+    if (lane == 0) h_after[k] = ent;
+}
+
+// ---------------------------------------------------------------------
 // Finalize: q[b] = H_before - (1/rowsum) * (arow . h_base
 //                 + sum_{pairs of b} arow[c]*(h_after - h_base[c]))
 // One wave per candidate; fixed reduction order (deterministic).
@@ -438,7 +576,33 @@ torch::Tensor pair_gemm_entropy(torch::Tensor a16, torch::Tensor egw,
         a16.data_ptr());
     const auto eb16 = reinterpret_cast<const hip_bfloat16*>(
         egw.data_ptr());
-    if (tile == 128) {
+    if (tile == 128 && 2 * H > 288) {
+        // wide-H split pipeline: M to global, entropy second pass
+        TORCH_CHECK(2 * H <= 4096, "pair engine caps at H = 2048");
+        const int twoH = 2 * H;
+        auto mout = torch::empty({(long)K, (long)twoH},
+                                 a16.options());
+        const size_t shmem = 4 * 128 * WSTRIDE * sizeof(hip_bfloat16);
+        dim3 grid(K / 128, (twoH + 127) / 128);
+        hipLaunchKernelGGL(pairops::pair_gemm_wide_kernel, grid,
+                           dim3(BLOCK), shmem, stream.stream(), ab16,
+                           eb16, pair_c.data_ptr<int>(),
+                           reinterpret_cast<hip_bfloat16*>(
+                               mout.data_ptr()),
+                           twoH);
+        hipLaunchKernelGGL(pairops::pair_entropy_wide_kernel,
+                           dim3((K + 3) / 4), dim3(BLOCK), 0,
+                           stream.stream(),
+                           reinterpret_cast<const hip_bfloat16*>(
+                               mout.data_ptr()),
+                           reinterpret_cast<const unsigned*>(
+                               vmask.data_ptr<int>()),
+                           pair_c.data_ptr<int>(), (int)vmask.size(1),
+                           pi_hat.data_ptr<float>(),
+                           pbest_before.data_ptr<float>(),
+                           mixture0.data_ptr<float>(),
+                           h_after.data_ptr<float>(), K, H);
+    } else if (tile == 128) {
         TORCH_CHECK(2 * H <= 288, "128-pair tile needs 2H <= 288");
         const size_t phase1 = (size_t)2 * H * BSTRIDE
                             * sizeof(hip_bfloat16);

@@ -50,12 +50,13 @@ PAIR_TILE = 16  # minimum tile; build_pairs may pad to 64 (see below)
 def tile_for(H: int, B: int) -> int:
     """MFMA tile height (pairs per class-uniform tile).
 
-    128-pair tiles keep the whole egw[c] B operand resident in LDS
-    (divides the dominant GEMM traffic by 8 vs 16-pair tiles) but need
-    2H <= 288 to fit, and only pay off when the padding (<= tile-1
-    pairs per class) is small against the real pair count.
+    128-pair tiles divide the dominant GEMM B traffic by 8 vs 16-pair
+    tiles (H <= 144: B resident in LDS, fused epilogue; larger H: the
+    split wide pipeline with a 128x128 M-writing GEMM). They only pay
+    off when the padding (<= tile-1 pairs per class) is small against
+    the real pair count - small candidate sets use 16-pair tiles.
     """
-    return 128 if (H <= 144 and B >= 4096) else PAIR_TILE
+    return 128 if B >= 4096 else PAIR_TILE
 
 
 class PairStructure(NamedTuple):

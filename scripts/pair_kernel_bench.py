@@ -115,3 +115,27 @@ for abl, name in [(0, "full"), (1, "no-epilogue"), (2, "no-mfma"),
                A16, tables.egw, ps.vmask, ps.pair_c,
                pi_hat.contiguous(), pbest_before.contiguous(),
                mixture0.contiguous(), 128, abl))
+
+# wide-H pipeline (the multi-GPU H>144 shapes): forced tile=128
+if args.h > 144:
+    pass  # main loop above covered it
+else:
+    # quick wide-path probe at H=256 on a subset
+    H2 = 256
+    preds2, _ = bench.synth_preds(list(range(H2)), 8192, C, dev)
+    cls2 = preds2.argmax(-1)
+    g2 = torch.Generator().manual_seed(1)
+    dl2 = (torch.rand(H2, C, C, generator=g2) * 2 + 0.5).to(dev)
+    a2, b2 = R.dirichlet_to_beta(dl2)
+    t2 = pops.attach_pair_tables(tops.table_precompute(a2, b2))
+    pb2 = R.pbest_from_beta(a2.t().contiguous(), b2.t().contiguous())
+    mix2, H0b = R.mixture_entropy(pb2, pi_hat)
+    ids2 = torch.arange(8192, device=dev)
+    cr2 = cls2[:, ids2].t().to(torch.int32).contiguous()
+    ps2 = pops.build_pairs(cr2, ids2, C, tile=128)
+    A2 = O._ext.pair_dsum_es(t2.delta16, ps2.pair_c, ps2.seg_off, ps2.seg_h)
+    print(f"--- wide H={H2} K={ps2.K}")
+    timeit("gemm_entropy[wide256]",
+           lambda: O._ext.pair_gemm_entropy(
+               A2, t2.egw, ps2.vmask, ps2.pair_c, pi_hat.contiguous(),
+               pb2.contiguous(), mix2.contiguous(), 128))
