@@ -560,3 +560,37 @@ class TestPairKernels:
             assert gap < 2e-3 * float(q_t[s].max()), \
                 (s, t_t[s], t_p[s], gap)
             break
+
+    def test_pair_wide_pipeline_vs_eager(self, dev):
+        """288 < 2H: the split wide pipeline (M-writing 128x128 GEMM +
+        wave-per-pair entropy) vs the eager pair math."""
+        from coda_amd.ops import pair as pops
+        from coda_amd.ops import table as tops
+        from coda_amd.ops import reference as R
+        from tests.test_pair import _random_problem
+        H, N, C = 160, 300, 30
+        (preds, cls, dirichlets, pi_hat, adjusted,
+         row_sums) = _random_problem(H, N, C, seed=7)
+        alpha_cc, beta_cc = R.dirichlet_to_beta(dirichlets)
+        alpha_cc, beta_cc = alpha_cc.to(dev), beta_cc.to(dev)
+        tables = pops.attach_pair_tables(
+            tops.table_precompute(alpha_cc, beta_cc))
+        pbest_before = R.pbest_from_beta(alpha_cc.t().contiguous(),
+                                         beta_cc.t().contiguous())
+        mixture0, H_before = R.mixture_entropy(pbest_before,
+                                               pi_hat.to(dev))
+        ids = torch.arange(N, device=dev)
+        cls_rows = cls.to(dev)[:, ids].t().to(torch.int32).contiguous()
+        ps = pops.build_pairs(cls_rows, ids, C, tile=128)
+        assert 2 * H > 288  # wide path engaged
+        eig_k = pops.eig_pairs(
+            tables, ps, cls_rows, pbest_before, pi_hat.to(dev),
+            mixture0, H_before, adjusted.to(dev), row_sums.to(dev))
+        h_eager = pops.pair_h_after(tables, ps, cls_rows, pbest_before,
+                                    pi_hat.to(dev), mixture0)
+        eig_e = pops.eig_from_pairs(h_eager, ps, adjusted.to(dev),
+                                    row_sums.to(dev),
+                                    H_before)[ps.cand_ids]
+        scale = float(eig_e.abs().max())
+        err = float((eig_k - eig_e).abs().max())
+        assert err < max(5e-3 * scale, 3e-4), (err, scale)
