@@ -85,7 +85,9 @@ __device__ __forceinline__ float wave_reduce(float v) {
 // ---------------------------------------------------------------------
 __global__ void __launch_bounds__(BLOCK)
 pair_dsum_es_kernel(const _Float16* __restrict__ delta16,  // (C, H, P)
+                    const float* __restrict__ dall,        // (C, P)
                     const int* __restrict__ pair_c,        // (K,)
+                    const int* __restrict__ pair_neg,      // (K,)
                     const int* __restrict__ seg_off,       // (K+1,)
                     const int* __restrict__ seg_h,         // (S,)
                     hip_bfloat16* __restrict__ a16,        // (K, P)
@@ -114,6 +116,19 @@ pair_dsum_es_kernel(const _Float16* __restrict__ delta16,  // (C, H, P)
             acc[4 + j] += (float)d1[j];
             acc[8 + j] += (float)d2[j];
             acc[12 + j] += (float)d3[j];
+        }
+    }
+    if (pair_neg[k]) {
+        // complement segment: dsum = (sum over ALL models) - partial
+        const float4* da = reinterpret_cast<const float4*>(
+            dall + (size_t)c * P_POINTS + p0);
+#pragma unroll
+        for (int q = 0; q < 4; ++q) {
+            const float4 d = da[q];
+            acc[4 * q] = d.x - acc[4 * q];
+            acc[4 * q + 1] = d.y - acc[4 * q + 1];
+            acc[4 * q + 2] = d.z - acc[4 * q + 2];
+            acc[4 * q + 3] = d.w - acc[4 * q + 3];
         }
     }
     ushort4 out[4];
@@ -536,21 +551,24 @@ __global__ void mfma_probe_kernel(const float* __restrict__ a,   // (16,32)
 // Host bindings
 // ---------------------------------------------------------------------
 
-torch::Tensor pair_dsum_es(torch::Tensor delta16, torch::Tensor pair_c,
+torch::Tensor pair_dsum_es(torch::Tensor delta16, torch::Tensor dall,
+                           torch::Tensor pair_c, torch::Tensor pair_neg,
                            torch::Tensor seg_off, torch::Tensor seg_h) {
     TORCH_CHECK(delta16.is_cuda() && delta16.dtype() == torch::kFloat16);
     TORCH_CHECK(delta16.size(-1) == P_POINTS);
+    TORCH_CHECK(dall.dtype() == torch::kFloat32);
     const int H = delta16.size(1);
     const int K = pair_c.size(0);
     auto a16 = torch::empty({K, P_POINTS},
                             delta16.options().dtype(torch::kBFloat16));
     auto stream = c10::hip::getCurrentHIPStream();
-    dim3 grid((K + 3) / 4);
+    dim3 grid((K + 15) / 16);
     hipLaunchKernelGGL(pairops::pair_dsum_es_kernel, grid, dim3(BLOCK), 0,
                        stream.stream(),
                        reinterpret_cast<const _Float16*>(
                            delta16.data_ptr()),
-                       pair_c.data_ptr<int>(),
+                       dall.data_ptr<float>(),
+                       pair_c.data_ptr<int>(), pair_neg.data_ptr<int>(),
                        seg_off.data_ptr<int>(), seg_h.data_ptr<int>(),
                        reinterpret_cast<hip_bfloat16*>(a16.data_ptr()),
                        K, H);

@@ -81,6 +81,11 @@ class PairStructure(NamedTuple):
     # static v-select; reading cls rows per pair costs 1.6 ms/step at
     # the headline shape, the bitmask 0.05 ms)
     vmask: torch.Tensor = None
+    # (K,) int32 — 1 where the stored segment is the COMPLEMENT of the
+    # hit set (majority pairs: most models predict the true-ish label,
+    # so summing the few non-hitting models' deltas and subtracting
+    # from the per-class total nearly halves the dsum read traffic)
+    pair_neg: torch.Tensor = None
 
     @property
     def K(self) -> int:
@@ -157,11 +162,40 @@ def build_pairs(cls_rows: torch.Tensor, cand_ids: torch.Tensor,
     vmask.index_put_((word,), bit, accumulate=True)
     vmask = vmask.view(K, W)
 
+    # complement majority segments: a pair hit by > H/2 models stores
+    # the NON-hitting models instead (dsum = dall[c] - sum(complement)).
+    # At most one such pair per candidate; its complement is exactly the
+    # candidate's other entries.
+    flip = seg_len_real > (H // 2)                          # (K_real,)
+    pair_neg = torch.zeros(K, dtype=torch.int32, device=device)
+    if bool(flip.any()):
+        pair_neg[pos[flip]] = 1
+        entry_pid = pid                                     # (B*H,)
+        entry_b = pr_b[entry_pid]
+        entry_h = (order % H)
+        q_of_b = torch.full((B,), -1, dtype=torch.long, device=device)
+        q_of_b[pr_b[flip]] = torch.nonzero(flip, as_tuple=True)[0]
+        keep = ~flip[entry_pid]
+        t1 = pos[entry_pid[keep]]
+        h1 = entry_h[keep]
+        qb = q_of_b[entry_b]
+        comp = (qb >= 0) & (entry_pid != qb)
+        t2 = pos[qb[comp]]
+        h2 = entry_h[comp]
+        tgt = torch.cat([t1, t2])
+        hh = torch.cat([h1, h2])
+        srt = torch.argsort(tgt * H + hh)                   # (tgt, h) order
+        seg_h = hh[srt].to(torch.int32)
+        seg_len2 = torch.bincount(tgt, minlength=K)
+        seg_off = torch.zeros(K + 1, dtype=torch.int32, device=device)
+        seg_off[1:] = seg_len2.cumsum(0).to(torch.int32)
+
     return PairStructure(cand_ids=cand_ids.long(), pair_b=pair_b,
                          pair_c=pair_c, seg_off=seg_off, seg_h=seg_h,
                          base_pos=run_off[:-1].clone(), n_real=K_real,
                          tile=tile, cand_off=cand_off,
-                         cand_pairs=cand_pairs, vmask=vmask)
+                         cand_pairs=cand_pairs, vmask=vmask,
+                         pair_neg=pair_neg)
 
 
 def pair_h_after(tables, ps: PairStructure, cls_rows: torch.Tensor,
@@ -188,6 +222,11 @@ def pair_h_after(tables, ps: PairStructure, cls_rows: torch.Tensor,
     if ps.seg_h.numel():
         dsum.index_add_(0, seg_pair,
                         delta[pc_long[seg_pair], ps.seg_h.long()])
+    if ps.pair_neg is not None and bool(ps.pair_neg.any()):
+        dall = tables.dall if getattr(tables, "dall", None) is not None \
+            else delta.sum(1)
+        neg = ps.pair_neg.bool()
+        dsum[neg] = dall[pc_long[neg]] - dsum[neg]
     A = torch.exp2(dsum + s_base[pc_long]) * w              # (K, P)
     M = torch.einsum('kp,kjp->kj', A,
                      EG.reshape(C, 2 * H, P)[pc_long])      # (K, 2H)
@@ -244,7 +283,8 @@ def eig_pairs(tables, ps: PairStructure, cls_rows: torch.Tensor,
     if (EG.is_cuda and P == PBEST_NUM_POINTS
             and getattr(tables, "egw", None) is not None
             and O._want_hip(EG)):
-        A16 = O._ext.pair_dsum_es(tables.delta16, ps.pair_c, ps.seg_off,
+        A16 = O._ext.pair_dsum_es(tables.delta16, tables.dall,
+                                  ps.pair_c, ps.pair_neg, ps.seg_off,
                                   ps.seg_h)                 # (K, P) bf16
         h_after = O._ext.pair_gemm_entropy(
             A16, tables.egw, ps.vmask, ps.pair_c,
@@ -278,7 +318,8 @@ def attach_pair_tables(tables):
     egw = (EG.reshape(C, 2 * H, P)
            * esb.unsqueeze(1)).to(torch.bfloat16).contiguous()
     return tables._replace(egw=egw,
-                           delta16=tables.delta.to(torch.float16))
+                           delta16=tables.delta.to(torch.float16),
+                           dall=tables.delta.sum(1).contiguous())
 
 
 def update_egw_rows(tables, rows) -> None:
@@ -291,3 +332,5 @@ def update_egw_rows(tables, rows) -> None:
                          * esb.unsqueeze(0)).to(torch.bfloat16)
         if tables.delta16 is not None:
             tables.delta16[c] = tables.delta[c].to(torch.float16)
+        if tables.dall is not None:
+            tables.dall[c] = tables.delta[c].sum(0)

@@ -78,6 +78,8 @@ class EigTables(NamedTuple):
     # (C, H, P) fp16 = the dsum kernel's halved-traffic delta table.
     egw: torch.Tensor = None
     delta16: torch.Tensor = None
+    # (C, P) fp32 per-class delta totals (complement-segment dsum)
+    dall: torch.Tensor = None
 
 
 def table_precompute(alpha_cc: torch.Tensor, beta_cc: torch.Tensor,

@@ -535,6 +535,8 @@ class CODA(ModelSelector):
                 t.delta16.index_copy_(
                     0, y_t, t.delta.index_select(0, y_t)
                     .to(torch.float16))
+                t.dall.index_copy_(
+                    0, y_t, t.delta.index_select(0, y_t).sum(1))
         # posterior rows for the next acquisition / get_pbest
         alpha_cc, beta_cc = ops.dirichlet_to_beta(self.dirichlets)
         rows = ops.pbest_from_beta(alpha_cc.t().contiguous(),

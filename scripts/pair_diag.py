@@ -34,8 +34,8 @@ for H, N, C in [(10, 120, 7), (3, 60, 2), (16, 200, 126), (128, 500, 50),
     h_e = pops.pair_h_after(tables, ps, cls_rows, pbest_before,
                             pi_hat.to(dev), mixture0)
     h_k = O._ext.pair_gemm_entropy(
-        O._ext.pair_dsum_es(tables.delta16, ps.pair_c, ps.seg_off,
-                            ps.seg_h),
+        O._ext.pair_dsum_es(tables.delta16, tables.dall, ps.pair_c,
+                            ps.pair_neg, ps.seg_off, ps.seg_h),
         tables.egw, ps.vmask, ps.pair_c,
         pi_hat.to(dev).contiguous(), pbest_before.contiguous(),
         mixture0.contiguous(), ps.tile)

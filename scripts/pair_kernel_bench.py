@@ -69,11 +69,11 @@ for tile in (16, 128):
     ps = pops.build_pairs(cls_rows, ids, C, tile=tile)
     print(f"--- tile={tile}  K={ps.K} n_real={ps.n_real} "
           f"avg_seg={ps.seg_h.numel()/max(ps.n_real,1):.1f}")
-    A16 = O._ext.pair_dsum_es(tables.delta16, ps.pair_c, ps.seg_off,
-                              ps.seg_h)
+    A16 = O._ext.pair_dsum_es(tables.delta16, tables.dall, ps.pair_c,
+                              ps.pair_neg, ps.seg_off, ps.seg_h)
     timeit(f"dsum_es[t{tile}]",
-           lambda: O._ext.pair_dsum_es(tables.delta16, ps.pair_c,
-                                       ps.seg_off, ps.seg_h))
+           lambda: O._ext.pair_dsum_es(tables.delta16, tables.dall, ps.pair_c,
+                                       ps.pair_neg, ps.seg_off, ps.seg_h))
     h_after = O._ext.pair_gemm_entropy(
         A16, tables.egw, ps.vmask, ps.pair_c,
         pi_hat.contiguous(), pbest_before.contiguous(),
@@ -93,8 +93,8 @@ for tile in (16, 128):
 ps16 = pops.build_pairs(cls_rows[:4096], ids[:4096], C, tile=16)
 ps64 = pops.build_pairs(cls_rows[:4096], ids[:4096], C, tile=128)
 for ps, name in ((ps16, "t16"), (ps64, "t64")):
-    A = O._ext.pair_dsum_es(tables.delta16, ps.pair_c, ps.seg_off,
-                            ps.seg_h)
+    A = O._ext.pair_dsum_es(tables.delta16, tables.dall, ps.pair_c,
+                            ps.pair_neg, ps.seg_off, ps.seg_h)
     h = O._ext.pair_gemm_entropy(A, tables.egw, ps.vmask, ps.pair_c,
                                  pi_hat.contiguous(),
                                  pbest_before.contiguous(),
@@ -107,7 +107,7 @@ for ps, name in ((ps16, "t16"), (ps64, "t64")):
 
 # phase ablation at tile=128 (guide: symptom -> diagnosis loop)
 ps = pops.build_pairs(cls_rows, ids, C, tile=128)
-A16 = O._ext.pair_dsum_es(tables.delta16, ps.pair_c, ps.seg_off, ps.seg_h)
+A16 = O._ext.pair_dsum_es(tables.delta16, tables.dall, ps.pair_c, ps.pair_neg, ps.seg_off, ps.seg_h)
 for abl, name in [(0, "full"), (1, "no-epilogue"), (2, "no-mfma"),
                   (3, "no-cls-reads")]:
     timeit(f"gemm128[{name}]",
@@ -133,7 +133,7 @@ else:
     ids2 = torch.arange(8192, device=dev)
     cr2 = cls2[:, ids2].t().to(torch.int32).contiguous()
     ps2 = pops.build_pairs(cr2, ids2, C, tile=128)
-    A2 = O._ext.pair_dsum_es(t2.delta16, ps2.pair_c, ps2.seg_off, ps2.seg_h)
+    A2 = O._ext.pair_dsum_es(t2.delta16, t2.dall, ps2.pair_c, ps2.pair_neg, ps2.seg_off, ps2.seg_h)
     print(f"--- wide H={H2} K={ps2.K}")
     timeit("gemm_entropy[wide256]",
            lambda: O._ext.pair_gemm_entropy(

@@ -44,11 +44,16 @@ class TestPairStructure:
             b = int(ps.pair_b[k])
             if b < 0:
                 assert int(ps.seg_off[k + 1]) == int(ps.seg_off[k])
+                assert int(ps.pair_neg[k]) == 0
                 continue
             c = int(ps.pair_c[k])
-            hs = ps.seg_h[int(ps.seg_off[k]):int(ps.seg_off[k + 1])]
+            hs = sorted(ps.seg_h[int(ps.seg_off[k]):
+                                 int(ps.seg_off[k + 1])].tolist())
+            if int(ps.pair_neg[k]):
+                # majority pair stores the complement of its hit set
+                hs = sorted(set(range(9)) - set(hs))
             assert (b, c) not in seen
-            seen[(b, c)] = sorted(hs.tolist())
+            seen[(b, c)] = hs
         for b in range(37):
             for c in range(6):
                 expect = [h for h in range(9) if int(cls_rows[b, h]) == c]
