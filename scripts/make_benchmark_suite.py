@@ -17,6 +17,10 @@ SUITE = [
     ("synth_wilds_like",    16, 8000,  62, 0.75, 0.40),
     ("synth_glue_like",     6,  1000,  3,  0.85, 0.60),
     ("synth_wide_pool",     48, 4000,  10, 0.88, 0.50),
+    # DomainNet-shaped at real scale: the largest
+    # reference tasks are H=10, N~100-200k, C=126 DomainNet pairs
+    # (reference paper/fig3.py:129-193 memory table)
+    ("synth_domainnet_full", 10, 200000, 126, 0.65, 0.30),
 ]
 
 
