@@ -75,6 +75,7 @@ def load_state_dict(selector, state: Dict[str, Any]):
         selector._posterior_version += 1
         selector._pbest_rows_cache = (-1, None)
         selector._label_graph = None  # graph buffers alias replaced state
+        selector._acq_graph = None
         selector._pairs_static = None  # hit structure covers a stale set
         selector._pair_row_of = None
         selector._active_mask = None

@@ -77,6 +77,11 @@ class CODA(ModelSelector):
         # ~20 static-shape launches replayed as one graph. Auto on
         # single-device GPU; CODA_AMD_NO_GRAPH=1 disables.
         self._label_graph = None
+        # hipGraph of the full-pool pair acquisition (static hit
+        # structure + static table/posterior buffers -> the whole
+        # get_next tensor pipeline incl. argmax/tie counting replays as
+        # one graph; one host sync per step)
+        self._acq_graph = None
         # (debug guards synchronize per op and cannot run under stream
         # capture - graph replay is a production-mode path)
         self._use_label_graph = (
@@ -442,8 +447,84 @@ class CODA(ModelSelector):
 
         return torch.cat(eig_chunks), candidate_ids
 
+    # -- hipGraph full-pool acquisition --------------------------------
+    def _acq_body(self):
+        """The whole get_next tensor pipeline on the STATIC buffers:
+        mixture entropy from the label-graph's posterior rows/pi_hat,
+        the three pair kernels, labeled-candidate masking, argmax and
+        tie counting. Captured once; identical math to _eig_pair."""
+        t = self._tables
+        ps, _ = self._pairs_static
+        rows, pi = self._g_rows, self._g_pi
+        mixture0, H0 = ops.mixture_entropy(rows, pi)
+        A16 = ops._ext.pair_dsum_es(t.delta16, t.dall, ps.pair_c,
+                                    ps.pair_neg, ps.seg_off, ps.seg_h)
+        h_after = ops._ext.pair_gemm_entropy(
+            A16, t.egw, ps.vmask, ps.pair_c, pi, rows,
+            mixture0.contiguous(), ps.tile)
+        # H_before enters as an in-graph tensor op (the kernel's scalar
+        # argument would be frozen at capture value)
+        q0 = ops._ext.pair_eig_finalize(
+            h_after, ps.pair_c, self._ps_base32, ps.cand_off,
+            ps.cand_pairs, ps.cand_ids, self._adjusted, self._row_sums,
+            0.0)
+        q = H0 + q0
+        q = torch.where(self._active_mask, q,
+                        torch.full_like(q, float("-inf")))
+        best_val, best_idx = q.max(0)
+        nt = (torch.isclose(q, best_val, rtol=1e-8)
+              & self._active_mask).sum()
+        self._acq_q = q
+        self._acq_out.copy_(torch.stack(
+            [best_val.double(), best_idx.double(), nt.double()]))
+
+    def _acq_result(self):
+        bv, bi, nt = self._acq_out.cpu().tolist()
+        bi, nt = int(bi), int(nt)
+        if nt > 1:
+            # same tie semantics as the eager path: active candidates
+            # ascend by point id in both orderings
+            q = self._acq_q
+            ties = (torch.isclose(q, q.max(), rtol=1e-8)
+                    & self._active_mask)
+            pos = random.choice(
+                torch.nonzero(ties, as_tuple=True)[0].tolist())
+            self.stochastic = True
+            return self._pairs_ids_host[pos], float(q[pos])
+        return self._pairs_ids_host[bi], bv
+
+    def _graphed_acquire(self):
+        if self._acq_graph is None:
+            ps, _ = self._pairs_static
+            self._ps_base32 = ps.base_pos.to(torch.int32)
+            self._pairs_ids_host = ps.cand_ids.cpu().tolist()
+            self._acq_out = torch.zeros(3, dtype=torch.float64,
+                                        device=self.device)
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                self._acq_body()
+            torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.synchronize()
+            first = self._acq_result()  # before capture rebinds _acq_q
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self._acq_body()
+            self._acq_graph = g
+            return first
+        self._acq_graph.replay()
+        return self._acq_result()
+
     # ------------------------------------------------------------------
     def get_next_item_to_label(self):
+        if (self.q == "eig" and self._use_label_graph
+                and self._label_graph is not None
+                and self._pairs_static is not None
+                and not DEBUG_VIZ
+                and not (self.prefilter_n
+                         and len(self._active_candidates)
+                         > self.prefilter_n)):
+            return self._graphed_acquire()
         if self.q == "eig":
             q_vals, cand = self.eig_batched()
         elif self.q == "iid":
