@@ -230,17 +230,39 @@ pair_gemm_entropy16_kernel(const hip_bfloat16* __restrict__ a16,
 }
 
 // ---------------------------------------------------------------------
-// 128-pair B-resident tile GEMM+entropy (2H <= 288): the tile's whole
+// 128-pair B-resident tile GEMM+entropy (2H <= 272): the tile's whole
 // egw[c] table (2H x P bf16) is staged into LDS ONCE with a coalesced
-// row-major copy, then 8 waves run the full K loop against it with no
-// further barriers - B traffic drops to K/128 reads of 131 KB. The A
-// operand (8 KB/wave) streams from L2. JT = #16-col tiles is a
-// template parameter so the accumulator array stays in registers.
+// row-major copy at a conflict-free stride (266 elems = 133 words; the
+// b128 16-lane service groups mix row16 0..3 with kgrp 0..3, so the
+// bank residues 5*r + 4*k must be distinct - a x264 stride collided
+// 4-way, 4.7 extra LDS cycles per read measured by PMC), then 8 waves
+// run the whole K loop against it barrier-free with the A operand
+// streamed from L2. The epilogue is REGISTER-RESIDENT: each lane holds
+// pairs kgrp*4+r at columns jt*16+row16, the static vmask selects the
+// v variant per (pair, h) (the mask word index (8*jt)>>5 is
+// compile-time), and 16-lane DPP row reductions produce tot/entropy -
+// the (128, 2H) M tile never exists in LDS or HBM.
 // ---------------------------------------------------------------------
-#define BSTRIDE (P_POINTS + 8)   // bf16 elems; +8 de-aliases banks
+#define BSTRIDE 266   // bf16 elems; 133 words == 5 mod 64
 #define BLOCK2 512
 
-template <int JT, int ABL = 0>   // ABL: 1=skip epilogue, 2=skip MFMA, 3=no cls reads
+__device__ __forceinline__ float row16_reduce(float v) {
+#ifdef CODA_DPP_SCAN
+    v = dpp_add<0x111, 0xf>(v);  // row_shr:1
+    v = dpp_add<0x112, 0xf>(v);  // row_shr:2
+    v = dpp_add<0x114, 0xf>(v);  // row_shr:4
+    v = dpp_add<0x118, 0xf>(v);  // row_shr:8 -> lane15 of each row
+#else
+#pragma unroll
+    for (int off = 1; off < 16; off <<= 1) {
+        float n = __shfl_up(v, off, 16);
+        if ((threadIdx.x & 15) >= off) v += n;
+    }
+#endif
+    return __shfl(v, 15, 16);    // broadcast row total to all 16 lanes
+}
+
+template <int JT>
 __global__ void __launch_bounds__(BLOCK2)
 pair_gemm_entropy128_kernel(const hip_bfloat16* __restrict__ a16,
                             const hip_bfloat16* __restrict__ egw,
@@ -285,7 +307,7 @@ pair_gemm_entropy128_kernel(const hip_bfloat16* __restrict__ a16,
     const hip_bfloat16* arow =
         a16 + (size_t)(k0 + wave * 16 + row16) * P_POINTS;
 #pragma unroll
-    for (int kk = 0; kk < (ABL == 2 ? 0 : P_POINTS); kk += 32) {
+    for (int kk = 0; kk < P_POINTS; kk += 32) {
         const bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
             arow + kk + kgrp * 8);
 #pragma unroll
@@ -299,53 +321,61 @@ pair_gemm_entropy128_kernel(const hip_bfloat16* __restrict__ a16,
                 afrag, bfrag, acc[jt], 0, 0, 0);
         }
     }
-    __syncthreads();
 
-    // phase 2: spill accumulators to LDS (overlapping the B buffer)
-    float* m_tile = reinterpret_cast<float*>(smem);
+    // register-resident epilogue: lane owns pairs kbase..kbase+3 (the
+    // accumulator rows) at columns jt*16+row16. For column j: h = j>>1
+    // and it contributes to pair r iff vmask bit h == (j&1).
+    const int kbase = k0 + wave * 16 + kgrp * 4;
+    unsigned vmw[4][(JT + 3) / 4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+#pragma unroll
+        for (int w = 0; w < (JT + 3) / 4; ++w)
+            vmw[r][w] = (w < W)
+                ? vmask[(size_t)(kbase + r) * W + w] : 0u;
+
+    float tot[4] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
     for (int jt = 0; jt < JT; ++jt) {
         const int j = jt * 16 + row16;
-        if (j < twoH) {
+        if (j >= twoH) continue;
+        const int h = j >> 1, v = j & 1;
 #pragma unroll
-            for (int r = 0; r < 4; ++r)
-                m_tile[(size_t)(wave * 16 + kgrp * 4 + r) * mstride + j]
-                    = acc[jt][r];
+        for (int r = 0; r < 4; ++r) {
+            const int bit = (vmw[r][jt >> 2] >> (h & 31)) & 1;
+            if (bit == v) tot[r] += acc[jt][r];
         }
     }
-    __syncthreads();
+    float inv[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+        inv[r] = 1.0f / fmaxf(row16_reduce(tot[r]), 1e-30f);
 
-    if (ABL == 1) {
-        if (tid < 128) h_after[k0 + tid] = m_tile[tid];
-        return;
-    }
-    for (int pi = wave; pi < 128; pi += 8) {
-        const int k = k0 + pi;
-        const unsigned* vm = vmask + (size_t)k * W;
-        const float* mrow = m_tile + (size_t)pi * mstride;
-        float val[3];          // pass-1 cache (H <= 144 -> <= 3 / lane)
-        float tot = 0.f;
-        int i = 0;
-        for (int h = lane; h < H; h += 64, ++i) {
-            const int v = ABL == 3 ? 0 : (vm[h >> 5] >> (h & 31)) & 1;
-            val[i] = mrow[2 * h + v];
-            tot += val[i];
+    const float pic = pi_hat[c];
+    float ent[4] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int jt = 0; jt < JT; ++jt) {
+        const int j = jt * 16 + row16;
+        if (j >= twoH) continue;
+        const int h = j >> 1, v = j & 1;
+        const float mix = mixture0[h];
+        const float pbb = pbest_before[(size_t)c * H + h];
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const int bit = (vmw[r][jt >> 2] >> (h & 31)) & 1;
+            if (bit == v) {
+                const float pb = acc[jt][r] * inv[r];
+                const float mm = fmaxf(mix + pic * (pb - pbb), 1e-12f);
+                ent[r] += -mm * __log2f(mm);
+            }
         }
-        tot = wave_reduce(tot);
-        const float inv = 1.0f / fmaxf(tot, 1e-30f);
-        const float pic = pi_hat[c];
-        float ent = 0.f;
-        i = 0;
-        for (int h = lane; h < H; h += 64, ++i) {
-            const float pb = val[i] * inv;
-            const float mm = fmaxf(
-                mixture0[h] + pic * (pb - pbest_before[(size_t)c * H + h]),
-                1e-12f);
-            ent += -mm * __log2f(mm);
-        }
-        ent = wave_reduce(ent);
-        if (lane == 0) h_after[k] = ent;
     }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        const float e = row16_reduce(ent[r]);
+        if (row16 == 0) h_after[kbase + r] = e;
+    }
+    (void)mstride;
 }
 
 // ---------------------------------------------------------------------
@@ -356,7 +386,7 @@ pair_gemm_entropy128_kernel(const hip_bfloat16* __restrict__ a16,
 // vs the 16-pair fused tile), and a wave-per-pair entropy kernel
 // consumes it with the same vmask epilogue.
 // ---------------------------------------------------------------------
-#define WSTRIDE 40   // 32 k-elems + 8 pad (bf16) per LDS row
+#define WSTRIDE 34   // 32 k-elems + 2 pad: 17 words, bank-clean for the mixed b128 lane groups
 
 __global__ void __launch_bounds__(BLOCK)
 pair_gemm_wide_kernel(const hip_bfloat16* __restrict__ a16,  // (K, P)
@@ -594,7 +624,7 @@ torch::Tensor pair_gemm_entropy(torch::Tensor a16, torch::Tensor egw,
         a16.data_ptr());
     const auto eb16 = reinterpret_cast<const hip_bfloat16*>(
         egw.data_ptr());
-    if (tile == 128 && 2 * H > 288) {
+    if (tile == 128 && 2 * H > 272) {
         // wide-H split pipeline: M to global, entropy second pass
         TORCH_CHECK(2 * H <= 4096, "pair engine caps at H = 2048");
         const int twoH = 2 * H;
@@ -621,11 +651,9 @@ torch::Tensor pair_gemm_entropy(torch::Tensor a16, torch::Tensor egw,
                            mixture0.data_ptr<float>(),
                            h_after.data_ptr<float>(), K, H);
     } else if (tile == 128) {
-        TORCH_CHECK(2 * H <= 288, "128-pair tile needs 2H <= 288");
-        const size_t phase1 = (size_t)2 * H * BSTRIDE
-                            * sizeof(hip_bfloat16);
-        const size_t phase2 = (size_t)128 * mstride * sizeof(float);
-        const size_t shmem = std::max(phase1, phase2);
+        TORCH_CHECK(2 * H <= 272, "128-pair tile needs 2H <= 272");
+        const size_t shmem = (size_t)2 * H * BSTRIDE
+                           * sizeof(hip_bfloat16);
         const int JT = (2 * H + 15) / 16;
         auto launch = [&](auto kern) {
             hipLaunchKernelGGL(kern, dim3(K / 128), dim3(BLOCK2), shmem,
@@ -639,18 +667,13 @@ torch::Tensor pair_gemm_entropy(torch::Tensor a16, torch::Tensor egw,
                                mixture0.data_ptr<float>(),
                                h_after.data_ptr<float>(), H, mstride);
         };
+        (void)ablate;  // phase-ablation diagnostics retired
         if (JT <= 4) launch(pairops::pair_gemm_entropy128_kernel<4>);
         else if (JT <= 8)
             launch(pairops::pair_gemm_entropy128_kernel<8>);
-        else if (JT <= 16) {
-            if (ablate == 1)
-                launch(pairops::pair_gemm_entropy128_kernel<16, 1>);
-            else if (ablate == 2)
-                launch(pairops::pair_gemm_entropy128_kernel<16, 2>);
-            else if (ablate == 3)
-                launch(pairops::pair_gemm_entropy128_kernel<16, 3>);
-            else launch(pairops::pair_gemm_entropy128_kernel<16>);
-        } else launch(pairops::pair_gemm_entropy128_kernel<18>);
+        else if (JT <= 16)
+            launch(pairops::pair_gemm_entropy128_kernel<16>);
+        else launch(pairops::pair_gemm_entropy128_kernel<17>);
     } else {
         TORCH_CHECK(tile == 16, "tile must be 16 or 64");
         const size_t shmem = 16 * P_POINTS * sizeof(hip_bfloat16)

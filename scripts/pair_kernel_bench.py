@@ -105,17 +105,6 @@ for ps, name in ((ps16, "t16"), (ps64, "t64")):
                                  adjusted, row_sums, float(H_before))
     print(name, "q[:4]", q[:4].tolist())
 
-# phase ablation at tile=128 (guide: symptom -> diagnosis loop)
-ps = pops.build_pairs(cls_rows, ids, C, tile=128)
-A16 = O._ext.pair_dsum_es(tables.delta16, tables.dall, ps.pair_c, ps.pair_neg, ps.seg_off, ps.seg_h)
-for abl, name in [(0, "full"), (1, "no-epilogue"), (2, "no-mfma"),
-                  (3, "no-cls-reads")]:
-    timeit(f"gemm128[{name}]",
-           lambda: O._ext.pair_gemm_entropy(
-               A16, tables.egw, ps.vmask, ps.pair_c,
-               pi_hat.contiguous(), pbest_before.contiguous(),
-               mixture0.contiguous(), 128, abl))
-
 # wide-H pipeline (the multi-GPU H>144 shapes): forced tile=128
 if args.h > 144:
     pass  # main loop above covered it
