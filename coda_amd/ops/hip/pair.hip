@@ -232,18 +232,17 @@ pair_gemm_entropy16_kernel(const hip_bfloat16* __restrict__ a16,
 // ---------------------------------------------------------------------
 // 128-pair B-resident tile GEMM+entropy (2H <= 272): the tile's whole
 // egw[c] table (2H x P bf16) is staged into LDS ONCE with a coalesced
-// row-major copy at a conflict-free stride (266 elems = 133 words; the
-// b128 16-lane service groups mix row16 0..3 with kgrp 0..3, so the
-// bank residues 5*r + 4*k must be distinct - a x264 stride collided
-// 4-way, 4.7 extra LDS cycles per read measured by PMC), then 8 waves
-// run the whole K loop against it barrier-free with the A operand
-// streamed from L2. The epilogue is REGISTER-RESIDENT: each lane holds
+// row-major copy (stride 264 elems - a 16-B-aligned row start is
+// required for the b128 fragment loads; an odd-word "bank-ideal"
+// stride measurably loses more to misalignment than it wins back in
+// conflicts), then 8 waves run the whole K loop against it
+// barrier-free with the A operand streamed from L2. The epilogue is REGISTER-RESIDENT: each lane holds
 // pairs kgrp*4+r at columns jt*16+row16, the static vmask selects the
 // v variant per (pair, h) (the mask word index (8*jt)>>5 is
 // compile-time), and 16-lane DPP row reductions produce tot/entropy -
 // the (128, 2H) M tile never exists in LDS or HBM.
 // ---------------------------------------------------------------------
-#define BSTRIDE 266   // bf16 elems; 133 words == 5 mod 64
+#define BSTRIDE 264   // bf16 elems (528 B rows keep 16-B alignment for b128)
 #define BLOCK2 512
 
 __device__ __forceinline__ float row16_reduce(float v) {
@@ -386,7 +385,7 @@ pair_gemm_entropy128_kernel(const hip_bfloat16* __restrict__ a16,
 // vs the 16-pair fused tile), and a wave-per-pair entropy kernel
 // consumes it with the same vmask epilogue.
 // ---------------------------------------------------------------------
-#define WSTRIDE 34   // 32 k-elems + 2 pad: 17 words, bank-clean for the mixed b128 lane groups
+#define WSTRIDE 40   // 32 k-elems + 8 pad (16-B-aligned rows)
 
 __global__ void __launch_bounds__(BLOCK)
 pair_gemm_wide_kernel(const hip_bfloat16* __restrict__ a16,  // (K, P)
