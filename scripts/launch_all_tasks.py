@@ -77,7 +77,7 @@ def main():
     p.add_argument("--pred-dir", default="data")
     p.add_argument("--methods",
                    default="iid,activetesting,vma,model_picker,"
-                           "uncertainty,coda")
+                           "uncertainty,coda-lr=0.01-mult=2.0-no-prefilter")
     p.add_argument("--seeds", type=int, default=5)
     p.add_argument("--gpus", type=int, default=None,
                    help="GPU slots to use (default: all visible, or 1 CPU "
