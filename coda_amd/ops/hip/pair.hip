@@ -552,6 +552,49 @@ pair_eig_finalize_kernel(const float* __restrict__ h_after,   // (K,)
     if (lane == 0) q[b] = H_before - tot * inv;
 }
 
+// cls-based wide entropy (H beyond the vmask regime, e.g. 10k-model
+// pools): v-select reads the candidate's argmax classes directly (the
+// B x H cls block is small for prefiltered candidate sets and
+// L2-resident); no per-lane value cache (H unbounded), so M is read
+// twice.
+__global__ void __launch_bounds__(BLOCK)
+pair_entropy_wide_cls_kernel(const hip_bfloat16* __restrict__ m,  // (K, 2H)
+                             const int* __restrict__ cls,         // (B, H)
+                             const int* __restrict__ pair_b,      // (K,)
+                             const int* __restrict__ pair_c,      // (K,)
+                             const float* __restrict__ pi_hat,
+                             const float* __restrict__ pbest_before,
+                             const float* __restrict__ mixture0,
+                             float* __restrict__ h_after,         // (K,)
+                             int K, int H) {
+    const int k = blockIdx.x * 4 + (threadIdx.x >> 6);
+    if (k >= K) return;
+    const int lane = threadIdx.x & 63;
+    const int c = pair_c[k];
+    const int b = pair_b[k];
+    const hip_bfloat16* mrow = m + (size_t)k * 2 * H;
+
+    float tot = 0.f;
+    for (int h = lane; h < H; h += 64) {
+        const int v = (b >= 0 && cls[(size_t)b * H + h] == c) ? 1 : 0;
+        tot += (float)mrow[2 * h + v];
+    }
+    tot = wave_reduce(tot);
+    const float inv = 1.0f / fmaxf(tot, 1e-30f);
+    const float pic = pi_hat[c];
+    float ent = 0.f;
+    for (int h = lane; h < H; h += 64) {
+        const int v = (b >= 0 && cls[(size_t)b * H + h] == c) ? 1 : 0;
+        const float pb = (float)mrow[2 * h + v] * inv;
+        const float mm = fmaxf(
+            mixture0[h] + pic * (pb - pbest_before[(size_t)c * H + h]),
+            1e-12f);
+        ent += -mm * __log2f(mm);
+    }
+    ent = wave_reduce(ent);
+    if (lane == 0) h_after[k] = ent;
+}
+
 // MFMA layout probe (correctness insurance, not a production op):
 // C (16,16) = A (16,32) x B stored row-major as BT (16 cols x 32 k).
 __global__ void mfma_probe_kernel(const float* __restrict__ a,   // (16,32)
@@ -719,6 +762,47 @@ torch::Tensor pair_eig_finalize(torch::Tensor h_after,
     return q;
 }
 
+torch::Tensor pair_gemm_entropy_cls(torch::Tensor a16, torch::Tensor egw,
+                                    torch::Tensor pair_b,
+                                    torch::Tensor pair_c,
+                                    torch::Tensor cls,
+                                    torch::Tensor pi_hat,
+                                    torch::Tensor pbest_before,
+                                    torch::Tensor mixture0) {
+    // wide-H pools (no v-select bitmask): 128x128 M-writing GEMM + the
+    // cls-based entropy pass; any 2H (memory-bound by the (K, 2H) M
+    // workspace)
+    TORCH_CHECK(a16.is_cuda() && a16.dtype() == torch::kBFloat16);
+    const int K = a16.size(0);
+    const int H = mixture0.size(0);
+    TORCH_CHECK(K % 128 == 0, "cls route needs 128-pair tiles");
+    const int twoH = 2 * H;
+    auto mout = torch::empty({(long)K, (long)twoH}, a16.options());
+    auto h_after = torch::empty({K}, pi_hat.options());
+    auto stream = c10::hip::getCurrentHIPStream();
+    const size_t shmem = 4 * 128 * WSTRIDE * sizeof(hip_bfloat16);
+    dim3 grid(K / 128, (twoH + 127) / 128);
+    hipLaunchKernelGGL(pairops::pair_gemm_wide_kernel, grid, dim3(BLOCK),
+                       shmem, stream.stream(),
+                       reinterpret_cast<const hip_bfloat16*>(
+                           a16.data_ptr()),
+                       reinterpret_cast<const hip_bfloat16*>(
+                           egw.data_ptr()),
+                       pair_c.data_ptr<int>(),
+                       reinterpret_cast<hip_bfloat16*>(mout.data_ptr()),
+                       twoH);
+    hipLaunchKernelGGL(pairops::pair_entropy_wide_cls_kernel,
+                       dim3((K + 3) / 4), dim3(BLOCK), 0, stream.stream(),
+                       reinterpret_cast<const hip_bfloat16*>(
+                           mout.data_ptr()),
+                       cls.data_ptr<int>(), pair_b.data_ptr<int>(),
+                       pair_c.data_ptr<int>(), pi_hat.data_ptr<float>(),
+                       pbest_before.data_ptr<float>(),
+                       mixture0.data_ptr<float>(),
+                       h_after.data_ptr<float>(), K, H);
+    return h_after;
+}
+
 torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor bt) {
     TORCH_CHECK(a.sizes() == torch::IntArrayRef({16, 32}));
     TORCH_CHECK(bt.sizes() == torch::IntArrayRef({16, 32}));
@@ -740,6 +824,8 @@ void register_pair_ops(pybind11::module_& m) {
           pybind11::arg("pi_hat"),
           pybind11::arg("pbest_before"), pybind11::arg("mixture0"),
           pybind11::arg("tile"), pybind11::arg("ablate") = 0);
+    m.def("pair_gemm_entropy_cls", &pair_gemm_entropy_cls,
+          "v3 wide-H pairing GEMM + cls-based entropy -> (K,)");
     m.def("pair_eig_finalize", &pair_eig_finalize,
           "v3 per-candidate EIG assembly (deterministic) -> (B,)");
     m.def("mfma_probe", &mfma_probe,

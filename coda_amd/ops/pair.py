@@ -93,7 +93,8 @@ class PairStructure(NamedTuple):
 
 
 def build_pairs(cls_rows: torch.Tensor, cand_ids: torch.Tensor,
-                C: int, tile: int = 0) -> PairStructure:
+                C: int, tile: int = 0,
+                with_vmask: bool = True) -> PairStructure:
     """Build the hit structure from candidate argmax classes.
 
     cls_rows: (B, H) — argmax class of every (global) model on each
@@ -152,15 +153,19 @@ def build_pairs(cls_rows: torch.Tensor, cand_ids: torch.Tensor,
     cand_off[1:] = b_counts.cumsum(0).to(torch.int32)
 
     # static v-select bitmask: bit h of vmask[k] iff model h in seg(k)
-    # (disjoint bits, so scatter-add == bitwise-or; base/pad rows stay 0)
-    W = (H + 31) // 32
-    seg_pair_pos = torch.repeat_interleave(pos, seg_len_real)
-    word = seg_pair_pos * W + (seg_h.long() >> 5)
-    bit = torch.bitwise_left_shift(
-        torch.ones_like(seg_h, dtype=torch.int32), seg_h & 31)
-    vmask = torch.zeros(K * W, dtype=torch.int32, device=device)
-    vmask.index_put_((word,), bit, accumulate=True)
-    vmask = vmask.view(K, W)
+    # (disjoint bits, so scatter-add == bitwise-or; base/pad rows stay
+    # 0). Wide pools (H > ~1024) skip it - K x H/8 bytes stops paying
+    # for itself and the cls-based entropy kernel reads classes instead.
+    vmask = None
+    if with_vmask:
+        W = (H + 31) // 32
+        seg_pair_pos = torch.repeat_interleave(pos, seg_len_real)
+        word = seg_pair_pos * W + (seg_h.long() >> 5)
+        bit = torch.bitwise_left_shift(
+            torch.ones_like(seg_h, dtype=torch.int32), seg_h & 31)
+        vmask = torch.zeros(K * W, dtype=torch.int32, device=device)
+        vmask.index_put_((word,), bit, accumulate=True)
+        vmask = vmask.view(K, W)
 
     # complement majority segments: a pair hit by > H/2 models stores
     # the NON-hitting models instead (dsum = dall[c] - sum(complement)).
@@ -286,10 +291,17 @@ def eig_pairs(tables, ps: PairStructure, cls_rows: torch.Tensor,
         A16 = O._ext.pair_dsum_es(tables.delta16, tables.dall,
                                   ps.pair_c, ps.pair_neg, ps.seg_off,
                                   ps.seg_h)                 # (K, P) bf16
-        h_after = O._ext.pair_gemm_entropy(
-            A16, tables.egw, ps.vmask, ps.pair_c,
-            pi_hat.contiguous(), pbest_before.contiguous(),
-            mixture0.contiguous(), ps.tile)                 # (K,)
+        if ps.vmask is None:
+            h_after = O._ext.pair_gemm_entropy_cls(
+                A16, tables.egw, ps.pair_b, ps.pair_c,
+                cls_rows.to(torch.int32).contiguous(),
+                pi_hat.contiguous(), pbest_before.contiguous(),
+                mixture0.contiguous())                      # (K,)
+        else:
+            h_after = O._ext.pair_gemm_entropy(
+                A16, tables.egw, ps.vmask, ps.pair_c,
+                pi_hat.contiguous(), pbest_before.contiguous(),
+                mixture0.contiguous(), ps.tile)             # (K,)
         q = O._ext.pair_eig_finalize(
             h_after, ps.pair_c, ps.base_pos.to(torch.int32),
             ps.cand_off, ps.cand_pairs, ps.cand_ids,

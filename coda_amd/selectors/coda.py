@@ -321,7 +321,10 @@ class CODA(ModelSelector):
         def build(ids_t):
             mine = ids_t[r::w] if self._replicated else ids_t
             cls_rows = self._global_classes(mine)
-            return pops.build_pairs(cls_rows, mine, self.C), cls_rows
+            wide = self.H > 1024  # beyond the vmask/fused-kernel regime
+            return pops.build_pairs(
+                cls_rows, mine, self.C, tile=(128 if wide else 0),
+                with_vmask=not wide), cls_rows
 
         if candidate_ids is self._active_candidates:
             # full-pool acquisition: the hit structure is static (argmax
@@ -384,13 +387,15 @@ class CODA(ModelSelector):
             if self._replicated:
                 impl0 = "pair"
             elif (self.device.type == "cuda"
-                    and not self.comm.is_distributed and self.H <= 1024
-                    and full_set):
+                    and not self.comm.is_distributed
+                    and (full_set or self.H > 1024)):
                 # full-pool acquisition: the static hit structure pays
                 # for itself; prefiltered SUBSETS resample every step,
-                # where the v2 table chunks (tuned round 1: 1.33
-                # ms/step at prefilter 256) stay faster than a per-step
-                # structure rebuild
+                # where at moderate H the v2 table chunks (tuned round
+                # 1: 1.33 ms/step at prefilter 256) beat a per-step
+                # structure rebuild - but at wide H (> 1024 models) the
+                # dense (B, C, P) table work loses to the hit-sparse
+                # pair pipeline even with the rebuild
                 impl0 = "pair"
         if impl0 == "pair" and (self._replicated
                                 or not self.comm.is_distributed):
@@ -520,6 +525,7 @@ class CODA(ModelSelector):
         if (self.q == "eig" and self._use_label_graph
                 and self._label_graph is not None
                 and self._pairs_static is not None
+                and self._pairs_static[0].vmask is not None
                 and not DEBUG_VIZ
                 and not (self.prefilter_n
                          and len(self._active_candidates)

@@ -594,3 +594,39 @@ class TestPairKernels:
         scale = float(eig_e.abs().max())
         err = float((eig_k - eig_e).abs().max())
         assert err < max(5e-3 * scale, 3e-4), (err, scale)
+
+    def test_pair_wide_cls_route_vs_table(self, dev):
+        """H > 1024 (the 10k-pool regime): the vmask-less cls-based pair
+        route vs the v2 table engine on the same chunk."""
+        from coda_amd.ops import pair as pops
+        from coda_amd.ops import table as tops
+        from coda_amd.ops import reference as R
+        from tests.test_pair import _random_problem
+        H, N, C = 1200, 200, 25
+        (preds, cls, dirichlets, pi_hat, adjusted,
+         row_sums) = _random_problem(H, N, C, seed=11)
+        alpha_cc, beta_cc = R.dirichlet_to_beta(dirichlets)
+        alpha_cc, beta_cc = alpha_cc.to(dev), beta_cc.to(dev)
+        tables = pops.attach_pair_tables(
+            tops.table_precompute(alpha_cc, beta_cc))
+        from coda_amd import ops as O
+        pbest_before = O.pbest_from_beta(alpha_cc.t().contiguous(),
+                                         beta_cc.t().contiguous())
+        mixture0, H_before = R.mixture_entropy(pbest_before,
+                                               pi_hat.to(dev))
+        ids = torch.arange(N, device=dev)
+        cls_rows = cls.to(dev)[:, ids].t().to(torch.int32).contiguous()
+        ps = pops.build_pairs(cls_rows, ids, C, tile=128,
+                              with_vmask=False)
+        assert ps.vmask is None
+        eig_k = pops.eig_pairs(
+            tables, ps, cls_rows, pbest_before, pi_hat.to(dev),
+            mixture0, H_before, adjusted.to(dev), row_sums.to(dev))
+        pi_xi = (adjusted / row_sums.clamp_min(1e-12).unsqueeze(-1)) \
+            .to(dev)
+        eig_t = tops.eig_chunk_table(tables, cls_rows.long(),
+                                     pbest_before, pi_hat.to(dev),
+                                     pi_xi, mixture0, H_before)
+        scale = float(eig_t.abs().max())
+        err = float((eig_k - eig_t).abs().max())
+        assert err < max(5e-3 * scale, 3e-4), (err, scale)
