@@ -387,15 +387,18 @@ class CODA(ModelSelector):
             if self._replicated:
                 impl0 = "pair"
             elif (self.device.type == "cuda"
-                    and not self.comm.is_distributed
-                    and (full_set or self.H > 1024)):
+                    and not self.comm.is_distributed and self.H <= 1024
+                    and full_set):
                 # full-pool acquisition: the static hit structure pays
                 # for itself; prefiltered SUBSETS resample every step,
-                # where at moderate H the v2 table chunks (tuned round
-                # 1: 1.33 ms/step at prefilter 256) beat a per-step
-                # structure rebuild - but at wide H (> 1024 models) the
-                # dense (B, C, P) table work loses to the hit-sparse
-                # pair pipeline even with the rebuild
+                # where the v2 table chunks (tuned round 1: 1.33
+                # ms/step at prefilter 256) stay faster than a per-step
+                # structure rebuild. Wide pools (H >> C, e.g. 10k
+                # models x 1000 classes) stay on v2 too: the hit
+                # sparsity collapses there (nearly every class is hit
+                # by SOME model, so K ~ B*C = the dense problem) and
+                # the pair pipeline's M workspace makes it strictly
+                # worse - measured 20.1 vs 4.0 ms/step at H=4096.
                 impl0 = "pair"
         if impl0 == "pair" and (self._replicated
                                 or not self.comm.is_distributed):
