@@ -85,41 +85,64 @@ class EigTables(NamedTuple):
 def table_precompute(alpha_cc: torch.Tensor, beta_cc: torch.Tensor,
                      update_weight: float = 1.0,
                      num_points: int = PBEST_NUM_POINTS) -> EigTables:
-    """Build the per-step curve tables from the (H, C) diagonal Betas."""
+    """Build the per-step curve tables from the (H, C) diagonal Betas.
+
+    Chunked over class rows: the naive form holds ~6 (C*H*2, P) fp32
+    transients at once (~120 GB at a 10k-model pool), which OOMs next
+    to a resident prediction tensor. Output tables are allocated once
+    and filled per chunk (a few GB of transients regardless of H*C).
+    """
     H, C = alpha_cc.shape
     dev = alpha_cc.device
-    # variants: v0 = (a, b+w) "predicted another class", v1 = (a+w, b)
-    a = torch.stack([alpha_cc, alpha_cc + update_weight], dim=-1)  # (H,C,2)
-    b = torch.stack([beta_cc + update_weight, beta_cc], dim=-1)
-    aR = a.permute(1, 0, 2).reshape(-1)   # (C*H*2,) in (c, h, v) order
-    bR = b.permute(1, 0, 2).reshape(-1)
+    P = num_points
 
-    x = torch.linspace(GRID_LO, GRID_HI, num_points, device=dev,
+    x = torch.linspace(GRID_LO, GRID_HI, P, device=dev,
                        dtype=torch.float64)
     lx = torch.log2(x)
     l1mx = torch.log2(1.0 - x)
-    lnB2 = (torch.lgamma(aR.double()) + torch.lgamma(bR.double())
-            - torch.lgamma((aR + bR).double())) * _LOG2E
-    # f64 log-pdf (as in the kernel's anchored form), then fp32
-    t2 = ((aR.double() - 1.0).unsqueeze(-1) * lx
-          + (bR.double() - 1.0).unsqueeze(-1) * l1mx
-          - lnB2.unsqueeze(-1)).float()                     # (R, P)
-    pdf = torch.exp2(t2)
     dx = float(x[1] - x[0])
-    csum = torch.cat([torch.zeros_like(pdf[:, :1]),
-                      torch.cumsum(0.5 * (pdf[:, 1:] + pdf[:, :-1]) * dx,
-                                   dim=-1)], dim=-1)
-    lc = torch.log2(csum.clamp_min(EPS_PROB))               # (R, P)
-    EG = torch.exp2(t2 - lc).reshape(C, H, 2, num_points)
-    lc = lc.reshape(C, H, 2, num_points)
-    delta = (lc[:, :, 1] - lc[:, :, 0]).contiguous()        # (C, H, P)
-    s_base = lc[:, :, 0].sum(dim=1)                         # (C, P)
 
-    w = torch.full((num_points,), dx, device=dev)
-    w[0] = w[-1] = 0.5 * dx
+    EG = torch.empty(C, H, 2, P, device=dev)
+    delta = torch.empty(C, H, P, device=dev)
+    s_base = torch.empty(C, P, device=dev)
     eg16 = None
     if EG.is_cuda and _use_bf16_gemm():
-        eg16 = EG.reshape(C, 2 * H, num_points).to(torch.bfloat16)
+        eg16 = torch.empty(C, 2 * H, P, device=dev, dtype=torch.bfloat16)
+
+    # rows per chunk so each (rows*H*2, P) fp32 transient stays <= ~1 GB
+    cchunk = max(1, min(C, int(1e9 // (H * 2 * P * 4))))
+    for c0 in range(0, C, cchunk):
+        c1 = min(C, c0 + cchunk)
+        # variants: v0 = (a, b+w) "predicted another class", v1 = (a+w, b)
+        a = torch.stack([alpha_cc[:, c0:c1],
+                         alpha_cc[:, c0:c1] + update_weight], dim=-1)
+        b = torch.stack([beta_cc[:, c0:c1] + update_weight,
+                         beta_cc[:, c0:c1]], dim=-1)        # (H, cc, 2)
+        aR = a.permute(1, 0, 2).reshape(-1)  # (cc*H*2,) in (c, h, v)
+        bR = b.permute(1, 0, 2).reshape(-1)
+        lnB2 = (torch.lgamma(aR.double()) + torch.lgamma(bR.double())
+                - torch.lgamma((aR + bR).double())) * _LOG2E
+        # f64 log-pdf (as in the kernel's anchored form), then fp32
+        t2 = ((aR.double() - 1.0).unsqueeze(-1) * lx
+              + (bR.double() - 1.0).unsqueeze(-1) * l1mx
+              - lnB2.unsqueeze(-1)).float()                 # (R, P)
+        pdf = torch.exp2(t2)
+        csum = torch.cat(
+            [torch.zeros_like(pdf[:, :1]),
+             torch.cumsum(0.5 * (pdf[:, 1:] + pdf[:, :-1]) * dx,
+                          dim=-1)], dim=-1)
+        lc = torch.log2(csum.clamp_min(EPS_PROB))           # (R, P)
+        cc = c1 - c0
+        EG[c0:c1] = torch.exp2(t2 - lc).reshape(cc, H, 2, P)
+        lc = lc.reshape(cc, H, 2, P)
+        delta[c0:c1] = lc[:, :, 1] - lc[:, :, 0]
+        s_base[c0:c1] = lc[:, :, 0].sum(dim=1)
+        if eg16 is not None:
+            eg16[c0:c1] = EG[c0:c1].reshape(cc, 2 * H, P) \
+                .to(torch.bfloat16)
+
+    w = torch.full((P,), dx, device=dev)
+    w[0] = w[-1] = 0.5 * dx
     return EigTables(EG, delta, s_base, w, eg16)
 
 

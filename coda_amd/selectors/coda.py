@@ -370,6 +370,19 @@ class CODA(ModelSelector):
             off += cnt
         return perm
 
+    def _hit_sparse(self) -> bool:
+        """True when candidates hit few distinct classes (the pair
+        engine's regime): estimated from a 1024-point sample once."""
+        if not hasattr(self, "_est_distinct"):
+            src = self.classes_global if self._replicated else self.classes
+            n = src.shape[1]
+            g = torch.Generator(device="cpu").manual_seed(0)
+            idx = torch.randperm(n, generator=g)[:1024].to(src.device)
+            sub = src[:, idx].long().sort(dim=0).values
+            distinct = (sub[1:] != sub[:-1]).sum(0) + 1
+            self._est_distinct = float(distinct.float().mean())
+        return self._est_distinct <= max(self.C / 4.0, 32.0)
+
     def _global_classes(self, ids: torch.Tensor) -> torch.Tensor:
         """(B, H) int32 argmax classes of every GLOBAL model on the
         given points."""
@@ -388,17 +401,16 @@ class CODA(ModelSelector):
                 impl0 = "pair"
             elif (self.device.type == "cuda"
                     and not self.comm.is_distributed and self.H <= 1024
-                    and full_set):
+                    and full_set and self._hit_sparse()):
                 # full-pool acquisition: the static hit structure pays
-                # for itself; prefiltered SUBSETS resample every step,
-                # where the v2 table chunks (tuned round 1: 1.33
-                # ms/step at prefilter 256) stay faster than a per-step
-                # structure rebuild. Wide pools (H >> C, e.g. 10k
-                # models x 1000 classes) stay on v2 too: the hit
-                # sparsity collapses there (nearly every class is hit
-                # by SOME model, so K ~ B*C = the dense problem) and
-                # the pair pipeline's M workspace makes it strictly
-                # worse - measured 20.1 vs 4.0 ms/step at H=4096.
+                # for itself when candidates hit FEW distinct classes.
+                # Prefiltered SUBSETS resample every step, where the v2
+                # table chunks (tuned round 1: 1.33 ms/step at
+                # prefilter 256) beat a per-step structure rebuild; and
+                # when H >~ C the sparsity collapses (nearly every
+                # class is hit by some model, K ~ B*C = the dense
+                # problem) so v2 stays - measured 20.1 vs 4.0 ms/step
+                # at H=4096, 15.2 vs ~5 at H=1024 full-pool.
                 impl0 = "pair"
         if impl0 == "pair" and (self._replicated
                                 or not self.comm.is_distributed):
