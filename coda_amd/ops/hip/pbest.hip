@@ -637,64 +637,91 @@ template <typename TOUT>
 __global__ void __launch_bounds__(BLOCK)
 es_build_kernel(const float* __restrict__ s_base,   // (C, P)
                 const float* __restrict__ delta,    // (C, H, P)
+                const float* __restrict__ dall,     // (C, P) sum over h
                 const int* __restrict__ hvals,      // (B, H) h sorted by class
                 const int* __restrict__ offsets,    // (B, C+1) CSR
                 const float* __restrict__ w,        // (P,)
                 TOUT* __restrict__ es,              // (C, B, P)
                 int B, int C, int H) {
-    const int r = blockIdx.x * ROWS_PER_BLOCK + (threadIdx.x >> 6);
+    // 16 lanes per (b, c) row, 16 grid points per lane. A majority
+    // class (e.g. ~75% of a 10k-model pool predicting the consensus
+    // label) used to serialize one wave on thousands of dependent
+    // row loads; rows whose bucket exceeds H/2 now sum the COMPLEMENT
+    // positions ([0,k0) u [k1,H) of the same sorted-permutation row)
+    // and subtract from the per-class total dall[c], capping the chain
+    // at H/2 and halving worst-case traffic.
+    const int r = blockIdx.x * 16 + (threadIdx.x >> 4);
     if (r >= B * C) return;
     const int b = r / C, c = r - b * C;
-    const int lane = threadIdx.x & 63;
-    const int p0 = lane * PTS_PER_LANE;
-
-    float acc[PTS_PER_LANE];
-    const float4 sb = *reinterpret_cast<const float4*>(
-        s_base + (size_t)c * P_POINTS + p0);
-    acc[0] = sb.x; acc[1] = sb.y; acc[2] = sb.z; acc[3] = sb.w;
+    const int sub = threadIdx.x & 15;
+    const int p0 = sub * 16;
     const size_t dbase = (size_t)c * H * P_POINTS + p0;
-    // CSR over "models whose argmax class on candidate b is c": avg H/C
-    // iterations instead of an H-long wave-uniform load chain. Four
-    // accumulator sets break the fp32 add dependency (the compiler may
-    // not reassociate), so four loads are in flight per wave.
+    const int* hrow = hvals + (size_t)b * H;
     const int k0 = offsets[(size_t)b * (C + 1) + c];
     const int k1 = offsets[(size_t)b * (C + 1) + c + 1];
-    float a1[4] = {0.f, 0.f, 0.f, 0.f}, a2[4] = {0.f, 0.f, 0.f, 0.f},
-          a3[4] = {0.f, 0.f, 0.f, 0.f};
-    int k = k0;
-    for (; k + 3 < k1; k += 4) {
-        const int ha = hvals[(size_t)b * H + k];
-        const int hb = hvals[(size_t)b * H + k + 1];
-        const int hc2 = hvals[(size_t)b * H + k + 2];
-        const int hd = hvals[(size_t)b * H + k + 3];
-        const float4 da = *reinterpret_cast<const float4*>(
-            delta + dbase + (size_t)ha * P_POINTS);
-        const float4 db = *reinterpret_cast<const float4*>(
-            delta + dbase + (size_t)hb * P_POINTS);
-        const float4 dc = *reinterpret_cast<const float4*>(
-            delta + dbase + (size_t)hc2 * P_POINTS);
-        const float4 dd = *reinterpret_cast<const float4*>(
-            delta + dbase + (size_t)hd * P_POINTS);
-        acc[0] += da.x; acc[1] += da.y; acc[2] += da.z; acc[3] += da.w;
-        a1[0] += db.x; a1[1] += db.y; a1[2] += db.z; a1[3] += db.w;
-        a2[0] += dc.x; a2[1] += dc.y; a2[2] += dc.z; a2[3] += dc.w;
-        a3[0] += dd.x; a3[1] += dd.y; a3[2] += dd.z; a3[3] += dd.w;
-    }
-    for (; k < k1; ++k) {
-        const int h = hvals[(size_t)b * H + k];
-        const float4 d = *reinterpret_cast<const float4*>(
+    const bool comp = (k1 - k0) > H / 2;
+
+    float a0[16] = {}, a1[16] = {};
+    auto addrow = [&](int h, float* acc) {
+        const float4* d = reinterpret_cast<const float4*>(
             delta + dbase + (size_t)h * P_POINTS);
-        acc[0] += d.x; acc[1] += d.y; acc[2] += d.z; acc[3] += d.w;
-    }
 #pragma unroll
-    for (int j = 0; j < PTS_PER_LANE; ++j)
-        acc[j] += (a1[j] + a2[j]) + a3[j];
-    const float4 wv = *reinterpret_cast<const float4*>(w + p0);
+        for (int q = 0; q < 4; ++q) {
+            const float4 v = d[q];
+            acc[4 * q] += v.x; acc[4 * q + 1] += v.y;
+            acc[4 * q + 2] += v.z; acc[4 * q + 3] += v.w;
+        }
+    };
+    if (!comp) {
+        int k = k0;
+        for (; k + 1 < k1; k += 2) {          // 2 rows in flight
+            addrow(hrow[k], a0);
+            addrow(hrow[k + 1], a1);
+        }
+        if (k < k1) addrow(hrow[k], a0);
+    } else {
+        int k = 0;
+        for (; k + 1 < k0; k += 2) {
+            addrow(hrow[k], a0);
+            addrow(hrow[k + 1], a1);
+        }
+        if (k < k0) addrow(hrow[k], a0);
+        k = k1;
+        for (; k + 1 < H; k += 2) {
+            addrow(hrow[k], a0);
+            addrow(hrow[k + 1], a1);
+        }
+        if (k < H) addrow(hrow[k], a0);
+    }
+
+    const float4* sb = reinterpret_cast<const float4*>(
+        s_base + (size_t)c * P_POINTS + p0);
+    const float4* da = comp ? reinterpret_cast<const float4*>(
+        dall + (size_t)c * P_POINTS + p0) : nullptr;
+    const float4* wv = reinterpret_cast<const float4*>(w + p0);
+    float out[16];
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+        const float4 s = sb[q];
+        const float4 ww = wv[q];
+        float sx[4] = {s.x, s.y, s.z, s.w};
+        float wx[4] = {ww.x, ww.y, ww.z, ww.w};
+        float dx[4] = {0.f, 0.f, 0.f, 0.f};
+        if (comp) {
+            const float4 dv = da[q];
+            dx[0] = dv.x; dx[1] = dv.y; dx[2] = dv.z; dx[3] = dv.w;
+        }
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            const int i = 4 * q + j;
+            const float dsum = comp ? dx[j] - (a0[i] + a1[i])
+                                    : a0[i] + a1[i];
+            out[i] = exp2f(sx[j] + dsum) * wx[j];
+        }
+    }
     TOUT* dst = es + ((size_t)c * B + b) * P_POINTS + p0;
-    dst[0] = (TOUT)(exp2f(acc[0]) * wv.x);
-    dst[1] = (TOUT)(exp2f(acc[1]) * wv.y);
-    dst[2] = (TOUT)(exp2f(acc[2]) * wv.z);
-    dst[3] = (TOUT)(exp2f(acc[3]) * wv.w);
+#pragma unroll
+    for (int i = 0; i < 16; ++i) dst[i] = (TOUT)out[i];
 }
 
 template <typename TM>
@@ -1263,10 +1290,12 @@ std::vector<torch::Tensor> eig_phase2(torch::Tensor alpha_cc,
 // ---- Table-path (v2) fusion host bindings ----
 
 torch::Tensor es_build(torch::Tensor s_base, torch::Tensor delta,
+                       torch::Tensor dall,
                        torch::Tensor hvals, torch::Tensor offsets,
                        torch::Tensor w, bool bf16_out) {
     check_f32_cuda(s_base, "s_base");
     check_f32_cuda(delta, "delta");
+    check_f32_cuda(dall, "dall");
     check_f32_cuda(w, "w");
     TORCH_CHECK(hvals.scalar_type() == torch::kInt32 &&
                 offsets.scalar_type() == torch::kInt32,
@@ -1279,13 +1308,15 @@ torch::Tensor es_build(torch::Tensor s_base, torch::Tensor delta,
                                bf16_out ? torch::kBFloat16
                                         : torch::kFloat32));
     const int R = B * C;
-    const int blocks = (R + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
+    const int blocks = (R + 15) / 16;  // 16 sublane rows per block
     auto stream = c10::hip::getCurrentHIPStream();
     if (bf16_out) {
         hipLaunchKernelGGL(es_build_kernel<hip_bfloat16>, dim3(blocks),
                            dim3(BLOCK), 0, stream.stream(),
                            s_base.data_ptr<float>(),
-                           delta.data_ptr<float>(), hvals.data_ptr<int>(),
+                           delta.data_ptr<float>(),
+                           dall.data_ptr<float>(),
+                           hvals.data_ptr<int>(),
                            offsets.data_ptr<int>(), w.data_ptr<float>(),
                            reinterpret_cast<hip_bfloat16*>(es.data_ptr()),
                            B, C, H);
@@ -1293,7 +1324,9 @@ torch::Tensor es_build(torch::Tensor s_base, torch::Tensor delta,
         hipLaunchKernelGGL(es_build_kernel<float>, dim3(blocks),
                            dim3(BLOCK), 0, stream.stream(),
                            s_base.data_ptr<float>(),
-                           delta.data_ptr<float>(), hvals.data_ptr<int>(),
+                           delta.data_ptr<float>(),
+                           dall.data_ptr<float>(),
+                           hvals.data_ptr<int>(),
                            offsets.data_ptr<int>(), w.data_ptr<float>(),
                            es.data_ptr<float>(), B, C, H);
     }

@@ -329,9 +329,11 @@ def attach_pair_tables(tables):
     esb = torch.exp2(s_base) * w                            # (C, P)
     egw = (EG.reshape(C, 2 * H, P)
            * esb.unsqueeze(1)).to(torch.bfloat16).contiguous()
+    dall = tables.dall if tables.dall is not None \
+        else tables.delta.sum(1).contiguous()
     return tables._replace(egw=egw,
                            delta16=tables.delta.to(torch.float16),
-                           dall=tables.delta.sum(1).contiguous())
+                           dall=dall)
 
 
 def update_egw_rows(tables, rows) -> None:

@@ -105,6 +105,7 @@ def table_precompute(alpha_cc: torch.Tensor, beta_cc: torch.Tensor,
     EG = torch.empty(C, H, 2, P, device=dev)
     delta = torch.empty(C, H, P, device=dev)
     s_base = torch.empty(C, P, device=dev)
+    dall = torch.empty(C, P, device=dev) if EG.is_cuda else None
     eg16 = None
     if EG.is_cuda and _use_bf16_gemm():
         eg16 = torch.empty(C, 2 * H, P, device=dev, dtype=torch.bfloat16)
@@ -137,13 +138,15 @@ def table_precompute(alpha_cc: torch.Tensor, beta_cc: torch.Tensor,
         lc = lc.reshape(cc, H, 2, P)
         delta[c0:c1] = lc[:, :, 1] - lc[:, :, 0]
         s_base[c0:c1] = lc[:, :, 0].sum(dim=1)
+        if dall is not None:
+            dall[c0:c1] = delta[c0:c1].sum(dim=1)
         if eg16 is not None:
             eg16[c0:c1] = EG[c0:c1].reshape(cc, 2 * H, P) \
                 .to(torch.bfloat16)
 
     w = torch.full((P,), dx, device=dev)
     w[0] = w[-1] = 0.5 * dx
-    return EigTables(EG, delta, s_base, w, eg16)
+    return EigTables(EG, delta, s_base, w, eg16, dall=dall)
 
 
 def pbest_hyp_table(tables: EigTables,
@@ -195,7 +198,8 @@ def eig_chunk_table(tables: EigTables, chunk_classes: torch.Tensor,
         cls32 = chunk_classes.to(torch.int32).contiguous()
         hvals, offsets = _class_csr(chunk_classes.long(), C)
         bf16 = tables.eg16 is not None
-        ES = O._ext.es_build(s_base, delta, hvals, offsets, w, bf16)
+        ES = O._ext.es_build(s_base, delta, tables.dall, hvals,
+                             offsets, w, bf16)
         if bf16:
             M = torch.bmm(ES, tables.eg16.transpose(1, 2))
         else:
@@ -235,6 +239,8 @@ def table_update_rows(tables: EigTables, alpha_cc: torch.Tensor,
                 tables.eg16[c] = eg.reshape(2 * H, -1).to(torch.bfloat16)
             tables.delta[c] = lc[:, 1] - lc[:, 0]
             tables.s_base[c] = lc[:, 0].sum(0)
+            if tables.dall is not None:
+                tables.dall[c] = tables.delta[c].sum(0)
         if tables.egw is not None:
             from .pair import update_egw_rows
             update_egw_rows(tables, rows.tolist())
@@ -248,6 +254,8 @@ def table_update_rows(tables: EigTables, alpha_cc: torch.Tensor,
             rows.numel(), 2 * H, -1).to(torch.bfloat16)
     tables.delta[rows] = sub.delta
     tables.s_base[rows] = sub.s_base
+    if tables.dall is not None:
+        tables.dall[rows] = tables.delta[rows].sum(1)
     if tables.egw is not None:
         from .pair import update_egw_rows
         update_egw_rows(tables, rows.tolist())

@@ -635,6 +635,9 @@ class CODA(ModelSelector):
                     .to(torch.bfloat16))
             t.delta.index_copy_(0, y_t, (lc[:, 1] - lc[:, 0]).unsqueeze(0))
             t.s_base.index_copy_(0, y_t, lc[:, 0].sum(0).unsqueeze(0))
+            if t.dall is not None:
+                t.dall.index_copy_(
+                    0, y_t, t.delta.index_select(0, y_t).sum(1))
             if t.egw is not None:
                 esb = torch.exp2(t.s_base.index_select(0, y_t)) \
                     * t.weights                             # (1, P)
@@ -644,8 +647,6 @@ class CODA(ModelSelector):
                 t.delta16.index_copy_(
                     0, y_t, t.delta.index_select(0, y_t)
                     .to(torch.float16))
-                t.dall.index_copy_(
-                    0, y_t, t.delta.index_select(0, y_t).sum(1))
         # posterior rows for the next acquisition / get_pbest
         alpha_cc, beta_cc = ops.dirichlet_to_beta(self.dirichlets)
         rows = ops.pbest_from_beta(alpha_cc.t().contiguous(),
