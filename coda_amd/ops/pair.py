@@ -73,10 +73,12 @@ class PairStructure(NamedTuple):
     seg_off: torch.Tensor    # (K+1,) int32 — segment offsets into seg_h
     seg_h: torch.Tensor      # (S,) int32 — models with argmax c on point b
     base_pos: torch.Tensor   # (C,) long — position of class c's base pair
-    n_real: int              # real (non-base, non-pad) pair count
+    n_real: int              # UNIQUE evaluated pair count (post-dedupe)
     tile: int                # pairs per kernel tile (16 or 128)
     cand_off: torch.Tensor = None    # (B+1,) int32 — per-candidate CSR
-    cand_pairs: torch.Tensor = None  # (n_real,) int32 — pair ids by cand
+    # (total hits,) int32 — representative pair position for every
+    # original (candidate, class) hit, grouped by candidate
+    cand_pairs: torch.Tensor = None
     # (K, ceil(H/32)) int32 — bit h set iff model h hits pair k (the
     # static v-select; reading cls rows per pair costs 1.6 ms/step at
     # the headline shape, the bitmask 0.05 ms)
@@ -94,13 +96,22 @@ class PairStructure(NamedTuple):
 
 def build_pairs(cls_rows: torch.Tensor, cand_ids: torch.Tensor,
                 C: int, tile: int = 0,
-                with_vmask: bool = True) -> PairStructure:
+                with_vmask: bool = True,
+                dedupe: bool = True) -> PairStructure:
     """Build the hit structure from candidate argmax classes.
 
     cls_rows: (B, H) — argmax class of every (global) model on each
     candidate; cand_ids: (B,) point ids. All work is batched torch ops
     (one sort over B*H keys), so the one-off cost at N=50k, H=128 is a
     few ms on device. tile overrides the MFMA tile height (benchmarks).
+
+    dedupe: h_after depends only on (class, hit SET), never on the
+    candidate - and hit sets repeat massively (singletons are bounded
+    by C*H across the whole pool). Duplicate sets collapse to one
+    evaluated pair; the per-candidate CSR maps every original (b, c)
+    hit to its representative. ~5-10x fewer evaluated pairs at the
+    headline shape. Set identity via two independent 64-bit random
+    per-model sums (collision odds ~K^2/2^128) plus (class, set size).
     """
     device = cls_rows.device
     B, H = cls_rows.shape
@@ -118,86 +129,121 @@ def build_pairs(cls_rows: torch.Tensor, cand_ids: torch.Tensor,
     pr_c = pair_key // B
     pr_b = pair_key - pr_c * B
     K_real = int(pair_key.shape[0])
-    pid = is_new.long().cumsum(0) - 1
+    pid = is_new.long().cumsum(0) - 1              # entry -> real pair
     seg_len_real = torch.bincount(pid, minlength=K_real)   # (K_real,)
-    seg_h = (order % H).to(torch.int32)            # grouped by pair
+    entry_h = order % H                            # (B*H,) grouped by pair
 
-    # class runs: base pair + real pairs, padded to the tile height
-    class_counts = torch.bincount(pr_c, minlength=C)        # (C,)
-    run_len = ((class_counts + 1 + tile - 1) // tile) * tile  # (C,)
+    # ---- dedupe identical (class, hit-set) pairs -------------------
+    if dedupe:
+        g = torch.Generator(device="cpu").manual_seed(0x5eed)
+        r1 = torch.randint(-2**63, 2**63 - 1, (H,), generator=g,
+                           dtype=torch.int64).to(device)
+        r2 = torch.randint(-2**63, 2**63 - 1, (H,), generator=g,
+                           dtype=torch.int64).to(device)
+        hash1 = torch.zeros(K_real, dtype=torch.int64, device=device)
+        hash2 = torch.zeros(K_real, dtype=torch.int64, device=device)
+        hash1.index_add_(0, pid, r1[entry_h])      # wrapping set-sum
+        hash2.index_add_(0, pid, r2[entry_h])
+        ukey = torch.stack([pr_c, seg_len_real, hash1, hash2], dim=1)
+        _, uid_of_real = torch.unique(ukey, dim=0, return_inverse=True)
+        U = int(uid_of_real.max()) + 1 if K_real else 0
+        rep_of_u = torch.full((U,), K_real, dtype=torch.int64,
+                              device=device)
+        rep_of_u.scatter_reduce_(0, uid_of_real,
+                                 torch.arange(K_real, device=device),
+                                 reduce="amin", include_self=True)
+        reps_idx = torch.sort(rep_of_u).values     # ascending = (c,b) order
+        new_of_u = torch.searchsorted(reps_idx, rep_of_u)  # uid -> new id
+        rep_mask = torch.zeros(K_real, dtype=torch.bool, device=device)
+        rep_mask[reps_idx] = True
+    else:
+        reps_idx = torch.arange(K_real, device=device)
+        uid_of_real = reps_idx
+        new_of_u = reps_idx
+        rep_mask = torch.ones(K_real, dtype=torch.bool, device=device)
+
+    pr_b2 = pr_b[reps_idx]
+    pr_c2 = pr_c[reps_idx]
+    seg_len2 = seg_len_real[reps_idx]
+    K2 = int(reps_idx.shape[0])
+
+    # class runs over the reduced set: base pair + reps, tile-padded
+    class_counts = torch.bincount(pr_c2, minlength=C)       # (C,)
+    run_len = ((class_counts + 1 + tile - 1) // tile) * tile
     run_off = torch.zeros(C + 1, dtype=torch.long, device=device)
     run_off[1:] = run_len.cumsum(0)
     K = int(run_off[-1])
 
-    class_off_real = torch.zeros(C, dtype=torch.long, device=device)
-    class_off_real[1:] = class_counts.cumsum(0)[:-1]
-    # position of real pair i: its class run start + 1 (base) + rank
-    rank = torch.arange(K_real, device=device) - class_off_real[pr_c]
-    pos = run_off[:-1][pr_c] + 1 + rank                     # (K_real,)
+    class_off = torch.zeros(C, dtype=torch.long, device=device)
+    class_off[1:] = class_counts.cumsum(0)[:-1]
+    rank = torch.arange(K2, device=device) - class_off[pr_c2]
+    pos = run_off[:-1][pr_c2] + 1 + rank                    # (K2,)
 
     pair_b = torch.full((K,), -1, dtype=torch.int32, device=device)
-    pair_b[pos] = pr_b.to(torch.int32)
+    pair_b[pos] = pr_b2.to(torch.int32)
     pair_c = torch.repeat_interleave(
         torch.arange(C, device=device), run_len).to(torch.int32)
-    seg_len = torch.zeros(K, dtype=torch.long, device=device)
-    seg_len[pos] = seg_len_real
-    seg_off = torch.zeros(K + 1, dtype=torch.int32, device=device)
-    seg_off[1:] = seg_len.cumsum(0).to(torch.int32)
 
-    # per-candidate CSR over real pairs (for the deterministic finalize
-    # kernel): pair positions grouped by candidate, class-ascending
+    # per-candidate CSR over ALL original (b, c) hits, mapped to their
+    # representative's padded position (the finalize kernel's view)
+    pos_of_real = pos[new_of_u[uid_of_real]]                # (K_real,)
     order_b = torch.argsort(pr_b, stable=True)
-    cand_pairs = pos[order_b].to(torch.int32)
+    cand_pairs = pos_of_real[order_b].to(torch.int32)
     b_counts = torch.bincount(pr_b, minlength=B)
     cand_off = torch.zeros(B + 1, dtype=torch.int32, device=device)
     cand_off[1:] = b_counts.cumsum(0).to(torch.int32)
 
-    # static v-select bitmask: bit h of vmask[k] iff model h in seg(k)
-    # (disjoint bits, so scatter-add == bitwise-or; base/pad rows stay
-    # 0). Wide pools (H > ~1024) skip it - K x H/8 bytes stops paying
-    # for itself and the cls-based entropy kernel reads classes instead.
+    # entry-level views (original, pre-dedupe)
+    entry_pid = pid
+    flip_real = seg_len_real > (H // 2)
+
+    # static v-select bitmask over REP pairs: bit h iff model h in the
+    # TRUE hit set (before any complement rewrite). Wide pools skip it.
     vmask = None
     if with_vmask:
         W = (H + 31) // 32
-        seg_pair_pos = torch.repeat_interleave(pos, seg_len_real)
-        word = seg_pair_pos * W + (seg_h.long() >> 5)
+        keep_v = rep_mask[entry_pid]
+        tpos = pos[new_of_u[uid_of_real[entry_pid[keep_v]]]]
+        hh_v = entry_h[keep_v]
+        word = tpos * W + (hh_v >> 5)
         bit = torch.bitwise_left_shift(
-            torch.ones_like(seg_h, dtype=torch.int32), seg_h & 31)
+            torch.ones_like(hh_v, dtype=torch.int32),
+            (hh_v & 31).to(torch.int32))
         vmask = torch.zeros(K * W, dtype=torch.int32, device=device)
         vmask.index_put_((word,), bit, accumulate=True)
         vmask = vmask.view(K, W)
 
-    # complement majority segments: a pair hit by > H/2 models stores
-    # the NON-hitting models instead (dsum = dall[c] - sum(complement)).
-    # At most one such pair per candidate; its complement is exactly the
-    # candidate's other entries.
-    flip = seg_len_real > (H // 2)                          # (K_real,)
+    # kernel segments: non-flipped reps store their hit models;
+    # flipped (majority) reps store the COMPLEMENT - the rep
+    # candidate's entries in OTHER classes - and dsum subtracts from
+    # the per-class total dall[c]
     pair_neg = torch.zeros(K, dtype=torch.int32, device=device)
-    if bool(flip.any()):
-        pair_neg[pos[flip]] = 1
-        entry_pid = pid                                     # (B*H,)
-        entry_b = pr_b[entry_pid]
-        entry_h = (order % H)
+    mask1 = rep_mask[entry_pid] & ~flip_real[entry_pid]
+    t1 = pos[new_of_u[uid_of_real[entry_pid[mask1]]]]
+    h1 = entry_h[mask1]
+    if bool(flip_real[reps_idx].any()):
+        flipped_reps = reps_idx[flip_real[reps_idx]]        # orig pids
+        pair_neg[pos[new_of_u[uid_of_real[flipped_reps]]]] = 1
         q_of_b = torch.full((B,), -1, dtype=torch.long, device=device)
-        q_of_b[pr_b[flip]] = torch.nonzero(flip, as_tuple=True)[0]
-        keep = ~flip[entry_pid]
-        t1 = pos[entry_pid[keep]]
-        h1 = entry_h[keep]
+        q_of_b[pr_b[flipped_reps]] = flipped_reps
+        entry_b = pr_b[entry_pid]
         qb = q_of_b[entry_b]
         comp = (qb >= 0) & (entry_pid != qb)
-        t2 = pos[qb[comp]]
+        t2 = pos[new_of_u[uid_of_real[qb[comp]]]]
         h2 = entry_h[comp]
         tgt = torch.cat([t1, t2])
         hh = torch.cat([h1, h2])
-        srt = torch.argsort(tgt * H + hh)                   # (tgt, h) order
-        seg_h = hh[srt].to(torch.int32)
-        seg_len2 = torch.bincount(tgt, minlength=K)
-        seg_off = torch.zeros(K + 1, dtype=torch.int32, device=device)
-        seg_off[1:] = seg_len2.cumsum(0).to(torch.int32)
+    else:
+        tgt, hh = t1, h1
+    srt = torch.argsort(tgt * H + hh)                       # (tgt, h)
+    seg_h = hh[srt].to(torch.int32)
+    seg_len_k = torch.bincount(tgt, minlength=K)
+    seg_off = torch.zeros(K + 1, dtype=torch.int32, device=device)
+    seg_off[1:] = seg_len_k.cumsum(0).to(torch.int32)
 
     return PairStructure(cand_ids=cand_ids.long(), pair_b=pair_b,
                          pair_c=pair_c, seg_off=seg_off, seg_h=seg_h,
-                         base_pos=run_off[:-1].clone(), n_real=K_real,
+                         base_pos=run_off[:-1].clone(), n_real=K2,
                          tile=tile, cand_off=cand_off,
                          cand_pairs=cand_pairs, vmask=vmask,
                          pair_neg=pair_neg)
@@ -252,21 +298,21 @@ def pair_h_after(tables, ps: PairStructure, cls_rows: torch.Tensor,
 def eig_from_pairs(h_after: torch.Tensor, ps: PairStructure,
                    adjusted: torch.Tensor, row_sums: torch.Tensor,
                    H_before) -> torch.Tensor:
-    """(N,) EIG over all points from per-pair entropies.
-
-    Points outside the candidate set get the baseline-only value (never
-    read - the caller gathers cand_ids).
-    """
+    """(N,) EIG over all points from per-(unique-)pair entropies, via
+    the per-candidate CSR (each original hit maps to its
+    representative pair). Points outside the candidate set get the
+    baseline-only value (never read - the caller gathers cand_ids)."""
     N = adjusted.shape[0]
     h_base = h_after[ps.base_pos]                           # (C,)
     inv_rs = 1.0 / row_sums.clamp_min(1e-12)
     base_n = (adjusted @ h_base) * inv_rs                   # (N,)
-    valid = ps.pair_b >= 0
-    ids = ps.cand_ids[ps.pair_b[valid].long()]              # (Kr,)
-    c_v = ps.pair_c[valid].long()
+    counts = (ps.cand_off[1:] - ps.cand_off[:-1]).long()
+    ids = torch.repeat_interleave(ps.cand_ids, counts)      # (K_real,)
+    k = ps.cand_pairs.long()
+    c_v = ps.pair_c[k].long()
     pix = adjusted[ids, c_v] * inv_rs[ids]
     corr = torch.zeros(N, device=adjusted.device)
-    corr.index_add_(0, ids, pix * (h_after[valid] - h_base[c_v]))
+    corr.index_add_(0, ids, pix * (h_after[k] - h_base[c_v]))
     if torch.is_tensor(H_before):
         H_before = H_before.to(base_n.dtype)
     return H_before - base_n - corr

@@ -34,34 +34,43 @@ def _random_problem(H=10, N=120, C=7, seed=0):
 
 class TestPairStructure:
     def test_covers_every_hit_exactly_once(self):
+        """Every (candidate, class) hit maps through the CSR to exactly
+        one representative pair whose (class, hit-set) matches the
+        brute-force set (pairs with identical sets are deduplicated)."""
         _, cls, *_ = _random_problem(H=9, N=50, C=6, seed=1)
         ids = torch.arange(37)
         cls_rows = cls[:, ids].t().contiguous()
         ps = pops.build_pairs(cls_rows, ids, 6)
-        # reconstruct (b, c, models) and compare with brute force
-        seen = {}
-        for k in range(ps.K):
-            b = int(ps.pair_b[k])
-            if b < 0:
-                assert int(ps.seg_off[k + 1]) == int(ps.seg_off[k])
-                assert int(ps.pair_neg[k]) == 0
-                continue
-            c = int(ps.pair_c[k])
-            hs = sorted(ps.seg_h[int(ps.seg_off[k]):
-                                 int(ps.seg_off[k + 1])].tolist())
+
+        def pair_set(k):
+            hs = set(ps.seg_h[int(ps.seg_off[k]):
+                              int(ps.seg_off[k + 1])].tolist())
             if int(ps.pair_neg[k]):
-                # majority pair stores the complement of its hit set
-                hs = sorted(set(range(9)) - set(hs))
-            assert (b, c) not in seen
-            seen[(b, c)] = hs
+                hs = set(range(9)) - hs
+            return hs
+
+        seen = {}
+        for b in range(37):
+            for s in range(int(ps.cand_off[b]), int(ps.cand_off[b + 1])):
+                k = int(ps.cand_pairs[s])
+                c = int(ps.pair_c[k])
+                assert (b, c) not in seen, "duplicate CSR entry"
+                seen[(b, c)] = pair_set(k)
         for b in range(37):
             for c in range(6):
-                expect = [h for h in range(9) if int(cls_rows[b, h]) == c]
+                expect = {h for h in range(9) if int(cls_rows[b, h]) == c}
                 if expect:
-                    assert seen[(b, c)] == expect
+                    assert seen[(b, c)] == expect, (b, c)
                 else:
                     assert (b, c) not in seen
-        assert len(seen) == ps.n_real
+        # dedupe really collapses: fewer evaluated pairs than hits
+        n_hits = len(seen)
+        assert ps.n_real <= n_hits
+        # base/pad pairs have empty segments and no complement flag
+        for k in range(ps.K):
+            if int(ps.pair_b[k]) < 0:
+                assert int(ps.seg_off[k + 1]) == int(ps.seg_off[k])
+                assert int(ps.pair_neg[k]) == 0
 
     def test_tiles_are_class_uniform_and_base_marked(self):
         _, cls, *_ = _random_problem(H=12, N=64, C=5, seed=2)
