@@ -522,10 +522,10 @@ pair_entropy_wide_kernel(const hip_bfloat16* __restrict__ m,  // (K, 2H)
 // ---------------------------------------------------------------------
 __global__ void __launch_bounds__(BLOCK)
 pair_eig_finalize_kernel(const float* __restrict__ h_after,   // (K,)
+                         const float* __restrict__ h_base,    // (C,)
                          const int* __restrict__ pair_c,      // (K,)
-                         const int* __restrict__ base_pos,    // (C,)
                          const int* __restrict__ cand_off,    // (B+1,)
-                         const int* __restrict__ cand_pairs,  // (n_real,)
+                         const int* __restrict__ cand_pairs,  // (hits,)
                          const long* __restrict__ cand_ids,   // (B,)
                          const float* __restrict__ adjusted,  // (N, C)
                          const float* __restrict__ row_sums,  // (N,)
@@ -539,14 +539,19 @@ pair_eig_finalize_kernel(const float* __restrict__ h_after,   // (K,)
     const float inv = 1.0f / fmaxf(row_sums[id], 1e-12f);
     const float* arow = adjusted + (size_t)id * C;
     float base = 0.f;
-    for (int cc = lane; cc < C; cc += 64)
-        base += arow[cc] * h_after[base_pos[cc]];
+    int cc = lane * 4;
+    for (; cc + 3 < C; cc += 256) {   // float4 over the class axis
+        const float4 a = *reinterpret_cast<const float4*>(arow + cc);
+        const float4 hb = *reinterpret_cast<const float4*>(h_base + cc);
+        base += a.x * hb.x + a.y * hb.y + a.z * hb.z + a.w * hb.w;
+    }
+    for (; cc < C; ++cc) base += arow[cc] * h_base[cc];
     float corr = 0.f;
     const int s1 = cand_off[b + 1];
     for (int s = cand_off[b] + lane; s < s1; s += 64) {
         const int k = cand_pairs[s];
-        const int cc = pair_c[k];
-        corr += arow[cc] * (h_after[k] - h_after[base_pos[cc]]);
+        const int c2 = pair_c[k];
+        corr += arow[c2] * (h_after[k] - h_base[c2]);
     }
     const float tot = wave_reduce(base + corr);
     if (lane == 0) q[b] = H_before - tot * inv;
@@ -738,8 +743,8 @@ torch::Tensor pair_gemm_entropy(torch::Tensor a16, torch::Tensor egw,
 }
 
 torch::Tensor pair_eig_finalize(torch::Tensor h_after,
+                                torch::Tensor h_base,
                                 torch::Tensor pair_c,
-                                torch::Tensor base_pos,
                                 torch::Tensor cand_off,
                                 torch::Tensor cand_pairs,
                                 torch::Tensor cand_ids,
@@ -748,12 +753,14 @@ torch::Tensor pair_eig_finalize(torch::Tensor h_after,
                                 double H_before) {
     const int B = cand_ids.size(0);
     const int C = adjusted.size(1);
+    TORCH_CHECK(h_base.is_contiguous() && h_base.size(0) == C);
     auto q = torch::empty({B}, adjusted.options());
     auto stream = c10::hip::getCurrentHIPStream();
     hipLaunchKernelGGL(pairops::pair_eig_finalize_kernel,
                        dim3((B + 3) / 4), dim3(BLOCK), 0, stream.stream(),
-                       h_after.data_ptr<float>(), pair_c.data_ptr<int>(),
-                       base_pos.data_ptr<int>(), cand_off.data_ptr<int>(),
+                       h_after.data_ptr<float>(),
+                       h_base.data_ptr<float>(), pair_c.data_ptr<int>(),
+                       cand_off.data_ptr<int>(),
                        cand_pairs.data_ptr<int>(),
                        cand_ids.data_ptr<long>(),
                        adjusted.data_ptr<float>(),

@@ -348,8 +348,9 @@ def eig_pairs(tables, ps: PairStructure, cls_rows: torch.Tensor,
                 A16, tables.egw, ps.vmask, ps.pair_c,
                 pi_hat.contiguous(), pbest_before.contiguous(),
                 mixture0.contiguous(), ps.tile)             # (K,)
+        h_base = h_after.index_select(0, ps.base_pos).contiguous()
         q = O._ext.pair_eig_finalize(
-            h_after, ps.pair_c, ps.base_pos.to(torch.int32),
+            h_after, h_base, ps.pair_c,
             ps.cand_off, ps.cand_pairs, ps.cand_ids,
             adjusted.contiguous(), row_sums.contiguous(),
             float(H_before))                                # (B,)

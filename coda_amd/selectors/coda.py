@@ -484,8 +484,9 @@ class CODA(ModelSelector):
             mixture0.contiguous(), ps.tile)
         # H_before enters as an in-graph tensor op (the kernel's scalar
         # argument would be frozen at capture value)
+        h_base = h_after.index_select(0, ps.base_pos).contiguous()
         q0 = ops._ext.pair_eig_finalize(
-            h_after, ps.pair_c, self._ps_base32, ps.cand_off,
+            h_after, h_base, ps.pair_c, ps.cand_off,
             ps.cand_pairs, ps.cand_ids, self._adjusted, self._row_sums,
             0.0)
         q = H0 + q0
@@ -516,7 +517,6 @@ class CODA(ModelSelector):
     def _graphed_acquire(self):
         if self._acq_graph is None:
             ps, _ = self._pairs_static
-            self._ps_base32 = ps.base_pos.to(torch.int32)
             self._pairs_ids_host = ps.cand_ids.cpu().tolist()
             self._acq_out = torch.zeros(3, dtype=torch.float64,
                                         device=self.device)
