@@ -79,8 +79,8 @@ def synth_preds(model_idxs, N, C, device, seed_base=1234,
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=200)
-    ap.add_argument("--warmup", type=int, default=25)
+    ap.add_argument("--steps", type=int, default=800)
+    ap.add_argument("--warmup", type=int, default=50)
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
