@@ -1013,19 +1013,33 @@ pi_marginal_kernel(const float* __restrict__ adjusted,  // (N, C)
         // (the (N, C) read is this kernel's whole cost at large N)
         const int c4 = tid * 4;
         if (c4 < C) {
+            // two accumulator sets x 4-row unroll: 4 row-loads in
+            // flight per lane and no serial add chain (measured 173 us
+            // at 2-row/1-acc vs a ~25 us traffic floor)
             float4 acc = {0.f, 0.f, 0.f, 0.f};
+            float4 acc2 = {0.f, 0.f, 0.f, 0.f};
             long long n = n0;
-            for (; n + 1 < n1; n += 2) {  // 2-row unroll
-                const float inv0 = 1.0f / fmaxf(row_sums[n], 1e-12f);
-                const float inv1 = 1.0f / fmaxf(row_sums[n + 1], 1e-12f);
+            for (; n + 3 < n1; n += 4) {
+                const float i0 = 1.0f / fmaxf(row_sums[n], 1e-12f);
+                const float i1 = 1.0f / fmaxf(row_sums[n + 1], 1e-12f);
+                const float i2 = 1.0f / fmaxf(row_sums[n + 2], 1e-12f);
+                const float i3 = 1.0f / fmaxf(row_sums[n + 3], 1e-12f);
                 const float4 v0 = *reinterpret_cast<const float4*>(
                     adjusted + n * C + c4);
                 const float4 v1 = *reinterpret_cast<const float4*>(
                     adjusted + (n + 1) * C + c4);
-                acc.x += v0.x * inv0 + v1.x * inv1;
-                acc.y += v0.y * inv0 + v1.y * inv1;
-                acc.z += v0.z * inv0 + v1.z * inv1;
-                acc.w += v0.w * inv0 + v1.w * inv1;
+                const float4 v2 = *reinterpret_cast<const float4*>(
+                    adjusted + (n + 2) * C + c4);
+                const float4 v3 = *reinterpret_cast<const float4*>(
+                    adjusted + (n + 3) * C + c4);
+                acc.x += v0.x * i0 + v1.x * i1;
+                acc.y += v0.y * i0 + v1.y * i1;
+                acc.z += v0.z * i0 + v1.z * i1;
+                acc.w += v0.w * i0 + v1.w * i1;
+                acc2.x += v2.x * i2 + v3.x * i3;
+                acc2.y += v2.y * i2 + v3.y * i3;
+                acc2.z += v2.z * i2 + v3.z * i3;
+                acc2.w += v2.w * i2 + v3.w * i3;
             }
             for (; n < n1; ++n) {
                 const float inv = 1.0f / fmaxf(row_sums[n], 1e-12f);
@@ -1034,6 +1048,8 @@ pi_marginal_kernel(const float* __restrict__ adjusted,  // (N, C)
                 acc.x += v.x * inv; acc.y += v.y * inv;
                 acc.z += v.z * inv; acc.w += v.w * inv;
             }
+            acc.x += acc2.x; acc.y += acc2.y;
+            acc.z += acc2.z; acc.w += acc2.w;
             atomicAdd(out + c4 + 0, acc.x);
             atomicAdd(out + c4 + 1, acc.y);
             atomicAdd(out + c4 + 2, acc.z);
