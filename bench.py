@@ -1,18 +1,24 @@
 """Flagship benchmark: acquisition steps/sec of the CODA engine.
 
-Measures the full active-model-selection serving step (EIG acquisition over
-the candidate pool -> oracle label -> Dirichlet posterior update -> pi_hat
-refresh -> best-model P(best)) on the ImageNet-1k-scale config named in
-BASELINE.json (config 2): H=128 candidate models x N=50k points x C=1000
-classes, synthetic prediction tensors, fp32 compute (the reference's compute
-dtype - its loader up-casts storage to fp32, coda/datasets.py:14).
+Measures the full active-model-selection serving step (the reference's
+CANONICAL no-prefilter acquisition - EIG over EVERY disagreeing
+unlabeled point - -> oracle label -> Dirichlet posterior update ->
+pi_hat refresh -> best-model P(best)) on the ImageNet-1k-scale config
+named in BASELINE.json (config 2): H=128 candidate models x 50k points
+per GPU x C=1000 classes, synthetic prediction tensors, fp32 compute
+(the reference's compute dtype - its loader up-casts storage to fp32,
+coda/datasets.py:14).
 
-Scaling is WEAK: each GPU contributes 128 models (1 GPU = the 128-model
-ImageNet-1k config; 8 GPUs = a 1024-model pool, BASELINE.json config 3),
-sharded over RCCL/xGMI, and the whole job runs ONE shared selection loop
-over the combined pool - so steps/sec at fixed per-GPU work is the
-whole-job metric. The reference publishes no absolute throughput numbers
- => vs_baseline = null.
+Scaling is WEAK along the CANDIDATE POOL: each GPU contributes 50k
+unlabeled points (the model axis stores sharded - 128/world models per
+rank - while EIG work shards by candidate; see the replicated-Beta
+design in PARITY.md). One shared selection loop runs over the combined
+pool, so whole-job steps/sec at fixed per-GPU work is the metric. The
+candidate axis is the one that scales per-GPU-flat for this algorithm:
+model-axis growth makes every candidate's acquisition intrinsically
+more expensive (distinct-classes x 2H per candidate), which no
+implementation can hold per-GPU-constant. The reference publishes no
+absolute throughput numbers => vs_baseline = null.
 
 Contract (driver): `python bench.py --gpus N --steps K --warmup W`; for
 N > 1 the driver launches it under torch.distributed.run with one rank per
@@ -37,8 +43,8 @@ from coda_amd.parallel import init_from_env, get_comm
 # PREFILTER=0 is the reference's canonical acquisition (main.py:49, run
 # name `coda-lr=0.01-mult=2.0-no-prefilter`, paper/tab1.py:60): EIG over
 # EVERY disagreeing unlabeled point each step - ~50k candidates here.
-H_PER_GPU = int(os.environ.get("CODA_BENCH_H", 128))
-N_POINTS = int(os.environ.get("CODA_BENCH_N", 50_000))
+H_MODELS = int(os.environ.get("CODA_BENCH_H", 128))
+N_PER_GPU = int(os.environ.get("CODA_BENCH_N", 50_000))
 C_CLASSES = int(os.environ.get("CODA_BENCH_C", 1000))
 PREFILTER_N = int(os.environ.get("CODA_BENCH_PREFILTER", 0))
 CHUNK = int(os.environ.get("CODA_BENCH_CHUNK", 256))
@@ -95,7 +101,8 @@ def main():
         raise SystemExit(
             f"--gpus {n_gpus} requested but WORLD_SIZE is unset; "
             "multi-GPU runs go through torch.distributed.run")
-    H_TOTAL = H_PER_GPU * comm.world  # weak scaling: 128 models per GPU
+    H_TOTAL = H_MODELS                # fixed pool of models
+    N_TOTAL = N_PER_GPU * comm.world  # weak scaling: 50k points per GPU
 
     if torch.cuda.is_available():
         device = comm.device or torch.device("cuda", 0)
@@ -107,7 +114,7 @@ def main():
         if shard else list(range(H_TOTAL))
 
     from coda_amd.datasets import STORAGE_DTYPES
-    preds, labels = synth_preds(model_idxs, N_POINTS, C_CLASSES, device,
+    preds, labels = synth_preds(model_idxs, N_TOTAL, C_CLASSES, device,
                                 dtype=STORAGE_DTYPES[STORAGE])
     ds = Dataset.from_tensors(preds, labels, device, shard=None)
     ds.total_models = H_TOTAL
@@ -166,13 +173,15 @@ def main():
                 "model": "coda-eig",
                 "task": "imagenet1k-scale",
                 "H_models": H_TOTAL,
-                "N_points": N_POINTS,
+                "N_points": N_TOTAL,
                 "C_classes": C_CLASSES,
                 "prefilter_n": PREFILTER_N,
                 "chunk_size": CHUNK,
-                "global_batch": PREFILTER_N or N_POINTS,
+                "global_batch": PREFILTER_N or N_TOTAL,
                 "seq_len": C_CLASSES,
-                "parallelism": f"model-shard{comm.world}",
+                "parallelism":
+                    f"cand-shard{comm.world}" if comm.world > 1
+                    else "single",
             },
         }))
 

@@ -214,8 +214,9 @@ def test_bench_torchrun_cpu():
     line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
     d = json.loads(line)
     assert d["n_gpus"] == 2 and d["steps"] == 2
-    assert d["config"]["H_models"] == 12  # weak scaling: 6 per rank
-    assert d["config"]["parallelism"] == "model-shard2"
+    assert d["config"]["H_models"] == 6     # fixed model pool
+    assert d["config"]["N_points"] == 300   # weak: 150 points per rank
+    assert d["config"]["parallelism"] == "cand-shard2"
     assert d["value"] > 0 and d["higher_is_better"] is True
 
 
