@@ -630,3 +630,42 @@ class TestPairKernels:
         scale = float(eig_t.abs().max())
         err = float((eig_k - eig_t).abs().max())
         assert err < max(5e-3 * scale, 3e-4), (err, scale)
+
+    def test_checkpoint_restore_with_pair_engine(self, dev):
+        """Mid-run checkpoint/restore on GPU: the pair structures,
+        hipGraphs and replicated caches must rebuild cleanly and the
+        resumed trajectory must continue identically."""
+        import bench
+        from coda_amd import CODA, Oracle, checkpoint as ckpt
+        from coda_amd.datasets import Dataset
+        from coda_amd.options import LOSS_FNS
+
+        preds, labels = bench.synth_preds(list(range(12)), 700, 10, dev)
+        ds = Dataset.from_tensors(preds, labels, dev)
+        oracle = Oracle(ds, LOSS_FNS["acc"])
+
+        def steps(sel, n, seed0):
+            out = []
+            for m in range(n):
+                random.seed(seed0 + m)
+                i, q = sel.get_next_item_to_label()
+                sel.add_label(i, oracle(int(i)), q)
+                out.append(int(i))
+            return out
+
+        random.seed(0); torch.manual_seed(0)
+        ref = CODA(ds)
+        t_ref = steps(ref, 8, 100)
+
+        random.seed(0); torch.manual_seed(0)
+        sel = CODA(ds)
+        t_a = steps(sel, 4, 100)
+        blob = ckpt.state_dict(sel)
+        random.seed(0); torch.manual_seed(0)
+        sel2 = CODA(ds)
+        ckpt.load_state_dict(sel2, blob)
+        t_b = steps(sel2, 4, 104)
+        assert t_a + t_b == t_ref
+        torch.testing.assert_close(sel2.get_pbest().cpu(),
+                                   ref.get_pbest().cpu(),
+                                   rtol=1e-4, atol=1e-6)
