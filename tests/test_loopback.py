@@ -71,3 +71,23 @@ def test_loopback_sharded_table_gpu():
     for traj, pbest in results:
         assert traj == st, (traj, st)
         torch.testing.assert_close(pbest, sp, rtol=2e-3, atol=1e-5)
+
+
+@pytest.mark.gpu
+def test_loopback_replicated_pair_gpu():
+    """World-4 replicated-beta candidate-sharded PAIR path (the v3
+    default for sharded runs: init gathers + pair kernels + the (B,)
+    EIG gather) on one GPU vs single-device."""
+    assert torch.cuda.is_available()
+    dev = "cuda:0"
+    preds, labels = make_synthetic_task(H=8, N=200, C=5, seed=12)
+    from coda_amd.parallel import Comm
+    st, sp = _trajectory(Comm(), preds, labels, dev, eig_impl="pair")
+    results = run_ranks(
+        4, lambda comm: _trajectory(comm, preds, labels, dev,
+                                    eig_impl="pair"), device=dev)
+    for traj, pbest in results:
+        assert [t[0] for t in traj] == [t[0] for t in st], (traj, st)
+        torch.testing.assert_close(pbest, sp, rtol=2e-3, atol=1e-5)
+        for (_, qa), (_, qb) in zip(traj, st):
+            assert abs(qa - qb) <= 1e-3 * max(1.0, abs(qb))
