@@ -82,10 +82,16 @@ def test_loopback_replicated_pair_gpu():
     dev = "cuda:0"
     preds, labels = make_synthetic_task(H=8, N=200, C=5, seed=12)
     from coda_amd.parallel import Comm
-    st, sp = _trajectory(Comm(), preds, labels, dev, eig_impl="pair")
+    # fp32 pi_hat isolates the sharding logic: the bf16 pi fast path
+    # rounds differently per GEMM shape (documented deviation), which
+    # would otherwise add ~1e-3-relative noise on top of sharding
+    st, sp = _trajectory(Comm(), preds, labels, dev, eig_impl="pair",
+                         pi_hat_precision="fp32")
     results = run_ranks(
         4, lambda comm: _trajectory(comm, preds, labels, dev,
-                                    eig_impl="pair"), device=dev)
+                                    eig_impl="pair",
+                                    pi_hat_precision="fp32"),
+        device=dev)
     for traj, pbest in results:
         assert [t[0] for t in traj] == [t[0] for t in st], (traj, st)
         torch.testing.assert_close(pbest, sp, rtol=2e-3, atol=1e-5)
