@@ -754,6 +754,9 @@ torch::Tensor pair_eig_finalize(torch::Tensor h_after,
     const int B = cand_ids.size(0);
     const int C = adjusted.size(1);
     TORCH_CHECK(h_base.is_contiguous() && h_base.size(0) == C);
+    TORCH_CHECK(cand_ids.is_contiguous() && cand_off.is_contiguous()
+                && cand_pairs.is_contiguous(),
+                "finalize inputs must be contiguous");
     auto q = torch::empty({B}, adjusted.options());
     auto stream = c10::hip::getCurrentHIPStream();
     hipLaunchKernelGGL(pairops::pair_eig_finalize_kernel,

@@ -241,7 +241,12 @@ def build_pairs(cls_rows: torch.Tensor, cand_ids: torch.Tensor,
     seg_off = torch.zeros(K + 1, dtype=torch.int32, device=device)
     seg_off[1:] = seg_len_k.cumsum(0).to(torch.int32)
 
-    return PairStructure(cand_ids=cand_ids.long(), pair_b=pair_b,
+    # contiguous() is load-bearing: sharded callers pass strided
+    # candidate slices (ids[rank::world]), and the finalize kernel
+    # walks the raw buffer - a strided view would silently read the
+    # WRONG candidates' rows (caught by the GPU loopback world-4 test)
+    return PairStructure(cand_ids=cand_ids.long().contiguous(),
+                         pair_b=pair_b,
                          pair_c=pair_c, seg_off=seg_off, seg_h=seg_h,
                          base_pos=run_off[:-1].clone(), n_real=K2,
                          tile=tile, cand_off=cand_off,
