@@ -76,6 +76,10 @@ def load_state_dict(selector, state: Dict[str, Any]):
         selector._pbest_rows_cache = (-1, None)
         selector._label_graph = None  # graph buffers alias replaced state
         selector._acq_graph = None
+        selector._acq_out = None      # sized for the stale pair set
+        selector._acq_fresh = False
+        selector._acq_saved = None
+        selector._merged_acq = False
         selector._pairs_static = None  # hit structure covers a stale set
         selector._pair_row_of = None
         selector._active_mask = None

@@ -495,7 +495,7 @@ class CODA(ModelSelector):
         best_val, best_idx = q.max(0)
         nt = (torch.isclose(q, best_val, rtol=1e-8)
               & self._active_mask).sum()
-        self._acq_q = q
+        self._acq_qbuf.copy_(q)   # stable storage across graph pools
         self._acq_out.copy_(torch.stack(
             [best_val.double(), best_idx.double(), nt.double()]))
 
@@ -505,7 +505,7 @@ class CODA(ModelSelector):
         if nt > 1:
             # same tie semantics as the eager path: active candidates
             # ascend by point id in both orderings
-            q = self._acq_q
+            q = self._acq_qbuf
             ties = (torch.isclose(q, q.max(), rtol=1e-8)
                     & self._active_mask)
             pos = random.choice(
@@ -514,12 +514,26 @@ class CODA(ModelSelector):
             return self._pairs_ids_host[pos], float(q[pos])
         return self._pairs_ids_host[bi], bv
 
-    def _graphed_acquire(self):
-        if self._acq_graph is None:
+    def _init_acq_buffers(self):
+        if getattr(self, "_acq_out", None) is None:
             ps, _ = self._pairs_static
             self._pairs_ids_host = ps.cand_ids.cpu().tolist()
             self._acq_out = torch.zeros(3, dtype=torch.float64,
                                         device=self.device)
+            self._acq_qbuf = torch.empty(ps.cand_ids.numel(),
+                                         device=self.device)
+
+    def _acq_eligible(self) -> bool:
+        return (self.q == "eig" and self._pairs_static is not None
+                and self._pairs_static[0].vmask is not None
+                and not DEBUG_VIZ
+                and not (self.prefilter_n
+                         and len(self._active_candidates)
+                         > self.prefilter_n))
+
+    def _graphed_acquire(self):
+        if self._acq_graph is None:
+            self._init_acq_buffers()
             side = torch.cuda.Stream()
             side.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(side):
@@ -537,14 +551,17 @@ class CODA(ModelSelector):
 
     # ------------------------------------------------------------------
     def get_next_item_to_label(self):
+        if getattr(self, "_acq_fresh", False) and self._acq_eligible():
+            # the merged label+acquire graph already computed this
+            # step's EIG/argmax during add_label
+            self._acq_fresh = False
+            if getattr(self, "_acq_saved", None) is not None:
+                res, self._acq_saved = self._acq_saved, None
+                return res
+            return self._acq_result()
         if (self.q == "eig" and self._use_label_graph
                 and self._label_graph is not None
-                and self._pairs_static is not None
-                and self._pairs_static[0].vmask is not None
-                and not DEBUG_VIZ
-                and not (self.prefilter_n
-                         and len(self._active_candidates)
-                         > self.prefilter_n)):
+                and self._acq_eligible()):
             return self._graphed_acquire()
         if self.q == "eig":
             q_vals, cand = self.eig_batched()
@@ -668,8 +685,19 @@ class CODA(ModelSelector):
             self._g_rows = self._pbest_rows_before().clone().contiguous()
             self._g_idx.fill_(idx)
             self._g_y.fill_(true_class)
-            # warmup on a side stream (required before capture), then
-            # capture a second execution
+            # When the full-pool pair acquisition is active, the NEXT
+            # step's acquisition is captured INTO the label graph: one
+            # replay per step computes posterior update + next EIG +
+            # argmax (the caller's get_next just reads the result).
+            self._merged_acq = self._acq_eligible()
+            if self._merged_acq:
+                self._init_acq_buffers()
+
+            def body():
+                self._label_update_body()
+                if self._merged_acq:
+                    self._acq_body()
+
             # warmup executes the body for real (THIS label's update) on
             # a side stream, as stream capture requires; the capture pass
             # then only RECORDS the ops (no execution, no state change),
@@ -677,16 +705,21 @@ class CODA(ModelSelector):
             side = torch.cuda.Stream()
             side.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(side):
-                self._label_update_body()
+                body()
             torch.cuda.current_stream().wait_stream(side)
             torch.cuda.synchronize()
+            if self._merged_acq:
+                # the warmup's acquisition result serves the next
+                # get_next (capture below records without executing)
+                self._acq_saved = self._acq_result()
             g = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g):
-                self._label_update_body()
+                body()
             self._label_graph = g
             self.pi_hat = self._g_pi
             self._pbest_rows_cache = (self._posterior_version + 1,
                                       self._g_rows)
+            self._acq_fresh = self._merged_acq
             return
         self._g_idx.fill_(idx)
         self._g_y.fill_(true_class)
@@ -694,6 +727,7 @@ class CODA(ModelSelector):
         self.pi_hat = self._g_pi
         self._pbest_rows_cache = (self._posterior_version + 1,
                                   self._g_rows)
+        self._acq_fresh = self._merged_acq
 
     def add_label(self, idx, true_class, selection_prob):
         """Posterior update (K13) + incremental pi_hat refresh.
@@ -705,6 +739,9 @@ class CODA(ModelSelector):
         idx = int(idx)
         if (self._use_label_graph and self._tables is not None
                 and not self._tables_dirty):
+            # the candidate mask update must be enqueued BEFORE the
+            # replay: the merged graph's acquisition half reads it
+            self._deactivate(idx)
             self._graphed_add_label(idx, int(true_class))
             self._posterior_version += 1
             self._pi_xi_cache = None
@@ -712,7 +749,6 @@ class CODA(ModelSelector):
             self.labels.append(int(true_class))
             self.q_vals.append(selection_prob)
             self.unlabeled_idxs.remove(idx)
-            self._deactivate(idx)
             return
         if self.dirichlets.is_cuda and ops.hip_available():
             # H-thread scatter kernel; torch's one_hot + dim-1 index_add_
@@ -768,6 +804,7 @@ class CODA(ModelSelector):
         if idx in self.unlabeled_idxs:
             self.unlabeled_idxs.remove(idx)
         self._deactivate(idx)
+        self._acq_fresh = False  # precomputed EIG predates the skip
 
     # ------------------------------------------------------------------
     def get_pbest(self):
