@@ -74,6 +74,7 @@
 #endif
 #define ROWS_PER_BLOCK 4
 #define PTS_PER_LANE 4
+#define ES_LONG_ROW 512
 
 namespace {
 
@@ -660,6 +661,8 @@ es_build_kernel(const float* __restrict__ s_base,   // (C, P)
     const int k0 = offsets[(size_t)b * (C + 1) + c];
     const int k1 = offsets[(size_t)b * (C + 1) + c + 1];
     const bool comp = (k1 - k0) > H / 2;
+    const int len_eff = comp ? (H - (k1 - k0)) : (k1 - k0);
+    if (len_eff > ES_LONG_ROW) return;  // block-per-row kernel's rows
 
     float a0[16] = {}, a1[16] = {};
     auto addrow = [&](int h, float* acc) {
@@ -724,6 +727,97 @@ es_build_kernel(const float* __restrict__ s_base,   // (C, P)
     for (int i = 0; i < 16; ++i) dst[i] = (TOUT)out[i];
 }
 
+// Block-per-row companion for rows whose (possibly complemented)
+// bucket is still long (wide pools: a 10k-model consensus class leaves
+// ~2500 rows even after complementing - a serial per-wave chain).
+// 4 waves split the k-range; lane owns 4 grid points; partials reduce
+// through LDS in fixed wave order (deterministic).
+template <typename TOUT>
+__global__ void __launch_bounds__(BLOCK)
+es_build_long_kernel(const float* __restrict__ s_base,
+                     const float* __restrict__ delta,
+                     const float* __restrict__ dall,
+                     const int* __restrict__ hvals,
+                     const int* __restrict__ offsets,
+                     const float* __restrict__ w,
+                     TOUT* __restrict__ es,
+                     int B, int C, int H) {
+    const int r = blockIdx.x;
+    if (r >= B * C) return;
+    const int b = r / C, c = r - b * C;
+    const int k0 = offsets[(size_t)b * (C + 1) + c];
+    const int k1 = offsets[(size_t)b * (C + 1) + c + 1];
+    const bool comp = (k1 - k0) > H / 2;
+    const int len_eff = comp ? (H - (k1 - k0)) : (k1 - k0);
+    if (len_eff <= ES_LONG_ROW) return;  // short kernel's rows
+
+    const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    const int p0 = lane * PTS_PER_LANE;
+    const size_t dbase = (size_t)c * H * P_POINTS + p0;
+    const int* hrow = hvals + (size_t)b * H;
+
+    float a0[4] = {0.f, 0.f, 0.f, 0.f}, a1[4] = {0.f, 0.f, 0.f, 0.f};
+    auto addrow = [&](int h, float* acc) {
+        const float4 d = *reinterpret_cast<const float4*>(
+            delta + dbase + (size_t)h * P_POINTS);
+        acc[0] += d.x; acc[1] += d.y; acc[2] += d.z; acc[3] += d.w;
+    };
+    // the wave's k-slice: [lo, hi) of the (complement) iteration space
+    auto scan = [&](int s0, int s1) {
+        const int n = s1 - s0;
+        const int per = (n + 3) / 4;
+        int lo = s0 + wave * per;
+        int hi = min(s1, lo + per);
+        int k = lo;
+        for (; k + 1 < hi; k += 2) {
+            addrow(hrow[k], a0);
+            addrow(hrow[k + 1], a1);
+        }
+        if (k < hi) addrow(hrow[k], a0);
+    };
+    if (!comp) {
+        scan(k0, k1);
+    } else {
+        // complement = positions outside [k0, k1) of the same row;
+        // treat as one logical range of length k0 + (H - k1)
+        const int n = k0 + (H - k1);
+        const int per = (n + 3) / 4;
+        int lo = wave * per;
+        int hi = min(n, lo + per);
+        for (int s = lo; s < hi; ++s) {
+            const int k = (s < k0) ? s : (k1 + (s - k0));
+            addrow(hrow[k], a0);
+        }
+    }
+
+    __shared__ float part[4][P_POINTS];
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+        part[wave][p0 + j] = a0[j] + a1[j];
+    __syncthreads();
+    if (wave == 0) {
+        const float4 sb = *reinterpret_cast<const float4*>(
+            s_base + (size_t)c * P_POINTS + p0);
+        const float4 wv = *reinterpret_cast<const float4*>(w + p0);
+        float sx[4] = {sb.x, sb.y, sb.z, sb.w};
+        float wx[4] = {wv.x, wv.y, wv.z, wv.w};
+        float dx[4] = {0.f, 0.f, 0.f, 0.f};
+        if (comp) {
+            const float4 dv = *reinterpret_cast<const float4*>(
+                dall + (size_t)c * P_POINTS + p0);
+            dx[0] = dv.x; dx[1] = dv.y; dx[2] = dv.z; dx[3] = dv.w;
+        }
+        TOUT* dst = es + ((size_t)c * B + b) * P_POINTS + p0;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            const float dsum0 = ((part[0][p0 + j] + part[1][p0 + j])
+                               + (part[2][p0 + j] + part[3][p0 + j]));
+            const float dsum = comp ? dx[j] - dsum0 : dsum0;
+            dst[j] = (TOUT)(exp2f(sx[j] + dsum) * wx[j]);
+        }
+    }
+}
+
 template <typename TM>
 __global__ void __launch_bounds__(BLOCK)
 eig_assemble_kernel(const TM* __restrict__ m,               // (C, B, 2H)
@@ -739,25 +833,37 @@ eig_assemble_kernel(const TM* __restrict__ m,               // (C, B, 2H)
     const int lane = threadIdx.x & 63;
     const TM* row = m + ((size_t)c * B + b) * (2 * H);
 
-    float total = 0.f;
-    for (int h = lane; h < H; h += 64) {
+    float t0 = 0.f, t1 = 0.f;
+    for (int h = lane; h < H; h += 128) {
         const int v = (cls[(size_t)b * H + h] == c) ? 1 : 0;
-        total += (float)row[2 * h + v];
+        t0 += (float)row[2 * h + v];
+        const int h2 = h + 64;
+        if (h2 < H) {
+            const int v2 = (cls[(size_t)b * H + h2] == c) ? 1 : 0;
+            t1 += (float)row[2 * h2 + v2];
+        }
     }
-    total = wave_reduce_sum(total);
+    float total = wave_reduce_sum(t0 + t1);
     const float inv = 1.0f / fmaxf(total, kEps);
 
     const float pi_c = pi_hat[c];
-    float ent = 0.f;
-    for (int h = lane; h < H; h += 64) {
+    float e0 = 0.f, e1 = 0.f;
+    for (int h = lane; h < H; h += 128) {
         const int v = (cls[(size_t)b * H + h] == c) ? 1 : 0;
         const float pb = (float)row[2 * h + v] * inv;
         float mm = mixture0[h]
                  + pi_c * (pb - pbest_before[(size_t)c * H + h]);
-        mm = fmaxf(mm, 1e-12f);
-        ent += -mm * __log2f(mm);
+        e0 += -fmaxf(mm, 1e-12f) * __log2f(fmaxf(mm, 1e-12f));
+        const int h2 = h + 64;
+        if (h2 < H) {
+            const int v2 = (cls[(size_t)b * H + h2] == c) ? 1 : 0;
+            const float pb2 = (float)row[2 * h2 + v2] * inv;
+            float m2 = mixture0[h2]
+                     + pi_c * (pb2 - pbest_before[(size_t)c * H + h2]);
+            e1 += -fmaxf(m2, 1e-12f) * __log2f(fmaxf(m2, 1e-12f));
+        }
     }
-    ent = wave_reduce_sum(ent);
+    const float ent = wave_reduce_sum(e0 + e1);
     if (lane == 0) h_after[r] = ent;
 }
 
@@ -985,10 +1091,19 @@ pi_hat_delta_part_kernel(const T* __restrict__ preds,   // (H, N, C)
     const int hend = min(H, hbeg + Hc);
     const long long n = (long long)blockIdx.x * BLOCK + threadIdx.x;
     if (n >= N) return;
-    float acc = 0.f;
-    for (int h = hbeg; h < hend; ++h)
-        acc += (float)preds[((long long)h * N + n) * C + cls[h]];
-    partial[(size_t)blockIdx.y * N + n] = acc;
+    // 4 accumulators: the single-chain form serializes ~H/KH scattered
+    // loads per thread (wide pools: 1.24 ms at 10k models, 3x floor)
+    float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+    int h = hbeg;
+    for (; h + 3 < hend; h += 4) {
+        a0 += (float)preds[((long long)h * N + n) * C + cls[h]];
+        a1 += (float)preds[((long long)(h + 1) * N + n) * C + cls[h + 1]];
+        a2 += (float)preds[((long long)(h + 2) * N + n) * C + cls[h + 2]];
+        a3 += (float)preds[((long long)(h + 3) * N + n) * C + cls[h + 3]];
+    }
+    for (; h < hend; ++h)
+        a0 += (float)preds[((long long)h * N + n) * C + cls[h]];
+    partial[(size_t)blockIdx.y * N + n] = (a0 + a1) + (a2 + a3);
 }
 
 
@@ -1345,6 +1460,31 @@ torch::Tensor es_build(torch::Tensor s_base, torch::Tensor delta,
                            hvals.data_ptr<int>(),
                            offsets.data_ptr<int>(), w.data_ptr<float>(),
                            es.data_ptr<float>(), B, C, H);
+    }
+    if (H > 2 * ES_LONG_ROW) {  // long rows only exist at wide pools
+        if (bf16_out) {
+            hipLaunchKernelGGL(es_build_long_kernel<hip_bfloat16>,
+                               dim3(R), dim3(BLOCK), 0, stream.stream(),
+                               s_base.data_ptr<float>(),
+                               delta.data_ptr<float>(),
+                               dall.data_ptr<float>(),
+                               hvals.data_ptr<int>(),
+                               offsets.data_ptr<int>(),
+                               w.data_ptr<float>(),
+                               reinterpret_cast<hip_bfloat16*>(
+                                   es.data_ptr()),
+                               B, C, H);
+        } else {
+            hipLaunchKernelGGL(es_build_long_kernel<float>,
+                               dim3(R), dim3(BLOCK), 0, stream.stream(),
+                               s_base.data_ptr<float>(),
+                               delta.data_ptr<float>(),
+                               dall.data_ptr<float>(),
+                               hvals.data_ptr<int>(),
+                               offsets.data_ptr<int>(),
+                               w.data_ptr<float>(),
+                               es.data_ptr<float>(), B, C, H);
+        }
     }
     C10_HIP_CHECK(hipGetLastError());
     return es;
