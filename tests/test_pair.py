@@ -165,3 +165,57 @@ class TestPairEig:
         sel.skip(i0)
         i1, _ = sel.get_next_item_to_label()
         assert int(i1) != int(i0)
+
+
+class TestDedupeProperties:
+    @pytest.mark.parametrize("H,N,C,seed", [(9, 200, 6, 0), (16, 300, 3, 1),
+                                            (5, 100, 50, 2), (32, 150, 8, 3)])
+    def test_dedupe_equals_no_dedupe(self, H, N, C, seed):
+        """EIG through the deduplicated structure == through the
+        1:1 structure, for random shapes (guards the set-hash grouping
+        and the CSR remap)."""
+        (preds, cls, dirichlets, pi_hat, adjusted,
+         row_sums) = _random_problem(H, N, C, seed=seed)
+        alpha_cc, beta_cc = R.dirichlet_to_beta(dirichlets)
+        tables = tops.table_precompute(alpha_cc, beta_cc)
+        pbest_before = R.pbest_from_beta(alpha_cc.t().contiguous(),
+                                         beta_cc.t().contiguous())
+        mixture0, H_before = R.mixture_entropy(pbest_before, pi_hat)
+        ids = torch.arange(N)
+        cls_rows = cls[:, ids].t().contiguous()
+        out = {}
+        for dd in (True, False):
+            ps = pops.build_pairs(cls_rows, ids, C, dedupe=dd)
+            out[dd] = pops.eig_pairs(tables, ps, cls_rows, pbest_before,
+                                     pi_hat, mixture0, H_before,
+                                     adjusted, row_sums)
+        assert out[True].shape == out[False].shape
+        torch.testing.assert_close(out[True], out[False], rtol=1e-5,
+                                   atol=1e-7)
+
+    def test_strided_candidate_slice_matches_full(self):
+        """Slice structures (the sharded per-rank view, incl. strided
+        cand_ids) reproduce the full-pool EIG exactly on CPU."""
+        (preds, cls, dirichlets, pi_hat, adjusted,
+         row_sums) = _random_problem(11, 180, 7, seed=9)
+        alpha_cc, beta_cc = R.dirichlet_to_beta(dirichlets)
+        tables = tops.table_precompute(alpha_cc, beta_cc)
+        pbest_before = R.pbest_from_beta(alpha_cc.t().contiguous(),
+                                         beta_cc.t().contiguous())
+        mixture0, H_before = R.mixture_entropy(pbest_before, pi_hat)
+        ids = torch.arange(180)
+        cls_all = cls[:, ids].t().contiguous()
+        ps = pops.build_pairs(cls_all, ids, 7)
+        q_full = pops.eig_pairs(tables, ps, cls_all, pbest_before,
+                                pi_hat, mixture0, H_before, adjusted,
+                                row_sums)
+        for r in range(3):
+            mine = ids[r::3]             # non-contiguous slice
+            cls_r = cls[:, mine].t().contiguous()
+            ps_r = pops.build_pairs(cls_r, mine, 7)
+            assert ps_r.cand_ids.is_contiguous()
+            q_r = pops.eig_pairs(tables, ps_r, cls_r, pbest_before,
+                                 pi_hat, mixture0, H_before, adjusted,
+                                 row_sums)
+            torch.testing.assert_close(q_r, q_full[r::3], rtol=1e-6,
+                                       atol=1e-8)
