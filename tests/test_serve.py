@@ -1,5 +1,8 @@
 """Serving layer: the interactive step-wise protocol over FastAPI."""
+import os
+
 import pytest
+import torch
 
 fastapi = pytest.importorskip("fastapi")
 from fastapi.testclient import TestClient  # noqa: E402
@@ -128,3 +131,85 @@ def test_interactive_loop_gpu():
     assert st["step"] == 2
     pb = c.get("/pbest").json()["pbest"]
     assert len(pb) == 6 and abs(sum(pb) - 1.0) < 1e-3
+
+
+def _make_tiny_clip(ckpt_dir: str):
+    """Random-init CLIP checkpoint + minimal byte-level tokenizer, built
+    entirely offline (the environment has no model weights): enough for
+    the zero-shot pipeline to execute its real code path."""
+    import json
+    from transformers import (CLIPConfig, CLIPModel, CLIPTokenizer,
+                              CLIPTextConfig, CLIPVisionConfig,
+                              CLIPProcessor, CLIPImageProcessor)
+    os.makedirs(ckpt_dir, exist_ok=True)
+    vocab = {"<|startoftext|>": 0, "<|endoftext|>": 1}
+    chars = "abcdefghijklmnopqrstuvwxyz "
+    for i, ch in enumerate(chars):
+        vocab[ch] = 2 + i
+        vocab[ch + "</w>"] = 2 + len(chars) + i
+    vp = os.path.join(ckpt_dir, "vocab.json")
+    json.dump(vocab, open(vp, "w"))
+    mp = os.path.join(ckpt_dir, "merges.txt")
+    open(mp, "w").write("#version: 0.2\n")
+    tok = CLIPTokenizer(vp, mp)
+    cfg = CLIPConfig(
+        text_config=CLIPTextConfig(
+            vocab_size=len(vocab), hidden_size=32, intermediate_size=64,
+            num_hidden_layers=2, num_attention_heads=2,
+            max_position_embeddings=32, bos_token_id=0,
+            eos_token_id=1).to_dict(),
+        vision_config=CLIPVisionConfig(
+            hidden_size=32, intermediate_size=64, num_hidden_layers=2,
+            num_attention_heads=2, image_size=32,
+            patch_size=8).to_dict(),
+        projection_dim=16)
+    CLIPModel(cfg).save_pretrained(ckpt_dir)
+    CLIPProcessor(
+        image_processor=CLIPImageProcessor(
+            size={"shortest_edge": 32},
+            crop_size={"height": 32, "width": 32}),
+        tokenizer=tok).save_pretrained(ckpt_dir)
+
+
+def test_hf_zero_shot_builder_end_to_end(tmp_path, monkeypatch):
+    """The HF zero-shot model path of serve/build_predictions executes
+    end-to-end offline (random-init tiny CLIP checkpoints; reference
+    counterpart demo/hf_zeroshot.py:71-219): image-folder scan -> two
+    zero-shot pipelines -> stacked (H, N, C) .pt + labels, loadable by
+    the Dataset."""
+    pytest.importorskip("transformers")
+    from PIL import Image
+
+    monkeypatch.setenv("HF_HUB_OFFLINE", "1")
+    monkeypatch.setenv("TRANSFORMERS_OFFLINE", "1")
+    imgdir = tmp_path / "imgs"
+    for ci, cls in enumerate(["cat", "dog"]):
+        (imgdir / cls).mkdir(parents=True)
+        for j in range(3):
+            Image.new("RGB", (32, 32),
+                      (40 * ci + 20 * j, 80, 120)).save(
+                imgdir / cls / f"{j}.png")
+    ck1 = tmp_path / "m1"
+    ck2 = tmp_path / "m2"
+    _make_tiny_clip(str(ck1))
+    _make_tiny_clip(str(ck2))
+
+    out = tmp_path / "hf_task.pt"
+    from coda_amd.serve import build_predictions as bp
+    import sys
+    argv = ["prog", "--images", str(imgdir),
+            "--models", f"{ck1},{ck2}", "--out", str(out)]
+    monkeypatch.setattr(sys, "argv", argv)
+    bp.main()
+
+    preds = torch.load(str(out), weights_only=False)
+    assert preds.shape == (2, 6, 2)
+    torch.testing.assert_close(preds.sum(-1),
+                               torch.ones(2, 6), rtol=1e-4, atol=1e-4)
+    labels = torch.load(str(out).replace(".pt", "_labels.pt"),
+                        weights_only=False)
+    assert labels.tolist() == [0, 0, 0, 1, 1, 1]
+
+    from coda_amd.datasets import Dataset
+    ds = Dataset(str(out), device="cpu")
+    assert ds.preds.shape == (2, 6, 2) and ds.labels is not None
