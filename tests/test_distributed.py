@@ -31,6 +31,13 @@ def _single_trajectory(preds, labels, steps=4):
     return out
 
 
+
+def _free_port() -> str:
+    import socket
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return str(s.getsockname()[1])
+
 def _worker(rank, world, init_file, preds, labels, steps, q):
     os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
                       LOCAL_RANK=str(rank))
@@ -177,7 +184,7 @@ def test_main_sharded_torchrun_cpu(tmp_path):
     r = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", "29659", "main.py", "--task", "sh",
+         "--master-port", _free_port(), "main.py", "--task", "sh",
          "--data-dir", "data", "--method", "coda", "--iters", "3",
          "--seeds", "1", "--sharded", "--device", "cpu",
          "--chunk-size", "32"],
@@ -207,7 +214,7 @@ def test_bench_torchrun_cpu():
     r = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", "29663", "bench.py", "--gpus", "2",
+         "--master-port", _free_port(), "bench.py", "--gpus", "2",
          "--steps", "2", "--warmup", "1"],
         cwd=repo, env=env, capture_output=True, text=True, timeout=500)
     assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
