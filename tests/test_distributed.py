@@ -193,6 +193,9 @@ def test_main_sharded_torchrun_cpu(tmp_path):
             timeout=500)
     r = launch()
     if r.returncode != 0:  # rendezvous under load is flaky; retry once
+        db = tmp_path / "coda.sqlite"
+        if db.exists():
+            db.unlink()  # a partial first attempt must not double-log
         r = launch()
     assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
     conn = sqlite3.connect(str(tmp_path / "coda.sqlite"))
