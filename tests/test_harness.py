@@ -240,3 +240,19 @@ def test_storage_flag_cpu(task_dir):
     _run_cli(task_dir, ["--method", "coda", "--iters", "2", "--seeds", "1",
                         "--no-mlflow", "--chunk-size", "64",
                         "--storage", "bf16"])
+
+
+def test_bench_rejects_gpu_count_mismatch():
+    """bench.py must fail loudly when --gpus disagrees with the launch
+    (single-process with --gpus 8 printed a fake rank count in r01)."""
+    import subprocess
+    import sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    env.pop("WORLD_SIZE", None)
+    r = subprocess.run([sys.executable, "bench.py", "--gpus", "8",
+                        "--steps", "1", "--warmup", "0"],
+                       cwd=repo, env=env, capture_output=True, text=True,
+                       timeout=120)
+    assert r.returncode != 0
+    assert "WORLD_SIZE" in (r.stdout + r.stderr)
