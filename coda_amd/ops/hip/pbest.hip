@@ -1027,126 +1027,6 @@ beta_row_tables_kernel(const float* __restrict__ alpha_col,  // (H,)
 }
 
 
-// Commit one class row's refreshed curves into ALL table mirrors in
-// two launches (the torch chain was ~13 graph nodes of index_copy /
-// cast / mul per label). A: per-point H-sums (s_base, dall). B: wave
-// per (h, v) curve writing EG / eg16 / delta / delta16 / egw.
-__global__ void __launch_bounds__(BLOCK)
-row_sums_commit_kernel(const float* __restrict__ lc,   // (H, 2, P)
-                       const long* __restrict__ y,     // (1,)
-                       float* __restrict__ s_base,     // (C, P)
-                       float* __restrict__ dall,       // (C, P) or null
-                       int H) {
-    const int p = threadIdx.x;
-    if (p >= P_POINTS || blockIdx.x != 0) return;
-    float s0 = 0.f, s1 = 0.f;
-    for (int h = 0; h < H; ++h) {
-        const float l0 = lc[((size_t)h * 2) * P_POINTS + p];
-        const float l1 = lc[((size_t)h * 2 + 1) * P_POINTS + p];
-        s0 += l0;
-        s1 += l1 - l0;
-    }
-    s_base[(size_t)y[0] * P_POINTS + p] = s0;
-    if (dall) dall[(size_t)y[0] * P_POINTS + p] = s1;
-}
-
-__global__ void __launch_bounds__(BLOCK)
-row_tables_commit_kernel(const float* __restrict__ eg,   // (H, 2, P)
-                         const float* __restrict__ lc,   // (H, 2, P)
-                         const long* __restrict__ y,     // (1,)
-                         const float* __restrict__ w,    // (P,)
-                         const float* __restrict__ s_base,  // (C, P)
-                         float* __restrict__ EG,            // (C, H, 2, P)
-                         hip_bfloat16* __restrict__ eg16,   // (C,2H,P)|0
-                         float* __restrict__ delta,         // (C, H, P)
-                         _Float16* __restrict__ delta16,    // (C,H,P)|0
-                         hip_bfloat16* __restrict__ egw,    // (C,2H,P)|0
-                         int H) {
-    const int q = blockIdx.x * ROWS_PER_BLOCK + (threadIdx.x >> 6);
-    if (q >= 2 * H) return;
-    const int h = q >> 1, v = q & 1;
-    const int lane = threadIdx.x & 63;
-    const long yy = y[0];
-    const size_t src = (size_t)q * P_POINTS + lane * PTS_PER_LANE;
-    const size_t sb = (size_t)yy * P_POINTS + lane * PTS_PER_LANE;
-    const float4 egv = *reinterpret_cast<const float4*>(eg + src);
-    *reinterpret_cast<float4*>(
-        EG + ((size_t)yy * H * 2) * P_POINTS + src) = egv;
-    const float e[4] = {egv.x, egv.y, egv.z, egv.w};
-    if (eg16) {
-        ushort4 o;
-        unsigned short* op = reinterpret_cast<unsigned short*>(&o);
-#pragma unroll
-        for (int j = 0; j < 4; ++j)
-            op[j] = hip_bfloat16(e[j]).data;
-        *reinterpret_cast<ushort4*>(
-            eg16 + ((size_t)yy * 2 * H) * P_POINTS + src) = o;
-    }
-    if (egw) {
-        const float4 sv = *reinterpret_cast<const float4*>(s_base + sb);
-        const float4 wv = *reinterpret_cast<const float4*>(
-            w + lane * PTS_PER_LANE);
-        const float s[4] = {sv.x, sv.y, sv.z, sv.w};
-        const float ww[4] = {wv.x, wv.y, wv.z, wv.w};
-        ushort4 o;
-        unsigned short* op = reinterpret_cast<unsigned short*>(&o);
-#pragma unroll
-        for (int j = 0; j < 4; ++j)
-            op[j] = hip_bfloat16(e[j] * exp2f(s[j]) * ww[j]).data;
-        *reinterpret_cast<ushort4*>(
-            egw + ((size_t)yy * 2 * H) * P_POINTS + src) = o;
-    }
-    if (v == 1) {   // delta rows need both variants: the v==1 wave owns
-        const size_t s0 = (size_t)(q - 1) * P_POINTS
-                        + lane * PTS_PER_LANE;
-        const float4 l0 = *reinterpret_cast<const float4*>(lc + s0);
-        const float4 l1 = *reinterpret_cast<const float4*>(lc + src);
-        float d[4] = {l1.x - l0.x, l1.y - l0.y, l1.z - l0.z,
-                      l1.w - l0.w};
-        const size_t dst = ((size_t)yy * H + h) * P_POINTS
-                         + lane * PTS_PER_LANE;
-        *reinterpret_cast<float4*>(delta + dst) =
-            make_float4(d[0], d[1], d[2], d[3]);
-        if (delta16) {
-            _Float16 o16[4];
-#pragma unroll
-            for (int j = 0; j < 4; ++j) o16[j] = (_Float16)d[j];
-            *reinterpret_cast<ushort4*>(delta16 + dst)
-                = *reinterpret_cast<ushort4*>(o16);
-        }
-    }
-}
-
-// Fused mixture entropy: mixture0[h] = sum_c pi[c]*rows[c,h];
-// H0 = -sum_h m log2 m (clamped). One block; H <= 1024.
-__global__ void __launch_bounds__(BLOCK)
-mixture_h0_kernel(const float* __restrict__ rows,  // (C, H)
-                  const float* __restrict__ pi,    // (C,)
-                  float* __restrict__ mixture0,    // (H,)
-                  float* __restrict__ h0,          // (1,)
-                  int C, int H) {
-    if (blockIdx.x != 0) return;
-    __shared__ float part[BLOCK / 64];
-    float ent = 0.f;
-    for (int h = threadIdx.x; h < H; h += BLOCK) {
-        float m = 0.f;
-        for (int c = 0; c < C; ++c)
-            m += pi[c] * rows[(size_t)c * H + h];
-        mixture0[h] = m;
-        const float mc = fmaxf(m, 1e-12f);
-        ent += -mc * __log2f(mc);
-    }
-    ent = wave_reduce_sum(ent);
-    const int wave = threadIdx.x >> 6;
-    if ((threadIdx.x & 63) == 0) part[wave] = ent;
-    __syncthreads();
-    if (threadIdx.x == 0) {
-        float t = 0.f;
-        for (int i = 0; i < BLOCK / 64; ++i) t += part[i];
-        h0[0] = t;
-    }
-}
-
 // Rank-1 pi_hat increment: out[n] = sum_h preds[h, n, cls[h]] - the exact
 // per-label posterior-marginal change (only Dirichlet row true_class
 // moves; coda/coda.py:316-317). One thread per point; consecutive
@@ -1901,52 +1781,6 @@ torch::Tensor pi_marginal(torch::Tensor adjusted, torch::Tensor row_sums) {
     return out;
 }
 
-
-void row_tables_commit(torch::Tensor eg, torch::Tensor lc,
-                       torch::Tensor y, torch::Tensor w,
-                       torch::Tensor EG, torch::Tensor eg16,
-                       torch::Tensor delta, torch::Tensor delta16,
-                       torch::Tensor egw, torch::Tensor s_base,
-                       torch::Tensor dall) {
-    const int H = eg.size(0);
-    auto stream = c10::hip::getCurrentHIPStream();
-    hipLaunchKernelGGL(row_sums_commit_kernel, dim3(1), dim3(BLOCK), 0,
-                       stream.stream(), lc.data_ptr<float>(),
-                       y.data_ptr<long>(), s_base.data_ptr<float>(),
-                       dall.defined() ? dall.data_ptr<float>() : nullptr,
-                       H);
-    hipLaunchKernelGGL(
-        row_tables_commit_kernel,
-        dim3((2 * H + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK),
-        dim3(BLOCK), 0, stream.stream(), eg.data_ptr<float>(),
-        lc.data_ptr<float>(), y.data_ptr<long>(), w.data_ptr<float>(),
-        s_base.data_ptr<float>(), EG.data_ptr<float>(),
-        eg16.defined()
-            ? reinterpret_cast<hip_bfloat16*>(eg16.data_ptr()) : nullptr,
-        delta.data_ptr<float>(),
-        delta16.defined()
-            ? reinterpret_cast<_Float16*>(delta16.data_ptr()) : nullptr,
-        egw.defined()
-            ? reinterpret_cast<hip_bfloat16*>(egw.data_ptr()) : nullptr,
-        H);
-    C10_HIP_CHECK(hipGetLastError());
-}
-
-std::vector<torch::Tensor> mixture_h0(torch::Tensor rows,
-                                      torch::Tensor pi) {
-    const int C = rows.size(0), H = rows.size(1);
-    TORCH_CHECK(H <= 1024, "mixture_h0 kernel caps at H=1024");
-    auto mixture0 = torch::empty({H}, rows.options());
-    auto h0 = torch::empty({1}, rows.options());
-    auto stream = c10::hip::getCurrentHIPStream();
-    hipLaunchKernelGGL(mixture_h0_kernel, dim3(1), dim3(BLOCK), 0,
-                       stream.stream(), rows.data_ptr<float>(),
-                       pi.data_ptr<float>(), mixture0.data_ptr<float>(),
-                       h0.data_ptr<float>(), C, H);
-    C10_HIP_CHECK(hipGetLastError());
-    return {mixture0, h0};
-}
-
 void register_pair_ops(pybind11::module_& m);  // pair.hip (v3 engine)
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -1990,8 +1824,4 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "rank-1 pi_hat increment: sum_h preds[h, :, cls_h]");
     m.def("pi_marginal", &pi_marginal,
           "pi[c] = sum_n adjusted[n,c]/clamp(rowsum[n]) in one pass");
-    m.def("row_tables_commit", &row_tables_commit,
-          "commit one refreshed class row into all table mirrors");
-    m.def("mixture_h0", &mixture_h0,
-          "fused mixture0 + log2 entropy from posterior rows");
 }

@@ -476,8 +476,7 @@ class CODA(ModelSelector):
         t = self._tables
         ps, _ = self._pairs_static
         rows, pi = self._g_rows, self._g_pi
-        mixture0, h0t = ops._ext.mixture_h0(rows, pi)
-        H0 = h0t[0]
+        mixture0, H0 = ops.mixture_entropy(rows, pi)
         A16 = ops._ext.pair_dsum_es(t.delta16, t.dall, ps.pair_c,
                                     ps.pair_neg, ps.seg_off, ps.seg_h)
         h_after = ops._ext.pair_gemm_entropy(
@@ -638,10 +637,7 @@ class CODA(ModelSelector):
         else:
             pi = (1.0 / self._row_sums.clamp_min(1e-12)) @ self._adjusted
         self._g_pi.copy_(pi / pi.sum())
-        # refresh the labeled class's table row (v2 tables): the curve
-        # kernel + one fused commit into every mirror (EG/eg16/delta/
-        # delta16/s_base/dall/egw - was ~13 graph nodes of index_copy/
-        # cast/mul chains)
+        # refresh the labeled class's table row (v2 tables)
         if self._tables is not None:
             row = self.dirichlets.index_select(1, y_t).squeeze(1)  # (Hl,C)
             a_col = row.gather(
@@ -649,9 +645,25 @@ class CODA(ModelSelector):
             b_col = (row.sum(1) - a_col).contiguous()
             eg, lc = ops._ext.beta_row_tables(a_col, b_col, 1.0)
             t = self._tables
-            ops._ext.row_tables_commit(
-                eg, lc, y_t, t.weights, t.EG, t.eg16, t.delta,
-                t.delta16, t.egw, t.s_base, t.dall)
+            t.EG.index_copy_(0, y_t, eg.unsqueeze(0))
+            if t.eg16 is not None:
+                t.eg16.index_copy_(
+                    0, y_t, eg.reshape(1, 2 * self.Hl, -1)
+                    .to(torch.bfloat16))
+            t.delta.index_copy_(0, y_t, (lc[:, 1] - lc[:, 0]).unsqueeze(0))
+            t.s_base.index_copy_(0, y_t, lc[:, 0].sum(0).unsqueeze(0))
+            if t.dall is not None:
+                t.dall.index_copy_(
+                    0, y_t, t.delta.index_select(0, y_t).sum(1))
+            if t.egw is not None:
+                esb = torch.exp2(t.s_base.index_select(0, y_t)) \
+                    * t.weights                             # (1, P)
+                t.egw.index_copy_(
+                    0, y_t, (eg.reshape(1, 2 * self.Hl, -1)
+                             * esb.unsqueeze(1)).to(torch.bfloat16))
+                t.delta16.index_copy_(
+                    0, y_t, t.delta.index_select(0, y_t)
+                    .to(torch.float16))
         # posterior rows: add_label moves only Dirichlet row y, so only
         # class y's Beta column - hence only pbest row y - changes.
         # _g_rows was seeded with the full (C, Hl) rows at graph init;
