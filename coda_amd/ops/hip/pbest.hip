@@ -1203,6 +1203,92 @@ pi_marginal_kernel(const float* __restrict__ adjusted,  // (N, C)
     }
 }
 
+
+// pi marginal, vector shapes (C % 4 == 0, C <= 1024): the single-kernel
+// schedule above is ATOMIC-bound at the headline shape (1536 blocks x
+// 250 lanes x 4 atomicAdds onto 1000 floats = 165 us measured vs a
+// ~25 us streaming floor; scripts/pi_marginal_probe.py has the sweep).
+// Three-kernel deterministic replacement, 36 us measured:
+//   1. slab partials  (G, Cpad)  - thread tid owns one float4 of
+//      columns, 8-row-unrolled stream over the block's row slab,
+//      plain coalesced float4 stores (no atomics);
+//   2. 16-way G-reduce (16, Cpad) - fixed slab boundaries;
+//   3. final 16-row sum -> out    - fixed order, so the whole chain is
+//      run-to-run DETERMINISTIC (the atomic schedule was not).
+__global__ void __launch_bounds__(BLOCK)
+pi_marginal_part_kernel(const float* __restrict__ adjusted,  // (N, C)
+                        const float* __restrict__ row_sums,  // (N,)
+                        float* __restrict__ partial,         // (G, Cpad)
+                        long long N, int C, int Cpad) {
+    const int tid = threadIdx.x;
+    const long long rows_per_block = (N + gridDim.x - 1) / gridDim.x;
+    const long long n0 = (long long)blockIdx.x * rows_per_block;
+    const long long n1 = min(n0 + rows_per_block, N);
+    const int c4 = tid * 4;
+    if (c4 >= C) return;
+    float4 acc = {0.f, 0.f, 0.f, 0.f};
+    float4 acc2 = {0.f, 0.f, 0.f, 0.f};
+    long long n = n0;
+    for (; n + 7 < n1; n += 8) {
+        float inv[8];
+        float4 v[8];
+#pragma unroll
+        for (int r = 0; r < 8; ++r)
+            inv[r] = 1.0f / fmaxf(row_sums[n + r], 1e-12f);
+#pragma unroll
+        for (int r = 0; r < 8; ++r)
+            v[r] = *reinterpret_cast<const float4*>(
+                adjusted + (n + r) * C + c4);
+#pragma unroll
+        for (int r = 0; r < 8; r += 2) {
+            acc.x += v[r].x * inv[r];  acc2.x += v[r + 1].x * inv[r + 1];
+            acc.y += v[r].y * inv[r];  acc2.y += v[r + 1].y * inv[r + 1];
+            acc.z += v[r].z * inv[r];  acc2.z += v[r + 1].z * inv[r + 1];
+            acc.w += v[r].w * inv[r];  acc2.w += v[r + 1].w * inv[r + 1];
+        }
+    }
+    for (; n < n1; ++n) {
+        const float inv = 1.0f / fmaxf(row_sums[n], 1e-12f);
+        const float4 v = *reinterpret_cast<const float4*>(
+            adjusted + n * C + c4);
+        acc.x += v.x * inv; acc.y += v.y * inv;
+        acc.z += v.z * inv; acc.w += v.w * inv;
+    }
+    acc.x += acc2.x; acc.y += acc2.y; acc.z += acc2.z; acc.w += acc2.w;
+    *reinterpret_cast<float4*>(
+        partial + (long long)blockIdx.x * Cpad + c4) = acc;
+}
+
+// rows-reduce: out[oy, c] = sum over this block.y's fixed slice of
+// in[g, c].  Used twice: (G -> 16 slices) then (16 -> 1, oy = 0).
+__global__ void __launch_bounds__(BLOCK)
+pi_marginal_reduce_kernel(const float* __restrict__ in,   // (G, Cpad)
+                          float* __restrict__ out,        // (gridDim.y, ostride)
+                          int G, int C, int Cpad, int ostride) {
+    const int c4 = (blockIdx.x * BLOCK + threadIdx.x) * 4;
+    if (c4 >= C) return;
+    const int gs = (G + gridDim.y - 1) / gridDim.y;
+    const int g0 = blockIdx.y * gs, g1 = min(g0 + gs, G);
+    float4 acc = {0.f, 0.f, 0.f, 0.f};
+    float4 acc2 = {0.f, 0.f, 0.f, 0.f};
+    int g = g0;
+    for (; g + 1 < g1; g += 2) {
+        const float4 a = *reinterpret_cast<const float4*>(
+            in + (long long)g * Cpad + c4);
+        const float4 b = *reinterpret_cast<const float4*>(
+            in + (long long)(g + 1) * Cpad + c4);
+        acc.x += a.x; acc.y += a.y; acc.z += a.z; acc.w += a.w;
+        acc2.x += b.x; acc2.y += b.y; acc2.z += b.z; acc2.w += b.w;
+    }
+    if (g < g1) {
+        const float4 a = *reinterpret_cast<const float4*>(
+            in + (long long)g * Cpad + c4);
+        acc.x += a.x; acc.y += a.y; acc.z += a.z; acc.w += a.w;
+    }
+    *reinterpret_cast<float4*>(out + (long long)blockIdx.y * ostride + c4) =
+        {acc.x + acc2.x, acc.y + acc2.y, acc.z + acc2.z, acc.w + acc2.w};
+}
+
 }  // namespace
 
 // ---------------------------------------------------------------------------
@@ -1769,10 +1855,35 @@ torch::Tensor pi_marginal(torch::Tensor adjusted, torch::Tensor row_sums) {
     const long long N = adjusted.size(0);
     const int C = adjusted.size(1);
     TORCH_CHECK(C <= 8 * BLOCK, "C too large for pi_marginal kernel");
+    auto stream = c10::hip::getCurrentHIPStream();
+    if ((C & 3) == 0 && C <= 4 * BLOCK) {
+        // deterministic two-stage-reduce path (probe: 36 us vs 165 us
+        // for the atomic single-kernel at N=50k, C=1000).  G sized so
+        // the partial stage keeps ~1024 ACTIVE waves streaming: a block
+        // covers ceil(C/4) lanes = awpb active waves.
+        const int awpb = (C / 4 + 63) / 64;
+        const int G = std::min(1024, std::max(256, 1024 / awpb));
+        constexpr int S = 16;  // G-reduce split
+        auto out = torch::empty({C}, adjusted.options());
+        auto work = torch::empty({G + S, C}, adjusted.options());
+        float* partial = work.data_ptr<float>();
+        float* part2 = partial + (long long)G * C;
+        const int cblocks = (C + 4 * BLOCK - 1) / (4 * BLOCK);
+        hipLaunchKernelGGL(pi_marginal_part_kernel, dim3(G), dim3(BLOCK),
+                           0, stream.stream(), adjusted.data_ptr<float>(),
+                           row_sums.data_ptr<float>(), partial, N, C, C);
+        hipLaunchKernelGGL(pi_marginal_reduce_kernel, dim3(cblocks, S),
+                           dim3(BLOCK), 0, stream.stream(), partial,
+                           part2, G, C, C, C);
+        hipLaunchKernelGGL(pi_marginal_reduce_kernel, dim3(cblocks, 1),
+                           dim3(BLOCK), 0, stream.stream(), part2,
+                           out.data_ptr<float>(), S, C, C, C);
+        C10_HIP_CHECK(hipGetLastError());
+        return out;
+    }
     auto out = torch::zeros({C}, adjusted.options());
     const int blocks = 1536;  // 6 blocks/CU: enough row-slabs in flight
                               // to cover HBM latency (512 ran at 1 TB/s)
-    auto stream = c10::hip::getCurrentHIPStream();
     hipLaunchKernelGGL(pi_marginal_kernel, dim3(blocks), dim3(BLOCK), 0,
                        stream.stream(), adjusted.data_ptr<float>(),
                        row_sums.data_ptr<float>(), out.data_ptr<float>(),
