@@ -358,6 +358,71 @@ pbest_kernel(const float* __restrict__ alpha, const float* __restrict__ beta,
         out[(size_t)r * H + h] = s_pb[h] * inv;
 }
 
+// Single-row P(best): R = 1 gives the generic kernel ONE wave (one
+// workgroup, 92 us at H=128 - the per-label incremental posterior-row
+// refresh is exactly this shape).  Here all ROWS_PER_BLOCK waves
+// cooperate on the one row by windowing the model axis: wave w owns
+// models [w*Hc, w*Hc+Hc), pass-A slog2 partials combine through LDS
+// (the in-workgroup analogue of the phase1/phase2 global coupling),
+// pass B integrates each window against the combined term.  Identical
+// math to the two-pass split, one launch.
+__global__ void CODA_LB
+pbest_row_kernel(const float* __restrict__ alpha,  // (H,)
+                 const float* __restrict__ beta,   // (H,)
+                 float* __restrict__ out,          // (H,)
+                 int H) {
+    extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+    const int Hc = (H + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
+    double* lnB_all = reinterpret_cast<double*>(smem_raw);
+    float* f_all = reinterpret_cast<float*>(lnB_all + ROWS_PER_BLOCK * Hc);
+    float* slds = f_all + ROWS_PER_BLOCK * 3 * Hc;     // (4, P) partials
+    float* tot_lds = slds + ROWS_PER_BLOCK * P_POINTS; // (4,)
+
+    for (int idx = threadIdx.x; idx < ROWS_PER_BLOCK * Hc; idx += BLOCK) {
+        const int w = idx / Hc, h = idx - w * Hc;
+        const int hg = w * Hc + h;
+        if (hg >= H) continue;
+        float a = alpha[hg], b = beta[hg];
+        f_all[w * 3 * Hc + h] = a;
+        f_all[w * 3 * Hc + Hc + h] = b;
+        lnB_all[w * Hc + h] = (lgamma((double)a) + lgamma((double)b)
+                            - lgamma((double)a + (double)b))
+                            * 1.4426950408889634;
+    }
+    __syncthreads();
+
+    const int w = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    const int Hcur = max(0, min(Hc, H - w * Hc));
+    LaneGrid g;
+    g.init(lane);
+    float slog2[PTS_PER_LANE];
+    // empty window: pass A leaves the identity (log2 1 = 0) partial
+    pbest_pass_a(f_all + w * 3 * Hc, f_all + w * 3 * Hc + Hc,
+                 lnB_all + w * Hc, Hcur, g, lane, slog2);
+#pragma unroll
+    for (int j = 0; j < PTS_PER_LANE; ++j)
+        slds[w * P_POINTS + lane * PTS_PER_LANE + j] = slog2[j];
+    __syncthreads();
+#pragma unroll
+    for (int j = 0; j < PTS_PER_LANE; ++j) {
+        float s = 0.f;
+        for (int q = 0; q < ROWS_PER_BLOCK; ++q)
+            s += slds[q * P_POINTS + lane * PTS_PER_LANE + j];
+        slog2[j] = s;
+    }
+    float* s_pb = f_all + w * 3 * Hc + 2 * Hc;
+    const float part = pbest_pass_b(f_all + w * 3 * Hc,
+                                    f_all + w * 3 * Hc + Hc,
+                                    lnB_all + w * Hc, s_pb,
+                                    Hcur, g, lane, slog2);
+    if (lane == 0) tot_lds[w] = part;
+    __syncthreads();
+    const float total = tot_lds[0] + tot_lds[1] + tot_lds[2] + tot_lds[3];
+    const float inv = 1.0f / fmaxf(total, kEps);
+    for (int h = lane; h < Hcur; h += 64)
+        out[w * Hc + h] = s_pb[h] * inv;
+}
+
 // ---------------------------------------------------------------------------
 // Kernel 2: fused hypothetical-update P(best) + entropy epilogue for EIG.
 // One wave per (candidate b, hypothesized class c) row:
@@ -1312,10 +1377,24 @@ torch::Tensor pbest_from_beta(torch::Tensor alpha, torch::Tensor beta,
     const int R = alpha.size(0), H = alpha.size(1);
     auto out = torch::empty_like(alpha);
     if (R == 0) return out;
+    auto stream = c10::hip::getCurrentHIPStream();
+    if (R == 1) {
+        // one row would give the generic kernel a single wave; the row
+        // kernel windows the model axis across the whole workgroup
+        const int Hc = (H + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
+        const size_t smem = lds_bytes(Hc)
+            + (ROWS_PER_BLOCK * P_POINTS + ROWS_PER_BLOCK) * sizeof(float);
+        TORCH_CHECK(smem <= 160 * 1024, "H too large for LDS: ", H);
+        hipLaunchKernelGGL(pbest_row_kernel, dim3(1), dim3(BLOCK), smem,
+                           stream.stream(), alpha.data_ptr<float>(),
+                           beta.data_ptr<float>(), out.data_ptr<float>(),
+                           H);
+        C10_HIP_CHECK(hipGetLastError());
+        return out;
+    }
     const size_t smem = lds_bytes(H);
     TORCH_CHECK(smem <= 160 * 1024, "H too large for LDS: ", H);
     const int blocks = (R + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
-    auto stream = c10::hip::getCurrentHIPStream();
     hipLaunchKernelGGL(pbest_kernel, dim3(blocks), dim3(BLOCK), smem,
                        stream.stream(), alpha.data_ptr<float>(),
                        beta.data_ptr<float>(), out.data_ptr<float>(), R, H);
