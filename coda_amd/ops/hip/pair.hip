@@ -600,6 +600,113 @@ pair_entropy_wide_cls_kernel(const hip_bfloat16* __restrict__ m,  // (K, 2H)
     if (lane == 0) h_after[k] = ent;
 }
 
+// ---------------------------------------------------------------------
+// Fused acquisition epilogue.  After pair_eig_finalize the graphed
+// get_next tail was ~10 small torch launches (add + where(-inf) + max
+// + isclose chain + sum + qbuf copy + stack/copy, ~120 us/step at
+// B=50k): three kernels replace it.
+//   K1: qbuf = active ? h0 + q0 : -inf, per-block (max, first-idx)
+//   K2: one block combines the partials -> out[0..1]
+//   K3: tie count |q - best| <= atol + rtol*|best| over active
+//       candidates (torch.isclose semantics), f64 atomic into out[2]
+//       (integer-valued doubles: exact, order-independent)
+// ---------------------------------------------------------------------
+#define ACQ_GRID 256
+
+__global__ void __launch_bounds__(BLOCK)
+acq_select_part_kernel(const float* __restrict__ q0,
+                       const float* __restrict__ h0,     // scalar (1,)
+                       const bool* __restrict__ active,
+                       float* __restrict__ qbuf,
+                       float* __restrict__ pmax, int* __restrict__ pidx,
+                       int B) {
+    const int per = (B + gridDim.x - 1) / gridDim.x;
+    const int b0 = blockIdx.x * per;
+    const int b1 = min(b0 + per, B);
+    const float h = h0[0];
+    float best = -INFINITY;
+    int bidx = -1;
+    for (int i = b0 + threadIdx.x; i < b1; i += BLOCK) {
+        const float q = active[i] ? h + q0[i] : -INFINITY;
+        qbuf[i] = q;
+        if (q > best) { best = q; bidx = i; }   // strict: first index wins
+    }
+    __shared__ float sv[BLOCK];
+    __shared__ int si[BLOCK];
+    sv[threadIdx.x] = best;
+    si[threadIdx.x] = bidx;
+    __syncthreads();
+    for (int s = BLOCK / 2; s > 0; s >>= 1) {
+        if (threadIdx.x < s) {
+            const float ov = sv[threadIdx.x + s];
+            const int oi = si[threadIdx.x + s];
+            if (ov > sv[threadIdx.x]
+                || (ov == sv[threadIdx.x] && oi != -1
+                    && (si[threadIdx.x] == -1 || oi < si[threadIdx.x]))) {
+                sv[threadIdx.x] = ov;
+                si[threadIdx.x] = oi;
+            }
+        }
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+        pmax[blockIdx.x] = sv[0];
+        pidx[blockIdx.x] = si[0];
+    }
+}
+
+__global__ void __launch_bounds__(ACQ_GRID)
+acq_select_combine_kernel(const float* __restrict__ pmax,
+                          const int* __restrict__ pidx,
+                          double* __restrict__ out) {   // (3,) f64
+    __shared__ float sv[ACQ_GRID];
+    __shared__ int si[ACQ_GRID];
+    sv[threadIdx.x] = pmax[threadIdx.x];
+    si[threadIdx.x] = pidx[threadIdx.x];
+    __syncthreads();
+    for (int s = ACQ_GRID / 2; s > 0; s >>= 1) {
+        if (threadIdx.x < s) {
+            const float ov = sv[threadIdx.x + s];
+            const int oi = si[threadIdx.x + s];
+            if (ov > sv[threadIdx.x]
+                || (ov == sv[threadIdx.x] && oi != -1
+                    && (si[threadIdx.x] == -1 || oi < si[threadIdx.x]))) {
+                sv[threadIdx.x] = ov;
+                si[threadIdx.x] = oi;
+            }
+        }
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+        out[0] = (double)sv[0];
+        out[1] = (double)si[0];
+        out[2] = 0.0;
+    }
+}
+
+__global__ void __launch_bounds__(BLOCK)
+acq_select_ties_kernel(const float* __restrict__ qbuf,
+                       const bool* __restrict__ active,
+                       double* __restrict__ out, int B) {
+    const float best = (float)out[0];
+    const float thr = 1e-8f + 1e-8f * fabsf(best);
+    const int per = (B + gridDim.x - 1) / gridDim.x;
+    const int b0 = blockIdx.x * per;
+    const int b1 = min(b0 + per, B);
+    int cnt = 0;
+    for (int i = b0 + threadIdx.x; i < b1; i += BLOCK)
+        cnt += (active[i] && fabsf(qbuf[i] - best) <= thr) ? 1 : 0;
+    __shared__ int sc[BLOCK];
+    sc[threadIdx.x] = cnt;
+    __syncthreads();
+    for (int s = BLOCK / 2; s > 0; s >>= 1) {
+        if (threadIdx.x < s) sc[threadIdx.x] += sc[threadIdx.x + s];
+        __syncthreads();
+    }
+    if (threadIdx.x == 0 && sc[0] > 0)
+        atomicAdd(out + 2, (double)sc[0]);
+}
+
 // MFMA layout probe (correctness insurance, not a production op):
 // C (16,16) = A (16,32) x B stored row-major as BT (16 cols x 32 k).
 __global__ void mfma_probe_kernel(const float* __restrict__ a,   // (16,32)
@@ -772,6 +879,38 @@ torch::Tensor pair_eig_finalize(torch::Tensor h_after,
     return q;
 }
 
+void acq_select(torch::Tensor q0, torch::Tensor h0, torch::Tensor active,
+                torch::Tensor qbuf, torch::Tensor out) {
+    TORCH_CHECK(q0.is_cuda() && q0.dtype() == torch::kFloat32
+                && q0.is_contiguous(), "q0 must be contiguous fp32");
+    TORCH_CHECK(h0.dtype() == torch::kFloat32 && h0.numel() == 1);
+    TORCH_CHECK(active.dtype() == torch::kBool && active.is_contiguous()
+                && active.numel() == q0.numel());
+    TORCH_CHECK(qbuf.dtype() == torch::kFloat32 && qbuf.is_contiguous()
+                && qbuf.numel() == q0.numel());
+    TORCH_CHECK(out.dtype() == torch::kFloat64 && out.is_contiguous()
+                && out.numel() == 3);
+    const int B = q0.numel();
+    auto pmax = torch::empty({ACQ_GRID}, q0.options());
+    auto pidx = torch::empty({ACQ_GRID},
+                             q0.options().dtype(torch::kInt32));
+    auto stream = c10::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(pairops::acq_select_part_kernel, dim3(ACQ_GRID),
+                       dim3(BLOCK), 0, stream.stream(),
+                       q0.data_ptr<float>(), h0.data_ptr<float>(),
+                       active.data_ptr<bool>(), qbuf.data_ptr<float>(),
+                       pmax.data_ptr<float>(), pidx.data_ptr<int>(), B);
+    hipLaunchKernelGGL(pairops::acq_select_combine_kernel, dim3(1),
+                       dim3(ACQ_GRID), 0, stream.stream(),
+                       pmax.data_ptr<float>(), pidx.data_ptr<int>(),
+                       out.data_ptr<double>());
+    hipLaunchKernelGGL(pairops::acq_select_ties_kernel, dim3(ACQ_GRID),
+                       dim3(BLOCK), 0, stream.stream(),
+                       qbuf.data_ptr<float>(), active.data_ptr<bool>(),
+                       out.data_ptr<double>(), B);
+    C10_HIP_CHECK(hipGetLastError());
+}
+
 torch::Tensor pair_gemm_entropy_cls(torch::Tensor a16, torch::Tensor egw,
                                     torch::Tensor pair_b,
                                     torch::Tensor pair_c,
@@ -838,6 +977,9 @@ void register_pair_ops(pybind11::module_& m) {
           "v3 wide-H pairing GEMM + cls-based entropy -> (K,)");
     m.def("pair_eig_finalize", &pair_eig_finalize,
           "v3 per-candidate EIG assembly (deterministic) -> (B,)");
+    m.def("acq_select", &acq_select,
+          "fused acquisition epilogue: qbuf = active ? h0+q0 : -inf; "
+          "out(f64[3]) = [masked max, first argmax, isclose tie count]");
     m.def("mfma_probe", &mfma_probe,
           "16x16x32 bf16 MFMA fragment-layout probe");
 }
