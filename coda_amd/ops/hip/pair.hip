@@ -658,7 +658,8 @@ acq_select_part_kernel(const float* __restrict__ q0,
 __global__ void __launch_bounds__(ACQ_GRID)
 acq_select_combine_kernel(const float* __restrict__ pmax,
                           const int* __restrict__ pidx,
-                          double* __restrict__ out) {   // (3,) f64
+                          double* __restrict__ out,     // (3,) f64
+                          int* __restrict__ tbuf) {
     __shared__ float sv[ACQ_GRID];
     __shared__ int si[ACQ_GRID];
     sv[threadIdx.x] = pmax[threadIdx.x];
@@ -681,13 +682,16 @@ acq_select_combine_kernel(const float* __restrict__ pmax,
         out[0] = (double)sv[0];
         out[1] = (double)si[0];
         out[2] = 0.0;
+        tbuf[0] = 0;                       // reset the tie-slot counter
     }
 }
 
 __global__ void __launch_bounds__(BLOCK)
 acq_select_ties_kernel(const float* __restrict__ qbuf,
                        const bool* __restrict__ active,
-                       double* __restrict__ out, int B) {
+                       double* __restrict__ out,
+                       int* __restrict__ tbuf,   // [0]=n, [1..cap]=idx
+                       int cap, int B) {
     const float best = (float)out[0];
     const float thr = 1e-8f + 1e-8f * fabsf(best);
     const int per = (B + gridDim.x - 1) / gridDim.x;
@@ -695,7 +699,13 @@ acq_select_ties_kernel(const float* __restrict__ qbuf,
     const int b1 = min(b0 + per, B);
     int cnt = 0;
     for (int i = b0 + threadIdx.x; i < b1; i += BLOCK)
-        cnt += (active[i] && fabsf(qbuf[i] - best) <= thr) ? 1 : 0;
+        if (active[i] && fabsf(qbuf[i] - best) <= thr) {
+            ++cnt;
+            // unordered slots; the host sorts the <= cap indices back
+            // to ascending before the seeded tie-break choice
+            const int slot = atomicAdd(tbuf, 1);
+            if (slot < cap) tbuf[1 + slot] = i;
+        }
     __shared__ int sc[BLOCK];
     sc[threadIdx.x] = cnt;
     __syncthreads();
@@ -880,7 +890,8 @@ torch::Tensor pair_eig_finalize(torch::Tensor h_after,
 }
 
 void acq_select(torch::Tensor q0, torch::Tensor h0, torch::Tensor active,
-                torch::Tensor qbuf, torch::Tensor out) {
+                torch::Tensor qbuf, torch::Tensor out,
+                torch::Tensor ties) {
     TORCH_CHECK(q0.is_cuda() && q0.dtype() == torch::kFloat32
                 && q0.is_contiguous(), "q0 must be contiguous fp32");
     TORCH_CHECK(h0.dtype() == torch::kFloat32 && h0.numel() == 1);
@@ -890,6 +901,9 @@ void acq_select(torch::Tensor q0, torch::Tensor h0, torch::Tensor active,
                 && qbuf.numel() == q0.numel());
     TORCH_CHECK(out.dtype() == torch::kFloat64 && out.is_contiguous()
                 && out.numel() == 3);
+    TORCH_CHECK(ties.dtype() == torch::kInt32 && ties.is_contiguous()
+                && ties.numel() >= 2, "ties buffer too small");
+    const int cap = ties.numel() - 1;
     const int B = q0.numel();
     auto pmax = torch::empty({ACQ_GRID}, q0.options());
     auto pidx = torch::empty({ACQ_GRID},
@@ -903,11 +917,12 @@ void acq_select(torch::Tensor q0, torch::Tensor h0, torch::Tensor active,
     hipLaunchKernelGGL(pairops::acq_select_combine_kernel, dim3(1),
                        dim3(ACQ_GRID), 0, stream.stream(),
                        pmax.data_ptr<float>(), pidx.data_ptr<int>(),
-                       out.data_ptr<double>());
+                       out.data_ptr<double>(), ties.data_ptr<int>());
     hipLaunchKernelGGL(pairops::acq_select_ties_kernel, dim3(ACQ_GRID),
                        dim3(BLOCK), 0, stream.stream(),
                        qbuf.data_ptr<float>(), active.data_ptr<bool>(),
-                       out.data_ptr<double>(), B);
+                       out.data_ptr<double>(), ties.data_ptr<int>(),
+                       cap, B);
     C10_HIP_CHECK(hipGetLastError());
 }
 

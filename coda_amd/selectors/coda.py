@@ -497,21 +497,28 @@ class CODA(ModelSelector):
         # seeded host random.choice either way (_acq_result).
         ops._ext.acq_select(q0.contiguous(), H0.reshape(1).contiguous(),
                             self._active_mask, self._acq_qbuf,
-                            self._acq_out)
+                            self._acq_out, self._acq_ties)
 
     def _acq_result(self):
         bv, bi, nt = self._acq_out.cpu().tolist()
         bi, nt = int(bi), int(nt)
         if nt > 1:
             # same tie semantics as the eager path: active candidates
-            # ascend by point id in both orderings
-            q = self._acq_qbuf
-            ties = (torch.isclose(q, q.max(), rtol=1e-8)
-                    & self._active_mask)
-            pos = random.choice(
-                torch.nonzero(ties, as_tuple=True)[0].tolist())
+            # ascend by point id in both orderings.  The fused epilogue
+            # collected the tie positions already (unordered slots);
+            # sorting restores the ascending order the seeded
+            # random.choice tie-break expects.
+            if nt < self._acq_ties.numel():
+                pos = random.choice(
+                    sorted(self._acq_ties[1:1 + nt].cpu().tolist()))
+            else:  # > buffer capacity: rescan (never seen in practice)
+                q = self._acq_qbuf
+                ties = (torch.isclose(q, q.max(), rtol=1e-8)
+                        & self._active_mask)
+                pos = random.choice(
+                    torch.nonzero(ties, as_tuple=True)[0].tolist())
             self.stochastic = True
-            return self._pairs_ids_host[pos], float(q[pos])
+            return self._pairs_ids_host[pos], float(self._acq_qbuf[pos])
         return self._pairs_ids_host[bi], bv
 
     def _init_acq_buffers(self):
@@ -521,6 +528,11 @@ class CODA(ModelSelector):
             self._acq_out = torch.zeros(3, dtype=torch.float64,
                                         device=self.device)
             self._acq_qbuf = torch.empty(ps.cand_ids.numel(),
+                                         device=self.device)
+            # fused-epilogue tie-index buffer: [0]=slot counter,
+            # [1:]=unordered tie positions (host-sorted back to
+            # ascending); overflow falls back to the full-scan path
+            self._acq_ties = torch.zeros(8192, dtype=torch.int32,
                                          device=self.device)
 
     def _acq_eligible(self) -> bool:
