@@ -1092,6 +1092,96 @@ beta_row_tables_kernel(const float* __restrict__ alpha_col,  // (H,)
 }
 
 
+// ---------------------------------------------------------------------------
+// Per-label table-row commit (v2/v3 tables).  After beta_row_tables
+// recomputes class y's eg/lc curves, the torch commit chain was ~14
+// small launches (~4.8 us each in-graph: index_copy x5, two 32-thread
+// strided reductions at 10-24 us, exp2/mul/convert kernels).  Three
+// kernels replace the whole chain:
+//   trc_cols   a_col/b_col from the Dirichlet row (the (H, C).sum(1)
+//              was a 24 us 32-thread torch reduce)
+//   trc_sums   s_base[y], dall[y], esb scratch (fixed-order H sums)
+//   trc_rows   EG/eg16/egw/delta/delta16 row writes + conversions
+// y is read from the device tensor so the label graph can replay with
+// a changed class.
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(BLOCK)
+trc_cols_kernel(const float* __restrict__ dir,   // (H, C, C)
+                const long long* __restrict__ y, // (1,)
+                float* __restrict__ a_col,       // (H,)
+                float* __restrict__ b_col,       // (H,)
+                int H, int C) {
+    const int h = blockIdx.x;
+    const long long yy = y[0];
+    const float* row = dir + ((size_t)h * C + yy) * C;
+    float s = 0.f;
+    for (int c = threadIdx.x; c < C; c += BLOCK) s += row[c];
+    __shared__ float ss[BLOCK];
+    ss[threadIdx.x] = s;
+    __syncthreads();
+    for (int k = BLOCK / 2; k > 0; k >>= 1) {
+        if (threadIdx.x < k) ss[threadIdx.x] += ss[threadIdx.x + k];
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+        const float a = row[yy];
+        a_col[h] = a;
+        b_col[h] = ss[0] - a;
+    }
+}
+
+__global__ void __launch_bounds__(P_POINTS)
+trc_sums_kernel(const float* __restrict__ lc,    // (H, 2, P)
+                const float* __restrict__ w,     // (P,)
+                const long long* __restrict__ y,
+                float* __restrict__ s_base,      // (C, P)
+                float* __restrict__ dall,        // (C, P)
+                float* __restrict__ esb,         // (P,) scratch
+                int H) {
+    const int p = threadIdx.x;
+    const long long yy = y[0];
+    float s = 0.f, d = 0.f;
+    for (int h = 0; h < H; ++h) {
+        const float l0 = lc[((size_t)h * 2) * P_POINTS + p];
+        const float l1 = lc[((size_t)h * 2 + 1) * P_POINTS + p];
+        s += l0;
+        d += l1 - l0;
+    }
+    s_base[yy * P_POINTS + p] = s;
+    dall[yy * P_POINTS + p] = d;
+    esb[p] = exp2f(s) * w[p];
+}
+
+__global__ void __launch_bounds__(BLOCK)
+trc_rows_kernel(const float* __restrict__ eg,    // (H, 2, P)
+                const float* __restrict__ lc,    // (H, 2, P)
+                const float* __restrict__ esb,   // (P,)
+                const long long* __restrict__ y,
+                float* __restrict__ EG,          // (C, H, 2, P)
+                float* __restrict__ delta,       // (C, H, P)
+                hip_bfloat16* __restrict__ eg16, // (C, 2H, P)
+                hip_bfloat16* __restrict__ egw,  // (C, 2H, P)
+                _Float16* __restrict__ delta16,  // (C, H, P)
+                int H) {
+    const int i = blockIdx.x * BLOCK + threadIdx.x;  // over (2H, P)
+    if (i >= 2 * H * P_POINTS) return;
+    const int p = i & (P_POINTS - 1);
+    const int q = i >> 8;                            // (h, v)
+    const long long yy = y[0];
+    const float e = eg[i];
+    const size_t rbase = (size_t)yy * 2 * H * P_POINTS;
+    EG[rbase + i] = e;
+    eg16[rbase + i] = hip_bfloat16(e);
+    egw[rbase + i] = hip_bfloat16(e * esb[p]);
+    if (q & 1) {   // one visit per (h, p): the v == 1 thread
+        const int h = q >> 1;
+        const float dv = lc[i] - lc[i - P_POINTS];
+        const size_t dbase = ((size_t)yy * H + h) * P_POINTS + p;
+        delta[dbase] = dv;
+        delta16[dbase] = (_Float16)dv;
+    }
+}
+
 // Rank-1 pi_hat increment: out[n] = sum_h preds[h, n, cls[h]] - the exact
 // per-label posterior-marginal change (only Dirichlet row true_class
 // moves; coda/coda.py:316-317). One thread per point; consecutive
@@ -1879,6 +1969,50 @@ std::vector<torch::Tensor> beta_row_tables(torch::Tensor alpha_col,
 }
 
 
+std::vector<torch::Tensor> table_commit_row(
+        torch::Tensor dirichlets, torch::Tensor y, torch::Tensor EG,
+        torch::Tensor delta, torch::Tensor s_base, torch::Tensor weights,
+        torch::Tensor eg16, torch::Tensor egw, torch::Tensor delta16,
+        torch::Tensor dall, double update_weight) {
+    check_f32_cuda(dirichlets, "dirichlets");
+    TORCH_CHECK(y.scalar_type() == torch::kInt64 && y.numel() == 1);
+    TORCH_CHECK(EG.is_contiguous() && delta.is_contiguous()
+                && s_base.is_contiguous() && weights.is_contiguous()
+                && eg16.is_contiguous() && egw.is_contiguous()
+                && delta16.is_contiguous() && dall.is_contiguous(),
+                "table tensors must be contiguous");
+    const int H = dirichlets.size(0), C = dirichlets.size(1);
+    auto a_col = torch::empty({H}, dirichlets.options());
+    auto b_col = torch::empty({H}, dirichlets.options());
+    auto stream = c10::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(trc_cols_kernel, dim3(H), dim3(BLOCK), 0,
+                       stream.stream(), dirichlets.data_ptr<float>(),
+                       reinterpret_cast<const long long*>(y.data_ptr<int64_t>()), a_col.data_ptr<float>(),
+                       b_col.data_ptr<float>(), H, C);
+    auto eglc = beta_row_tables(a_col, b_col, update_weight);
+    auto& eg = eglc[0];
+    auto& lc = eglc[1];
+    auto esb = torch::empty({P_POINTS}, dirichlets.options());
+    hipLaunchKernelGGL(trc_sums_kernel, dim3(1), dim3(P_POINTS), 0,
+                       stream.stream(), lc.data_ptr<float>(),
+                       weights.data_ptr<float>(),
+                       reinterpret_cast<const long long*>(y.data_ptr<int64_t>()), s_base.data_ptr<float>(),
+                       dall.data_ptr<float>(), esb.data_ptr<float>(), H);
+    const int total = 2 * H * P_POINTS;
+    hipLaunchKernelGGL(trc_rows_kernel,
+                       dim3((total + BLOCK - 1) / BLOCK), dim3(BLOCK), 0,
+                       stream.stream(), eg.data_ptr<float>(),
+                       lc.data_ptr<float>(), esb.data_ptr<float>(),
+                       reinterpret_cast<const long long*>(y.data_ptr<int64_t>()), EG.data_ptr<float>(),
+                       delta.data_ptr<float>(),
+                       reinterpret_cast<hip_bfloat16*>(eg16.data_ptr()),
+                       reinterpret_cast<hip_bfloat16*>(egw.data_ptr()),
+                       reinterpret_cast<_Float16*>(delta16.data_ptr()),
+                       H);
+    C10_HIP_CHECK(hipGetLastError());
+    return {a_col, b_col};
+}
+
 torch::Tensor pi_hat_delta(torch::Tensor preds, torch::Tensor cls) {
     TORCH_CHECK(preds.is_cuda() && preds.is_contiguous(),
                 "preds must be contiguous on a ROCm device");
@@ -2098,6 +2232,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "rank-1 pi_hat increment: sum_h preds[h, :, cls_h]");
     m.def("mixture_entropy", &mixture_entropy,
           "mixture0 (H,) = sum_c pi[c]*rows[c,h] + fused log2 entropy");
+    m.def("table_commit_row", &table_commit_row,
+          "per-label class-row table refresh (cols + curves + commits)");
     m.def("pi_marginal", &pi_marginal,
           "pi[c] = sum_n adjusted[n,c]/clamp(rowsum[n]) in one pass");
 }

@@ -651,31 +651,44 @@ class CODA(ModelSelector):
         self._g_pi.copy_(pi / pi.sum())
         # refresh the labeled class's table row (v2 tables)
         if self._tables is not None:
-            row = self.dirichlets.index_select(1, y_t).squeeze(1)  # (Hl,C)
-            a_col = row.gather(
-                1, y_t.view(1, 1).expand(self.Hl, 1)).squeeze(1).contiguous()
-            b_col = (row.sum(1) - a_col).contiguous()
-            eg, lc = ops._ext.beta_row_tables(a_col, b_col, 1.0)
             t = self._tables
-            t.EG.index_copy_(0, y_t, eg.unsqueeze(0))
-            if t.eg16 is not None:
-                t.eg16.index_copy_(
-                    0, y_t, eg.reshape(1, 2 * self.Hl, -1)
-                    .to(torch.bfloat16))
-            t.delta.index_copy_(0, y_t, (lc[:, 1] - lc[:, 0]).unsqueeze(0))
-            t.s_base.index_copy_(0, y_t, lc[:, 0].sum(0).unsqueeze(0))
-            if t.dall is not None:
-                t.dall.index_copy_(
-                    0, y_t, t.delta.index_select(0, y_t).sum(1))
-            if t.egw is not None:
-                esb = torch.exp2(t.s_base.index_select(0, y_t)) \
-                    * t.weights                             # (1, P)
-                t.egw.index_copy_(
-                    0, y_t, (eg.reshape(1, 2 * self.Hl, -1)
-                             * esb.unsqueeze(1)).to(torch.bfloat16))
-                t.delta16.index_copy_(
-                    0, y_t, t.delta.index_select(0, y_t)
-                    .to(torch.float16))
+            if (t.eg16 is not None and t.egw is not None
+                    and t.delta16 is not None and t.dall is not None
+                    and ops.hip_available()):
+                # one fused commit (trc_* kernels in pbest.hip): the
+                # torch chain below was ~14 in-graph launches incl. two
+                # 32-thread strided reductions (10-24 us each)
+                a_col, b_col = ops._ext.table_commit_row(
+                    self.dirichlets, y_t, t.EG, t.delta, t.s_base,
+                    t.weights, t.eg16, t.egw, t.delta16, t.dall, 1.0)
+            else:
+                row = self.dirichlets.index_select(1, y_t) \
+                    .squeeze(1)                            # (Hl, C)
+                a_col = row.gather(
+                    1, y_t.view(1, 1).expand(self.Hl, 1)) \
+                    .squeeze(1).contiguous()
+                b_col = (row.sum(1) - a_col).contiguous()
+                eg, lc = ops._ext.beta_row_tables(a_col, b_col, 1.0)
+                t.EG.index_copy_(0, y_t, eg.unsqueeze(0))
+                if t.eg16 is not None:
+                    t.eg16.index_copy_(
+                        0, y_t, eg.reshape(1, 2 * self.Hl, -1)
+                        .to(torch.bfloat16))
+                t.delta.index_copy_(0, y_t,
+                                    (lc[:, 1] - lc[:, 0]).unsqueeze(0))
+                t.s_base.index_copy_(0, y_t, lc[:, 0].sum(0).unsqueeze(0))
+                if t.dall is not None:
+                    t.dall.index_copy_(
+                        0, y_t, t.delta.index_select(0, y_t).sum(1))
+                if t.egw is not None:
+                    esb = torch.exp2(t.s_base.index_select(0, y_t)) \
+                        * t.weights                         # (1, P)
+                    t.egw.index_copy_(
+                        0, y_t, (eg.reshape(1, 2 * self.Hl, -1)
+                                 * esb.unsqueeze(1)).to(torch.bfloat16))
+                    t.delta16.index_copy_(
+                        0, y_t, t.delta.index_select(0, y_t)
+                        .to(torch.float16))
         # posterior rows: add_label moves only Dirichlet row y, so only
         # class y's Beta column - hence only pbest row y - changes.
         # _g_rows was seeded with the full (C, Hl) rows at graph init;
