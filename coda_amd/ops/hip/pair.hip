@@ -104,7 +104,35 @@ pair_dsum_es_kernel(const _Float16* __restrict__ delta16,  // (C, H, P)
     const size_t dbase = (size_t)c * H * P_POINTS + p0;
 
     float acc[16] = {};
-    for (int s = s0; s < s1; ++s) {
+    float acc2[16] = {};
+    int s = s0;
+    // 2-row software pipeline: the single-row loop is a dependent
+    // seg_h -> 4-load -> FMA chain (the table lives in the LLC, so
+    // this kernel is latency-bound, not HBM-bound); two rows with
+    // split accumulators keep 8 loads in flight per lane (a 4-row
+    // variant measured the same)
+    for (; s + 1 < s1; s += 2) {
+        const _Float16* ra = delta16 + dbase
+            + (size_t)seg_h[s] * P_POINTS;
+        const _Float16* rb = delta16 + dbase
+            + (size_t)seg_h[s + 1] * P_POINTS;
+        const half4v a0 = *reinterpret_cast<const half4v*>(ra);
+        const half4v a1 = *reinterpret_cast<const half4v*>(ra + 4);
+        const half4v a2 = *reinterpret_cast<const half4v*>(ra + 8);
+        const half4v a3 = *reinterpret_cast<const half4v*>(ra + 12);
+        const half4v b0 = *reinterpret_cast<const half4v*>(rb);
+        const half4v b1 = *reinterpret_cast<const half4v*>(rb + 4);
+        const half4v b2 = *reinterpret_cast<const half4v*>(rb + 8);
+        const half4v b3 = *reinterpret_cast<const half4v*>(rb + 12);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            acc[j] += (float)a0[j];      acc2[j] += (float)b0[j];
+            acc[4 + j] += (float)a1[j];  acc2[4 + j] += (float)b1[j];
+            acc[8 + j] += (float)a2[j];  acc2[8 + j] += (float)b2[j];
+            acc[12 + j] += (float)a3[j]; acc2[12 + j] += (float)b3[j];
+        }
+    }
+    if (s < s1) {
         const _Float16* row = delta16 + dbase + (size_t)seg_h[s] * P_POINTS;
         const half4v d0 = *reinterpret_cast<const half4v*>(row);
         const half4v d1 = *reinterpret_cast<const half4v*>(row + 4);
@@ -118,6 +146,8 @@ pair_dsum_es_kernel(const _Float16* __restrict__ delta16,  // (C, H, P)
             acc[12 + j] += (float)d3[j];
         }
     }
+#pragma unroll
+    for (int j = 0; j < 16; ++j) acc[j] += acc2[j];
     if (pair_neg[k]) {
         // complement segment: dsum = (sum over ALL models) - partial
         const float4* da = reinterpret_cast<const float4*>(
