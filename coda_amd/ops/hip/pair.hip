@@ -302,13 +302,18 @@ pair_gemm_entropy128_kernel(const hip_bfloat16* __restrict__ a16,
                             const float* __restrict__ pbest_before,
                             const float* __restrict__ mixture0,
                             float* __restrict__ h_after,
+                            const int* __restrict__ grp_off,
                             int H, int mstride) {
     extern __shared__ char smem[];
     const int twoH = 2 * H;
     hip_bfloat16* b_lds = reinterpret_cast<hip_bfloat16*>(smem);
 
-    const int k0 = blockIdx.x * 128;
-    const int c = pair_c[k0];
+    // block owns tiles [t0, t1) - all the same class, so the 131 KB
+    // B stage amortizes over up to GRP_MAX tiles (classes average
+    // ~2.4 tiles at the headline hit distribution; B staging was
+    // ~60% of this kernel's global traffic)
+    const int t0 = grp_off[blockIdx.x], t1 = grp_off[blockIdx.x + 1];
+    const int c = pair_c[t0 * 128];
     const int tid = threadIdx.x;
     const int wave = tid >> 6, lane = tid & 63;
     const int row16 = lane & 15;
@@ -329,6 +334,8 @@ pair_gemm_entropy128_kernel(const hip_bfloat16* __restrict__ a16,
     }
     __syncthreads();
 
+    for (int tile = t0; tile < t1; ++tile) {
+    const int k0 = tile * 128;
     f32x4 acc[JT];
 #pragma unroll
     for (int jt = 0; jt < JT; ++jt) acc[jt] = {0.f, 0.f, 0.f, 0.f};
@@ -404,6 +411,7 @@ pair_gemm_entropy128_kernel(const hip_bfloat16* __restrict__ a16,
         const float e = row16_reduce(ent[r]);
         if (row16 == 0) h_after[kbase + r] = e;
     }
+    }  // tile loop
     (void)mstride;
 }
 
@@ -804,7 +812,7 @@ torch::Tensor pair_gemm_entropy(torch::Tensor a16, torch::Tensor egw,
                                 torch::Tensor pi_hat,
                                 torch::Tensor pbest_before,
                                 torch::Tensor mixture0, int64_t tile,
-                                int64_t ablate) {
+                                int64_t ablate, torch::Tensor grp) {
     TORCH_CHECK(a16.is_cuda() && a16.dtype() == torch::kBFloat16);
     TORCH_CHECK(egw.dtype() == torch::kBFloat16);
     const int K = a16.size(0);
@@ -849,8 +857,16 @@ torch::Tensor pair_gemm_entropy(torch::Tensor a16, torch::Tensor egw,
         const size_t shmem = (size_t)2 * H * BSTRIDE
                            * sizeof(hip_bfloat16);
         const int JT = (2 * H + 15) / 16;
+        // tile groups (same-class runs, capped): default 1 tile/block
+        torch::Tensor grp_t = grp;
+        if (grp_t.numel() == 0)
+            grp_t = torch::arange(K / 128 + 1,
+                                  pair_c.options().dtype(torch::kInt32));
+        TORCH_CHECK(grp_t.scalar_type() == torch::kInt32
+                    && grp_t.is_contiguous());
+        const int ngrp = grp_t.numel() - 1;
         auto launch = [&](auto kern) {
-            hipLaunchKernelGGL(kern, dim3(K / 128), dim3(BLOCK2), shmem,
+            hipLaunchKernelGGL(kern, dim3(ngrp), dim3(BLOCK2), shmem,
                                stream.stream(), ab16, eb16,
                                reinterpret_cast<const unsigned*>(
                                    vmask.data_ptr<int>()),
@@ -859,7 +875,8 @@ torch::Tensor pair_gemm_entropy(torch::Tensor a16, torch::Tensor egw,
                                pi_hat.data_ptr<float>(),
                                pbest_before.data_ptr<float>(),
                                mixture0.data_ptr<float>(),
-                               h_after.data_ptr<float>(), H, mstride);
+                               h_after.data_ptr<float>(),
+                               grp_t.data_ptr<int>(), H, mstride);
         };
         (void)ablate;  // phase-ablation diagnostics retired
         if (JT <= 4) launch(pairops::pair_gemm_entropy128_kernel<4>);
@@ -1017,7 +1034,8 @@ void register_pair_ops(pybind11::module_& m) {
           pybind11::arg("vmask"), pybind11::arg("pair_c"),
           pybind11::arg("pi_hat"),
           pybind11::arg("pbest_before"), pybind11::arg("mixture0"),
-          pybind11::arg("tile"), pybind11::arg("ablate") = 0);
+          pybind11::arg("tile"), pybind11::arg("ablate") = 0,
+          pybind11::arg("grp") = torch::Tensor());
     m.def("pair_gemm_entropy_cls", &pair_gemm_entropy_cls,
           "v3 wide-H pairing GEMM + cls-based entropy -> (K,)");
     m.def("pair_eig_finalize", &pair_eig_finalize,

@@ -88,6 +88,11 @@ class PairStructure(NamedTuple):
     # so summing the few non-hitting models' deltas and subtracting
     # from the per-class total nearly halves the dsum read traffic)
     pair_neg: torch.Tensor = None
+    # (G+1,) int32 — 128-pair TILE groups: runs of same-class tiles
+    # capped at GRP_MAX, so the GEMM's 131 KB B stage amortizes over a
+    # class's tiles instead of re-staging per tile (classes average
+    # ~2.4 tiles at the headline distribution)
+    grp_off: torch.Tensor = None
 
     @property
     def K(self) -> int:
@@ -245,13 +250,30 @@ def build_pairs(cls_rows: torch.Tensor, cand_ids: torch.Tensor,
     # candidate slices (ids[rank::world]), and the finalize kernel
     # walks the raw buffer - a strided view would silently read the
     # WRONG candidates' rows (caught by the GPU loopback world-4 test)
+    grp_off = None
+    if tile == 128 and K % 128 == 0 and K > 0:
+        GRP_MAX = 4
+        tc = pair_c[::128]                         # (T,) tile classes
+        T = tc.numel()
+        newc = torch.ones(T, dtype=torch.bool, device=device)
+        newc[1:] = tc[1:] != tc[:-1]
+        idx = torch.arange(T, device=device)
+        run_start = idx[newc]
+        run_id = newc.long().cumsum(0) - 1
+        off_in_run = idx - run_start[run_id]
+        gstart = newc | (off_in_run % GRP_MAX == 0)
+        grp_off = torch.cat([
+            idx[gstart].to(torch.int32),
+            torch.tensor([T], dtype=torch.int32, device=device),
+        ]).contiguous()
+
     return PairStructure(cand_ids=cand_ids.long().contiguous(),
                          pair_b=pair_b,
                          pair_c=pair_c, seg_off=seg_off, seg_h=seg_h,
                          base_pos=run_off[:-1].clone(), n_real=K2,
                          tile=tile, cand_off=cand_off,
                          cand_pairs=cand_pairs, vmask=vmask,
-                         pair_neg=pair_neg)
+                         pair_neg=pair_neg, grp_off=grp_off)
 
 
 def pair_h_after(tables, ps: PairStructure, cls_rows: torch.Tensor,
@@ -349,10 +371,13 @@ def eig_pairs(tables, ps: PairStructure, cls_rows: torch.Tensor,
                 pi_hat.contiguous(), pbest_before.contiguous(),
                 mixture0.contiguous())                      # (K,)
         else:
+            grp = (ps.grp_off if ps.grp_off is not None
+                   else torch.empty(0, dtype=torch.int32,
+                                    device=A16.device))
             h_after = O._ext.pair_gemm_entropy(
                 A16, tables.egw, ps.vmask, ps.pair_c,
                 pi_hat.contiguous(), pbest_before.contiguous(),
-                mixture0.contiguous(), ps.tile)             # (K,)
+                mixture0.contiguous(), ps.tile, 0, grp)     # (K,)
         h_base = h_after.index_select(0, ps.base_pos).contiguous()
         q = O._ext.pair_eig_finalize(
             h_after, h_base, ps.pair_c,

@@ -479,9 +479,11 @@ class CODA(ModelSelector):
         mixture0, H0 = ops.mixture_entropy(rows, pi)
         A16 = ops._ext.pair_dsum_es(t.delta16, t.dall, ps.pair_c,
                                     ps.pair_neg, ps.seg_off, ps.seg_h)
+        grp = (ps.grp_off if ps.grp_off is not None
+               else torch.empty(0, dtype=torch.int32, device=A16.device))
         h_after = ops._ext.pair_gemm_entropy(
             A16, t.egw, ps.vmask, ps.pair_c, pi, rows,
-            mixture0.contiguous(), ps.tile)
+            mixture0.contiguous(), ps.tile, 0, grp)
         # H_before enters as an in-graph tensor op (the kernel's scalar
         # argument would be frozen at capture value)
         h_base = h_after.index_select(0, ps.base_pos).contiguous()
