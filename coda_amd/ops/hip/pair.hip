@@ -272,7 +272,9 @@ pair_gemm_entropy16_kernel(const hip_bfloat16* __restrict__ a16,
 // compile-time), and 16-lane DPP row reductions produce tot/entropy -
 // the (128, 2H) M tile never exists in LDS or HBM.
 // ---------------------------------------------------------------------
-#define BSTRIDE 264   // bf16 elems (528 B rows keep 16-B alignment for b128)
+#define BSTRIDE 272  // B row stride in halfs: 136 words, 8-word aligned (132
+                     // put odd-row b128 reads at =4 mod 8 words: +60
+                     // stall cycles each on CDNA4)
 #define BLOCK2 512
 
 __device__ __forceinline__ float row16_reduce(float v) {
