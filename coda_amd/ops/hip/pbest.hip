@@ -1140,13 +1140,26 @@ trc_sums_kernel(const float* __restrict__ lc,    // (H, 2, P)
                 int H) {
     const int p = threadIdx.x;
     const long long yy = y[0];
-    float s = 0.f, d = 0.f;
-    for (int h = 0; h < H; ++h) {
+    // 2-row unroll with split accumulators: the serial strided walk
+    // over H rows was ~34 us latency-bound on one workgroup
+    float s = 0.f, d = 0.f, s2 = 0.f, d2 = 0.f;
+    int h = 0;
+    for (; h + 1 < H; h += 2) {
+        const float l0 = lc[((size_t)h * 2) * P_POINTS + p];
+        const float l1 = lc[((size_t)h * 2 + 1) * P_POINTS + p];
+        const float m0 = lc[((size_t)(h + 1) * 2) * P_POINTS + p];
+        const float m1 = lc[((size_t)(h + 1) * 2 + 1) * P_POINTS + p];
+        s += l0;       d += l1 - l0;
+        s2 += m0;      d2 += m1 - m0;
+    }
+    if (h < H) {
         const float l0 = lc[((size_t)h * 2) * P_POINTS + p];
         const float l1 = lc[((size_t)h * 2 + 1) * P_POINTS + p];
         s += l0;
         d += l1 - l0;
     }
+    s += s2;
+    d += d2;
     s_base[yy * P_POINTS + p] = s;
     dall[yy * P_POINTS + p] = d;
     esb[p] = exp2f(s) * w[p];
@@ -1451,12 +1464,35 @@ mix_combine_kernel(const float* __restrict__ partial,  // (G, H)
     const int h4 = tid * 4;
     float ent = 0.f;
     if (h4 < H) {
+        // 4-way G unroll: at H=128 only 32 threads are active and the
+        // serial strided walk was ~17 us latency-bound; independent
+        // accumulators keep 4 loads in flight
         float4 acc = {0.f, 0.f, 0.f, 0.f};
-        for (int g = 0; g < G; ++g) {
+        float4 a1 = {0.f, 0.f, 0.f, 0.f};
+        float4 a2 = {0.f, 0.f, 0.f, 0.f};
+        float4 a3 = {0.f, 0.f, 0.f, 0.f};
+        int g = 0;
+        for (; g + 3 < G; g += 4) {
+            const float4 v0 = *reinterpret_cast<const float4*>(
+                partial + (size_t)g * H + h4);
+            const float4 v1 = *reinterpret_cast<const float4*>(
+                partial + (size_t)(g + 1) * H + h4);
+            const float4 v2 = *reinterpret_cast<const float4*>(
+                partial + (size_t)(g + 2) * H + h4);
+            const float4 v3 = *reinterpret_cast<const float4*>(
+                partial + (size_t)(g + 3) * H + h4);
+            acc.x += v0.x; acc.y += v0.y; acc.z += v0.z; acc.w += v0.w;
+            a1.x += v1.x; a1.y += v1.y; a1.z += v1.z; a1.w += v1.w;
+            a2.x += v2.x; a2.y += v2.y; a2.z += v2.z; a2.w += v2.w;
+            a3.x += v3.x; a3.y += v3.y; a3.z += v3.z; a3.w += v3.w;
+        }
+        for (; g < G; ++g) {
             const float4 v = *reinterpret_cast<const float4*>(
                 partial + (size_t)g * H + h4);
             acc.x += v.x; acc.y += v.y; acc.z += v.z; acc.w += v.w;
         }
+        acc.x += a1.x + a2.x + a3.x; acc.y += a1.y + a2.y + a3.y;
+        acc.z += a1.z + a2.z + a3.z; acc.w += a1.w + a2.w + a3.w;
         *reinterpret_cast<float4*>(mixture0 + h4) = acc;
         const float m[4] = {fmaxf(acc.x, 1e-12f), fmaxf(acc.y, 1e-12f),
                             fmaxf(acc.z, 1e-12f), fmaxf(acc.w, 1e-12f)};
