@@ -699,7 +699,7 @@ __global__ void __launch_bounds__(ACQ_GRID)
 acq_select_combine_kernel(const float* __restrict__ pmax,
                           const int* __restrict__ pidx,
                           double* __restrict__ out,     // (3,) f64
-                          int* __restrict__ tbuf) {
+                          long long* __restrict__ tbuf) {
     __shared__ float sv[ACQ_GRID];
     __shared__ int si[ACQ_GRID];
     sv[threadIdx.x] = pmax[threadIdx.x];
@@ -730,7 +730,7 @@ __global__ void __launch_bounds__(BLOCK)
 acq_select_ties_kernel(const float* __restrict__ qbuf,
                        const bool* __restrict__ active,
                        double* __restrict__ out,
-                       int* __restrict__ tbuf,   // [0]=n, [1..cap]=idx
+                       long long* __restrict__ tbuf,  // [0]=n, [1..cap]
                        int cap, int B) {
     const float best = (float)out[0];
     const float thr = 1e-8f + 1e-8f * fabsf(best);
@@ -741,10 +741,14 @@ acq_select_ties_kernel(const float* __restrict__ qbuf,
     for (int i = b0 + threadIdx.x; i < b1; i += BLOCK)
         if (active[i] && fabsf(qbuf[i] - best) <= thr) {
             ++cnt;
-            // unordered slots; the host sorts the <= cap indices back
-            // to ascending before the seeded tie-break choice
-            const int slot = atomicAdd(tbuf, 1);
-            if (slot < cap) tbuf[1 + slot] = i;
+            // unordered slots, each packing (position, q value) so the
+            // host tie-break needs NO second device fetch; the host
+            // sorts the <= cap entries back to ascending position
+            const int slot = (int)atomicAdd(
+                reinterpret_cast<unsigned long long*>(tbuf), 1ull);
+            if (slot < cap)
+                tbuf[1 + slot] = ((long long)i << 32)
+                    | (unsigned int)__float_as_int(qbuf[i]);
         }
     __shared__ int sc[BLOCK];
     sc[threadIdx.x] = cnt;
@@ -950,7 +954,7 @@ void acq_select(torch::Tensor q0, torch::Tensor h0, torch::Tensor active,
                 && qbuf.numel() == q0.numel());
     TORCH_CHECK(out.dtype() == torch::kFloat64 && out.is_contiguous()
                 && out.numel() == 3);
-    TORCH_CHECK(ties.dtype() == torch::kInt32 && ties.is_contiguous()
+    TORCH_CHECK(ties.dtype() == torch::kInt64 && ties.is_contiguous()
                 && ties.numel() >= 2, "ties buffer too small");
     const int cap = ties.numel() - 1;
     const int B = q0.numel();
@@ -966,11 +970,15 @@ void acq_select(torch::Tensor q0, torch::Tensor h0, torch::Tensor active,
     hipLaunchKernelGGL(pairops::acq_select_combine_kernel, dim3(1),
                        dim3(ACQ_GRID), 0, stream.stream(),
                        pmax.data_ptr<float>(), pidx.data_ptr<int>(),
-                       out.data_ptr<double>(), ties.data_ptr<int>());
+                       out.data_ptr<double>(),
+                       reinterpret_cast<long long*>(
+                           ties.data_ptr<int64_t>()));
     hipLaunchKernelGGL(pairops::acq_select_ties_kernel, dim3(ACQ_GRID),
                        dim3(BLOCK), 0, stream.stream(),
                        qbuf.data_ptr<float>(), active.data_ptr<bool>(),
-                       out.data_ptr<double>(), ties.data_ptr<int>(),
+                       out.data_ptr<double>(),
+                       reinterpret_cast<long long*>(
+                           ties.data_ptr<int64_t>()),
                        cap, B);
     C10_HIP_CHECK(hipGetLastError());
 }

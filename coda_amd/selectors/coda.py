@@ -502,25 +502,34 @@ class CODA(ModelSelector):
                             self._acq_out, self._acq_ties)
 
     def _acq_result(self):
-        bv, bi, nt = self._acq_out.cpu().tolist()
+        import struct
+        self._acq_out_host.copy_(self._acq_out)
+        bv, bi, nt = self._acq_out_host.tolist()
         bi, nt = int(bi), int(nt)
         if nt > 1:
             # same tie semantics as the eager path: active candidates
             # ascend by point id in both orderings.  The fused epilogue
-            # collected the tie positions already (unordered slots);
-            # sorting restores the ascending order the seeded
+            # collected (position, q value) pairs already (unordered
+            # slots); sorting restores the ascending order the seeded
             # random.choice tie-break expects.
             if nt < self._acq_ties.numel():
-                pos = random.choice(
-                    sorted(self._acq_ties[1:1 + nt].cpu().tolist()))
+                th = self._acq_ties_host[:1 + nt]
+                th.copy_(self._acq_ties[:1 + nt])
+                packed = sorted(v & 0xFFFFFFFFFFFFFFFF
+                                for v in th[1:].tolist())
+                pk = random.choice(packed)
+                pos = pk >> 32
+                qv = struct.unpack(
+                    "<f", struct.pack("<I", pk & 0xFFFFFFFF))[0]
             else:  # > buffer capacity: rescan (never seen in practice)
                 q = self._acq_qbuf
                 ties = (torch.isclose(q, q.max(), rtol=1e-8)
                         & self._active_mask)
                 pos = random.choice(
                     torch.nonzero(ties, as_tuple=True)[0].tolist())
+                qv = float(q[pos])
             self.stochastic = True
-            return self._pairs_ids_host[pos], float(self._acq_qbuf[pos])
+            return self._pairs_ids_host[pos], qv
         return self._pairs_ids_host[bi], bv
 
     def _init_acq_buffers(self):
@@ -531,11 +540,18 @@ class CODA(ModelSelector):
                                         device=self.device)
             self._acq_qbuf = torch.empty(ps.cand_ids.numel(),
                                          device=self.device)
-            # fused-epilogue tie-index buffer: [0]=slot counter,
-            # [1:]=unordered tie positions (host-sorted back to
-            # ascending); overflow falls back to the full-scan path
-            self._acq_ties = torch.zeros(8192, dtype=torch.int32,
+            # fused-epilogue tie buffer: [0]=slot counter, [1:] packs
+            # (position << 32) | q-value-bits per tie (host-sorted back
+            # to ascending position - one D2H fetch serves both the
+            # seeded tie-break AND its q value); overflow falls back to
+            # the full-scan path.  Host mirrors are PINNED so the
+            # per-step result fetches take the fast DMA path.
+            self._acq_ties = torch.zeros(8192, dtype=torch.int64,
                                          device=self.device)
+            self._acq_out_host = torch.empty(
+                3, dtype=torch.float64, pin_memory=True)
+            self._acq_ties_host = torch.empty(
+                8192, dtype=torch.int64, pin_memory=True)
 
     def _acq_eligible(self) -> bool:
         return (self.q == "eig" and self._pairs_static is not None
