@@ -2005,6 +2005,36 @@ std::vector<torch::Tensor> beta_row_tables(torch::Tensor alpha_col,
 }
 
 
+static void commit_row_launch(torch::Tensor& eg, torch::Tensor& lc,
+                              torch::Tensor& y, torch::Tensor& EG,
+                              torch::Tensor& delta, torch::Tensor& s_base,
+                              torch::Tensor& weights, torch::Tensor& eg16,
+                              torch::Tensor& egw, torch::Tensor& delta16,
+                              torch::Tensor& dall, int H) {
+    auto stream = c10::hip::getCurrentHIPStream();
+    auto esb = torch::empty({P_POINTS}, eg.options());
+    hipLaunchKernelGGL(trc_sums_kernel, dim3(1), dim3(P_POINTS), 0,
+                       stream.stream(), lc.data_ptr<float>(),
+                       weights.data_ptr<float>(),
+                       reinterpret_cast<const long long*>(
+                           y.data_ptr<int64_t>()),
+                       s_base.data_ptr<float>(),
+                       dall.data_ptr<float>(), esb.data_ptr<float>(), H);
+    const int total = 2 * H * P_POINTS;
+    hipLaunchKernelGGL(trc_rows_kernel,
+                       dim3((total + BLOCK - 1) / BLOCK), dim3(BLOCK), 0,
+                       stream.stream(), eg.data_ptr<float>(),
+                       lc.data_ptr<float>(), esb.data_ptr<float>(),
+                       reinterpret_cast<const long long*>(
+                           y.data_ptr<int64_t>()),
+                       EG.data_ptr<float>(), delta.data_ptr<float>(),
+                       reinterpret_cast<hip_bfloat16*>(eg16.data_ptr()),
+                       reinterpret_cast<hip_bfloat16*>(egw.data_ptr()),
+                       reinterpret_cast<_Float16*>(delta16.data_ptr()),
+                       H);
+    C10_HIP_CHECK(hipGetLastError());
+}
+
 std::vector<torch::Tensor> table_commit_row(
         torch::Tensor dirichlets, torch::Tensor y, torch::Tensor EG,
         torch::Tensor delta, torch::Tensor s_base, torch::Tensor weights,
@@ -2026,27 +2056,27 @@ std::vector<torch::Tensor> table_commit_row(
                        reinterpret_cast<const long long*>(y.data_ptr<int64_t>()), a_col.data_ptr<float>(),
                        b_col.data_ptr<float>(), H, C);
     auto eglc = beta_row_tables(a_col, b_col, update_weight);
-    auto& eg = eglc[0];
-    auto& lc = eglc[1];
-    auto esb = torch::empty({P_POINTS}, dirichlets.options());
-    hipLaunchKernelGGL(trc_sums_kernel, dim3(1), dim3(P_POINTS), 0,
-                       stream.stream(), lc.data_ptr<float>(),
-                       weights.data_ptr<float>(),
-                       reinterpret_cast<const long long*>(y.data_ptr<int64_t>()), s_base.data_ptr<float>(),
-                       dall.data_ptr<float>(), esb.data_ptr<float>(), H);
-    const int total = 2 * H * P_POINTS;
-    hipLaunchKernelGGL(trc_rows_kernel,
-                       dim3((total + BLOCK - 1) / BLOCK), dim3(BLOCK), 0,
-                       stream.stream(), eg.data_ptr<float>(),
-                       lc.data_ptr<float>(), esb.data_ptr<float>(),
-                       reinterpret_cast<const long long*>(y.data_ptr<int64_t>()), EG.data_ptr<float>(),
-                       delta.data_ptr<float>(),
-                       reinterpret_cast<hip_bfloat16*>(eg16.data_ptr()),
-                       reinterpret_cast<hip_bfloat16*>(egw.data_ptr()),
-                       reinterpret_cast<_Float16*>(delta16.data_ptr()),
-                       H);
-    C10_HIP_CHECK(hipGetLastError());
+    commit_row_launch(eglc[0], eglc[1], y, EG, delta, s_base, weights,
+                      eg16, egw, delta16, dall, H);
     return {a_col, b_col};
+}
+
+// eager-path variant: caller already has the class column (the
+// distributed ranks' table refresh); y is a host int.
+void table_commit_cols(torch::Tensor a_col, torch::Tensor b_col,
+                       int64_t y, torch::Tensor EG, torch::Tensor delta,
+                       torch::Tensor s_base, torch::Tensor weights,
+                       torch::Tensor eg16, torch::Tensor egw,
+                       torch::Tensor delta16, torch::Tensor dall,
+                       double update_weight) {
+    check_f32_cuda(a_col, "a_col");
+    check_f32_cuda(b_col, "b_col");
+    const int H = a_col.size(0);
+    auto y_t = torch::full({1}, y,
+                           a_col.options().dtype(torch::kInt64));
+    auto eglc = beta_row_tables(a_col, b_col, update_weight);
+    commit_row_launch(eglc[0], eglc[1], y_t, EG, delta, s_base, weights,
+                      eg16, egw, delta16, dall, H);
 }
 
 torch::Tensor pi_hat_delta(torch::Tensor preds, torch::Tensor cls) {
@@ -2268,6 +2298,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "rank-1 pi_hat increment: sum_h preds[h, :, cls_h]");
     m.def("mixture_entropy", &mixture_entropy,
           "mixture0 (H,) = sum_c pi[c]*rows[c,h] + fused log2 entropy");
+    m.def("table_commit_cols", &table_commit_cols,
+          "per-class table refresh from Beta columns (eager path)");
     m.def("table_commit_row", &table_commit_row,
           "per-label class-row table refresh (cols + curves + commits)");
     m.def("pi_marginal", &pi_marginal,

@@ -230,6 +230,21 @@ def table_update_rows(tables: EigTables, alpha_cc: torch.Tensor,
     import coda_amd.ops as O
     if (alpha_cc.is_cuda and tables.EG.shape[-1] == PBEST_NUM_POINTS
             and rows.numel() <= 4 and O._want_hip(alpha_cc)):
+        if (tables.eg16 is not None and tables.egw is not None
+                and tables.delta16 is not None
+                and tables.dall is not None):
+            # fully-fused per-class commit (curves + sums + all table
+            # writes in 3 kernels - the torch chain below spends
+            # ~50 us/class in 32-thread strided reductions and small
+            # conversion launches on the eager/distributed path)
+            for c in rows.tolist():
+                O._ext.table_commit_cols(
+                    alpha_cc[:, c].contiguous(),
+                    beta_cc[:, c].contiguous(), int(c),
+                    tables.EG, tables.delta, tables.s_base,
+                    tables.weights, tables.eg16, tables.egw,
+                    tables.delta16, tables.dall, float(update_weight))
+            return tables
         # per-class refresh kernel: one wave per (model, variant) curve
         H = alpha_cc.shape[0]
         for c in rows.tolist():
