@@ -22,6 +22,7 @@ the math):
 from __future__ import annotations
 
 import random
+import struct
 
 import torch
 from sortedcontainers import SortedList
@@ -502,10 +503,12 @@ class CODA(ModelSelector):
                             self._acq_out, self._acq_ties)
 
     def _acq_result(self):
-        import struct
         self._acq_out_host.copy_(self._acq_out)
         bv, bi, nt = self._acq_out_host.tolist()
         bi, nt = int(bi), int(nt)
+        if bi < 0:
+            raise RuntimeError("acquisition ran with no active "
+                               "candidates (pool exhausted?)")
         if nt > 1:
             # same tie semantics as the eager path: active candidates
             # ascend by point id in both orderings.  The fused epilogue
