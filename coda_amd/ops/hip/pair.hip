@@ -565,7 +565,7 @@ pair_eig_finalize_kernel(const float* __restrict__ h_after,   // (K,)
                          const float* __restrict__ h_base,    // (C,)
                          const int* __restrict__ pair_c,      // (K,)
                          const int* __restrict__ cand_off,    // (B+1,)
-                         const int* __restrict__ cand_pairs,  // (hits,)
+                         const long long* __restrict__ cand_ck,  // (hits,)
                          const long* __restrict__ cand_ids,   // (B,)
                          const float* __restrict__ adjusted,  // (N, C)
                          const float* __restrict__ row_sums,  // (N,)
@@ -588,11 +588,17 @@ pair_eig_finalize_kernel(const float* __restrict__ h_after,   // (K,)
     for (; cc < C; ++cc) base += arow[cc] * h_base[cc];
     float corr = 0.f;
     const int s1 = cand_off[b + 1];
+    // (pair, class) packed per hit: one load decodes both, so the
+    // chase is packed -> {h_after, h_base, arow} (2 dependent levels,
+    // the last three loads independent) instead of
+    // cand_pairs -> pair_c -> gathers (3 levels)
     for (int s = cand_off[b] + lane; s < s1; s += 64) {
-        const int k = cand_pairs[s];
-        const int c2 = pair_c[k];
+        const long long pk = cand_ck[s];
+        const int k = (int)(pk >> 32);
+        const int c2 = (int)(pk & 0xffffffff);
         corr += arow[c2] * (h_after[k] - h_base[c2]);
     }
+    (void)pair_c;
     const float tot = wave_reduce(base + corr);
     if (lane == 0) q[b] = H_before - tot * inv;
 }
@@ -916,7 +922,7 @@ torch::Tensor pair_eig_finalize(torch::Tensor h_after,
                                 torch::Tensor h_base,
                                 torch::Tensor pair_c,
                                 torch::Tensor cand_off,
-                                torch::Tensor cand_pairs,
+                                torch::Tensor cand_ck,
                                 torch::Tensor cand_ids,
                                 torch::Tensor adjusted,
                                 torch::Tensor row_sums,
@@ -924,8 +930,10 @@ torch::Tensor pair_eig_finalize(torch::Tensor h_after,
     const int B = cand_ids.size(0);
     const int C = adjusted.size(1);
     TORCH_CHECK(h_base.is_contiguous() && h_base.size(0) == C);
+    TORCH_CHECK(cand_ck.scalar_type() == torch::kInt64,
+                "cand_ck must be the packed int64 hit table");
     TORCH_CHECK(cand_ids.is_contiguous() && cand_off.is_contiguous()
-                && cand_pairs.is_contiguous(),
+                && cand_ck.is_contiguous(),
                 "finalize inputs must be contiguous");
     auto q = torch::empty({B}, adjusted.options());
     auto stream = c10::hip::getCurrentHIPStream();
@@ -934,7 +942,8 @@ torch::Tensor pair_eig_finalize(torch::Tensor h_after,
                        h_after.data_ptr<float>(),
                        h_base.data_ptr<float>(), pair_c.data_ptr<int>(),
                        cand_off.data_ptr<int>(),
-                       cand_pairs.data_ptr<int>(),
+                       reinterpret_cast<const long long*>(
+                           cand_ck.data_ptr<int64_t>()),
                        cand_ids.data_ptr<long>(),
                        adjusted.data_ptr<float>(),
                        row_sums.data_ptr<float>(),

@@ -490,7 +490,7 @@ class CODA(ModelSelector):
         h_base = h_after.index_select(0, ps.base_pos).contiguous()
         q0 = ops._ext.pair_eig_finalize(
             h_after, h_base, ps.pair_c, ps.cand_off,
-            ps.cand_pairs, ps.cand_ids, self._adjusted, self._row_sums,
+            ps.cand_ck, ps.cand_ids, self._adjusted, self._row_sums,
             0.0)
         # fused epilogue: qbuf = active ? H0 + q0 : -inf; _acq_out =
         # [masked max, first argmax, isclose tie count].  Replaces the

@@ -83,10 +83,11 @@ for tile in (16, 128):
                A16, tables.egw, ps.vmask, ps.pair_c,
                pi_hat.contiguous(), pbest_before.contiguous(),
                mixture0.contiguous(), tile))
+    h_base_t = h_after.index_select(0, ps.base_pos).contiguous()
     timeit(f"finalize[t{tile}]",
            lambda: O._ext.pair_eig_finalize(
-               h_after, ps.pair_c, ps.base_pos.to(torch.int32),
-               ps.cand_off, ps.cand_pairs, ps.cand_ids, adjusted,
+               h_after, h_base_t, ps.pair_c,
+               ps.cand_off, ps.cand_ck, ps.cand_ids, adjusted,
                row_sums, float(H_before)))
 
 # numerics cross-check tile16 vs tile64 h_after on the base rows
@@ -99,9 +100,9 @@ for ps, name in ((ps16, "t16"), (ps64, "t64")):
                                  pi_hat.contiguous(),
                                  pbest_before.contiguous(),
                                  mixture0.contiguous(), ps.tile)
-    q = O._ext.pair_eig_finalize(h, ps.pair_c,
-                                 ps.base_pos.to(torch.int32),
-                                 ps.cand_off, ps.cand_pairs, ps.cand_ids,
+    hb = h.index_select(0, ps.base_pos).contiguous()
+    q = O._ext.pair_eig_finalize(h, hb, ps.pair_c,
+                                 ps.cand_off, ps.cand_ck, ps.cand_ids,
                                  adjusted, row_sums, float(H_before))
     print(name, "q[:4]", q[:4].tolist())
 
