@@ -137,8 +137,15 @@ def mixture_entropy(pbest_rows: torch.Tensor, pi_hat: torch.Tensor):
 
 
 def pi_hat_delta(preds: torch.Tensor, point_classes: torch.Tensor,
-                 chunk_h: int = 32) -> torch.Tensor:
-    """sum_h preds[h, :, cls_h] -> (N,) fp32 (the rank-1 pi_hat term)."""
+                 chunk_h: int = 32,
+                 preds_t: torch.Tensor = None) -> torch.Tensor:
+    """sum_h preds[h, :, cls_h] -> (N,) fp32 (the rank-1 pi_hat term).
+
+    preds_t, when given, is the static class-major (H, C, N) mirror of
+    preds: the row-major gather touches one element per 64-B sector
+    (at the random-sector floor, 150 us at the headline shape); the
+    mirror makes the same read a coalesced stream (~15x). The kernels
+    share the accumulation pattern, so the route is bitwise-neutral."""
     if (preds.is_cuda and preds.is_contiguous()
             and preds.dtype in (torch.float32, torch.bfloat16,
                                 torch.float8_e4m3fn)
@@ -152,8 +159,13 @@ def pi_hat_delta(preds: torch.Tensor, point_classes: torch.Tensor,
         kh = -(-512 // bx)
         if kh > 1 and H >= 2 * kh:
             hc = -(-H // kh)
-            out = _ext.pi_hat_delta_part(preds, cls32, hc).sum(0)
+            if preds_t is not None:
+                out = _ext.pi_hat_delta_t_part(preds_t, cls32, hc).sum(0)
+            else:
+                out = _ext.pi_hat_delta_part(preds, cls32, hc).sum(0)
         else:
+            # small-N shapes keep the row-major kernel (its single-chain
+            # accumulation differs bitwise from the 4-acc part kernels)
             out = _ext.pi_hat_delta(preds, cls32)
         if DEBUG:
             _check(out, "pi_hat_delta(kernel)")

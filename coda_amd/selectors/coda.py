@@ -132,6 +132,22 @@ class CODA(ModelSelector):
         self.pi_hat_precision = pi_hat_precision
         self._adjusted = None  # maintained incrementally after init
 
+        # static class-major mirror of the prediction pool (H, C, N):
+        # the per-label rank-1 pi_hat gather preds[h, :, cls_h] reads
+        # one element per 64-B sector in row-major layout (at the
+        # random-sector floor - 150 us headline, 1.2 ms at 10k models);
+        # the mirror makes it a coalesced stream. Built only when HBM
+        # has the pool's size + 16 GB headroom to spare.
+        self._preds_t = None
+        if preds.is_cuda and ops.hip_available() and preds.dim() == 3:
+            try:
+                free, _ = torch.cuda.mem_get_info(self.device)
+                need = preds.numel() * preds.element_size()
+                if free > need + 16 * (1 << 30):
+                    self._preds_t = preds.permute(0, 2, 1).contiguous()
+            except (RuntimeError, torch.OutOfMemoryError):
+                self._preds_t = None
+
         # consensus prior: global mean over H (all-reduce site K1)
         self.comm.all_reduce_(ens_sum)
         pseudo = (ens_sum / self.H).argmax(-1)            # (N,) global pseudo-labels
@@ -656,8 +672,8 @@ class CODA(ModelSelector):
             onehot = torch.nn.functional.one_hot(col, self.C).to(
                 self.dirichlets.dtype)
             self.dirichlets.index_add_(1, y_t, (lr * onehot).unsqueeze(1))
-        delta = ops.pi_hat_delta(self.dataset.preds,
-                                 col) * lr                       # (N,)
+        delta = ops.pi_hat_delta(self.dataset.preds, col,
+                                 preds_t=self._preds_t) * lr     # (N,)
         if self._adjusted.is_cuda and ops.hip_available():
             # fused column update (torch index_add_ over dim 1 with one
             # index runs ~90x slower than this elementwise pass)
@@ -821,7 +837,8 @@ class CODA(ModelSelector):
                 self.update_strength * (~hit).float()
         self._tables_dirty.add(int(true_class))
         self._posterior_version += 1
-        delta = ops.pi_hat_delta(self.dataset.preds, self.classes[:, idx])
+        delta = ops.pi_hat_delta(self.dataset.preds, self.classes[:, idx],
+                                 preds_t=self._preds_t)
         self.comm.all_reduce_(delta)
         self._adjusted[:, int(true_class)] += self.update_strength * delta
         self._row_sums += self.update_strength * delta

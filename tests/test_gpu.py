@@ -775,3 +775,18 @@ class TestSessionKernels:
              * (torch.exp2(t.s_base[3]) * t.weights)).to(
                 torch.bfloat16).float(),
             rtol=1e-2, atol=1e-3)
+
+    def test_pi_hat_delta_class_major_mirror(self):
+        from coda_amd import ops
+        torch.manual_seed(5)
+        for (H, N, C) in [(128, 20_000, 50), (64, 4_000, 33)]:
+            preds = torch.rand(H, N, C, device="cuda")
+            preds_t = preds.permute(0, 2, 1).contiguous()
+            cls = torch.randint(0, C, (H,), device="cuda")
+            a = ops.pi_hat_delta(preds, cls)
+            b = ops.pi_hat_delta(preds, cls, preds_t=preds_t)
+            # same accumulation pattern -> bitwise when the chunked
+            # route is taken on both sides
+            assert torch.equal(a, b)
+            ref = preds[torch.arange(H, device="cuda"), :, cls].sum(0)
+            torch.testing.assert_close(a, ref, rtol=1e-5, atol=1e-4)
