@@ -137,13 +137,14 @@ class CODA(ModelSelector):
         # one element per 64-B sector in row-major layout (at the
         # random-sector floor - 150 us headline, 1.2 ms at 10k models);
         # the mirror makes it a coalesced stream. Built only when HBM
-        # has the pool's size + 16 GB headroom to spare.
+        # has the pool's size + 24 GB headroom to spare (the 128 GB
+        # fp8 million-point pool deliberately skips it).
         self._preds_t = None
         if preds.is_cuda and ops.hip_available() and preds.dim() == 3:
             try:
                 free, _ = torch.cuda.mem_get_info(self.device)
                 need = preds.numel() * preds.element_size()
-                if free > need + 16 * (1 << 30):
+                if free > need + 24 * (1 << 30):
                     self._preds_t = preds.permute(0, 2, 1).contiguous()
             except (RuntimeError, torch.OutOfMemoryError):
                 self._preds_t = None
